@@ -1,0 +1,140 @@
+"""AL experiment driver — query -> update -> re-init -> train -> load-best ->
+test -> save, with resume.
+
+Structural parity with src/main_al.py:43-188 (round loop :145-184, per-phase
+wall-clock prints :160-178, resume branch :123-125, early exit when the pool
+is exhausted :182-184, Comet metric names in the module docstring :24-40).
+"""
+
+import os
+from datetime import date
+from time import time
+
+import numpy as np
+import torch
+
+from .arg_pools import get_arg_pool
+from .cli import get_args
+from .data import get_data
+from .models import get_networks
+from .strategies import get_strategy
+from .utils.checkpoint import load_experiment, save_experiment
+from .utils.logging_setup import setup_logging
+from .utils.pool_init import generate_eval_idxs, generate_init_lb_idxs
+from .utils.tracking import Experiment
+
+
+def main(args):
+    os.makedirs(args.ckpt_path, exist_ok=True)
+
+    train_args = get_arg_pool(args.arg_pool)[args.dataset]
+
+    imbalance_args = {"imbalance_type": args.imbalance_type,
+                      "imbalance_factor": args.imbalance_factor,
+                      "imbalance_seed": args.imbalance_seed}
+
+    train_set, test_set, al_set = get_data(data_path=args.dataset_dir,
+                                           data_name=args.dataset, supervised=True,
+                                           debug_mode=args.debug_mode,
+                                           imbalance_args=imbalance_args)
+
+    n_devices = torch.cuda.device_count()
+    print(f"Using {n_devices} GPU devices.")
+    net = get_networks(args.dataset, args.model)
+    net.freeze_feature = args.freeze_feature
+
+    eval_idxs = generate_eval_idxs(train_set, train_args["eval_split"], random_seed=99)
+
+    init_pool_size = args.init_pool_size
+    if init_pool_size == -1:
+        init_pool_size = int(args.round_budget)
+
+    if init_pool_size == 0:
+        init_lb_idxs = np.array([], dtype=np.int64)
+    else:
+        init_lb_idxs = generate_init_lb_idxs(train_set, eval_idxs, init_pool_size,
+                                             init_pool_type=args.init_pool_type,
+                                             random_seed=98)
+
+    if args.debug_mode:
+        # tiny fixed pool (main_al.py:87-92)
+        init_lb_idxs = np.arange(5) if init_pool_size != 0 else np.array([], dtype=np.int64)
+        eval_idxs = np.arange(15, 20)
+        test_set = torch.utils.data.Subset(test_set, list(range(10)))
+        test_set.num_classes = al_set.num_classes
+
+    if args.world_size is None:
+        args.world_size = max(1, n_devices)
+
+    if not args.resume_training:
+        experiment = Experiment(project_name=args.project_name,
+                                disabled=not args.enable_comet, log_dir=args.log_dir)
+        experiment.add_tag(args.exp_name)
+        experiment.add_tag(args.strategy)
+        exp_hash = os.path.basename(os.path.normpath(experiment.url))[:9]
+        if exp_hash == ".":
+            exp_hash = "debug"
+        if not args.exp_hash:
+            args.exp_hash = exp_hash
+        experiment.set_name(args.exp_name)
+        experiment.log_parameters(vars(args))
+
+        strategy_cls = get_strategy(args.strategy)
+        strategy = strategy_cls(train_set, al_set, net, train_args, eval_idxs,
+                                experiment, test_set, **vars(args))
+        strategy.update(init_lb_idxs, len(init_lb_idxs))
+        start_round = 0
+    else:
+        strategy, start_round, experiment = load_experiment(args)
+
+    strategy.world_size = args.world_size
+
+    os.makedirs(args.log_dir, exist_ok=True)
+    today = date.today()
+    log_filename = f"{args.exp_hash}_{today.month:02d}{today.day:02d}.log"
+    logger = setup_logging(args.log_dir, log_filename)
+    logger.info(f"Experiment Name: {args.exp_name}")
+    logger.info(f"Dataset: {args.dataset}")
+    logger.info(f"Strategy: {args.strategy}")
+    logger.info(f"Budget used before starting: {len(init_lb_idxs)}")
+    logger.info(f"Log file name: {log_filename}")
+
+    for rd in range(start_round, args.rounds):
+        strategy.round = rd
+        logger.info(f"Active Learning Round {rd} start.")
+
+        al_round_0 = rd == 0 and init_pool_size == 0
+        if rd > 0 or al_round_0:
+            if al_round_0:
+                strategy.init_network_weights()
+            t0 = time()
+            labeled_idxs, cur_cost = strategy.query(args.round_budget)
+            print(f"Rd {rd} query_time is {time() - t0}")
+            strategy.update(labeled_idxs, cur_cost)
+
+        t0 = time()
+        strategy.init_network_weights()
+        print(f"Rd {rd} init_network_weights_time is {time() - t0}")
+
+        t0 = time()
+        strategy.train()
+        print(f"Rd {rd} train_time is {time() - t0}")
+
+        t0 = time()
+        strategy.load_best_ckpt()
+        print(f"Rd {rd} load_best_ckpt_time is {time() - t0}")
+
+        t0 = time()
+        strategy.test()
+        print(f"Rd {rd} test_time is {time() - t0}")
+
+        save_experiment(strategy, args, logger)
+        args.resume_training = True
+        if len(strategy.available_query_idxs()) == 0:
+            logger.info("Finished querying all Images!")
+            break
+    return strategy
+
+
+if __name__ == "__main__":
+    main(get_args())

@@ -1,0 +1,265 @@
+"""Autograd-integrated NHWC ops: conv / BN(+residual)+ReLU / pooling.
+
+Layout contract (MI355X-native, chosen for MFMA/LDS tiling and coalescing):
+  activations: (N, H, W, C) contiguous, bf16 on GPU / fp32 on CPU
+  conv weight: (K, R, S, C) fp32 master parameter; cast to the activation
+               dtype inside the Function (so weight gradients are produced in
+               fp32 straight from the accumulator — no bf16 round-trip).
+
+Reference counterparts: these are the kernels the reference reaches through
+cuDNN/ATen: conv fwd/bwd-data/bwd-weight (strategy.py:268,270),
+fused BN+ReLU incl. the frozen-stats eval variant (strategy.py:366-367),
+SyncBN stats exchange (strategy.py:292), maxpool/avgpool. The BN op
+additionally fuses the residual add + ReLU that ends every ResNet block, so a
+bottleneck tail is one kernel instead of three memory passes.
+
+GPU dispatch goes to active_learning_amd._C (HIP/gfx950); CPU fallback uses
+plain PyTorch and defines the semantics the kernels are tested against.
+"""
+
+import torch
+import torch.distributed as dist
+import torch.nn.functional as F
+from torch.autograd import Function
+
+from .extension import require_extension
+
+
+def _cpu_conv_args(x_nhwc, w_krsc):
+    return x_nhwc.permute(0, 3, 1, 2), w_krsc.permute(0, 3, 1, 2)
+
+
+class Conv2dNHWC(Function):
+    """y[N,P,Q,K] = conv(x[N,H,W,C], w[K,R,S,C]; stride, pad), no bias
+    (ResNet convs carry no bias; BN follows)."""
+
+    @staticmethod
+    def forward(ctx, x, weight, stride, padding):
+        w_c = weight.to(x.dtype)  # fp32 master -> compute dtype
+        ctx.save_for_backward(x, w_c)
+        ctx.stride, ctx.padding = stride, padding
+        ctx.weight_dtype = weight.dtype
+        if x.is_cuda:
+            ext = require_extension()
+            return ext.conv2d_fwd(x, w_c, stride, padding)
+        xc, wc = _cpu_conv_args(x, w_c)
+        y = F.conv2d(xc.float(), wc.float(), stride=stride, padding=padding)
+        return y.to(x.dtype).permute(0, 2, 3, 1).contiguous()
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w_c = ctx.saved_tensors
+        dy = dy.contiguous()
+        dx = dw = None
+        if x.is_cuda:
+            ext = require_extension()
+            if ctx.needs_input_grad[0]:
+                dx = ext.conv2d_bwd_data(dy, w_c, ctx.stride, ctx.padding,
+                                         x.shape[1], x.shape[2])
+            if ctx.needs_input_grad[1]:
+                dw = ext.conv2d_bwd_weight(dy, x, w_c.shape[1], w_c.shape[2],
+                                           ctx.stride, ctx.padding)
+        else:
+            xc, wc = _cpu_conv_args(x, w_c)
+            dyc = dy.permute(0, 3, 1, 2).float()
+            if ctx.needs_input_grad[0]:
+                dx = torch.nn.grad.conv2d_input(list(xc.shape), wc.float(), dyc,
+                                                stride=ctx.stride, padding=ctx.padding)
+                dx = dx.to(x.dtype).permute(0, 2, 3, 1).contiguous()
+            if ctx.needs_input_grad[1]:
+                dw = torch.nn.grad.conv2d_weight(xc.float(), list(wc.shape), dyc,
+                                                 stride=ctx.stride, padding=ctx.padding)
+                dw = dw.permute(0, 2, 3, 1).contiguous()
+        if dw is not None:
+            dw = dw.to(ctx.weight_dtype)
+        return dx, dw, None, None
+
+
+def conv2d(x, weight, stride=1, padding=0):
+    return Conv2dNHWC.apply(x, weight, stride, padding)
+
+
+class BatchNormAct(Function):
+    """Fused BatchNorm (+ optional residual add + optional ReLU) over NHWC.
+
+    Three stat modes:
+    * batch stats (training)          — per-channel mean/var over N*H*W
+    * synced batch stats (SyncBN)     — partial sums all-reduced over `pg`
+      (reference: SyncBatchNorm conversion at strategy.py:292)
+    * frozen running stats            — the reference's net.eval()-while-
+      training semantics (strategy.py:366-367): grads flow, stats frozen.
+
+    y = act(bn(x) + residual); residual may be None.
+    """
+
+    @staticmethod
+    def forward(ctx, x, weight, bias, running_mean, running_var, use_batch_stats,
+                momentum, eps, relu, residual, pg):
+        n_local = x.numel() // x.shape[-1]
+        if use_batch_stats:
+            if x.is_cuda:
+                ext = require_extension()
+                s, ss = ext.bn_stats(x)  # fp32 per-channel sum / sum of squares
+            else:
+                xf = x.float()
+                s = xf.sum(dim=(0, 1, 2))
+                ss = (xf * xf).sum(dim=(0, 1, 2))
+            count = torch.tensor([float(n_local)], device=x.device)
+            if pg is not None and dist.is_initialized() and dist.get_world_size(pg) > 1:
+                packed = torch.cat([s, ss, count])
+                dist.all_reduce(packed, group=pg)
+                s, ss, count = packed[:len(s)], packed[len(s):2 * len(s)], packed[-1:]
+            n = float(count.item())
+            mean = s / n
+            var = (ss / n - mean * mean).clamp_min_(0)
+            invstd = (var + eps).rsqrt()
+            with torch.no_grad():
+                if running_mean is not None:
+                    unbiased = var * (n / max(n - 1, 1))
+                    running_mean.mul_(1 - momentum).add_(mean, alpha=momentum)
+                    running_var.mul_(1 - momentum).add_(unbiased, alpha=momentum)
+        else:
+            mean = running_mean.to(torch.float32)
+            invstd = (running_var.to(torch.float32) + eps).rsqrt()
+            n = float(n_local)
+
+        if x.is_cuda:
+            ext = require_extension()
+            y = ext.bn_norm_fwd(x, mean, invstd, weight, bias, relu,
+                                residual if residual is not None else x.new_empty(0))
+        else:
+            xf = x.float()
+            y = (xf - mean) * invstd * weight + bias
+            if residual is not None:
+                y = y + residual.float()
+            if relu:
+                y = F.relu(y)
+            y = y.to(x.dtype)
+        ctx.save_for_backward(x, y, weight, mean, invstd)
+        ctx.use_batch_stats = use_batch_stats
+        ctx.relu = relu
+        ctx.has_residual = residual is not None
+        ctx.pg = pg
+        ctx.n = n
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, y, weight, mean, invstd = ctx.saved_tensors
+        dy = dy.contiguous()
+        if x.is_cuda:
+            ext = require_extension()
+            # reduce pass: per-channel sums of dy~ and dy~*xhat (dy~ = mask * dy)
+            sum_dy, sum_dy_xhat = ext.bn_bwd_reduce(dy, x, y, mean, invstd, ctx.relu)
+        else:
+            dyf = dy.float()
+            if ctx.relu:
+                dyf = dyf * (y > 0).float()
+            xhat = (x.float() - mean) * invstd
+            sum_dy = dyf.sum(dim=(0, 1, 2))
+            sum_dy_xhat = (dyf * xhat).sum(dim=(0, 1, 2))
+
+        dgamma = sum_dy_xhat.clone()
+        dbeta = sum_dy.clone()
+
+        if ctx.use_batch_stats and ctx.pg is not None and dist.is_initialized() \
+                and dist.get_world_size(ctx.pg) > 1:
+            packed = torch.cat([sum_dy, sum_dy_xhat])
+            dist.all_reduce(packed, group=ctx.pg)
+            sum_dy, sum_dy_xhat = packed[:len(sum_dy)], packed[len(sum_dy):]
+        # ctx.n is already the GLOBAL element count when stats were synced
+        # (the forward all-reduced the counts), else the local count.
+        n_global = ctx.n
+
+        dres = None
+        if x.is_cuda:
+            ext = require_extension()
+            dx, dres_t = ext.bn_bwd(dy, x, y, mean, invstd, weight, sum_dy, sum_dy_xhat,
+                                    n_global, ctx.use_batch_stats, ctx.relu,
+                                    ctx.has_residual)
+            if ctx.has_residual:
+                dres = dres_t
+        else:
+            dyf = dy.float()
+            if ctx.relu:
+                dyf = dyf * (y > 0).float()
+            if ctx.has_residual:
+                dres = dyf.to(x.dtype)
+            g = weight * invstd
+            if ctx.use_batch_stats:
+                xhat = (x.float() - mean) * invstd
+                dx = g * (dyf - sum_dy / n_global - xhat * (sum_dy_xhat / n_global))
+            else:
+                dx = g * dyf
+            dx = dx.to(x.dtype)
+        return dx, dgamma, dbeta, None, None, None, None, None, None, dres, None
+
+
+def batch_norm_act(x, weight, bias, running_mean, running_var, use_batch_stats,
+                   momentum=0.1, eps=1e-5, relu=True, residual=None, pg=None):
+    return BatchNormAct.apply(x, weight, bias, running_mean, running_var,
+                              use_batch_stats, momentum, eps, relu, residual, pg)
+
+
+class MaxPool2dNHWC(Function):
+    @staticmethod
+    def forward(ctx, x, kernel, stride, padding):
+        if x.is_cuda:
+            ext = require_extension()
+            y, idx = ext.maxpool2d_fwd(x, kernel, stride, padding)
+        else:
+            xc = x.permute(0, 3, 1, 2).float()
+            y, idx = F.max_pool2d(xc, kernel, stride, padding, return_indices=True)
+            y = y.to(x.dtype).permute(0, 2, 3, 1).contiguous()
+        ctx.save_for_backward(idx)
+        ctx.x_shape = x.shape
+        ctx.params = (kernel, stride, padding)
+        ctx.x_dtype = x.dtype
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (idx,) = ctx.saved_tensors
+        kernel, stride, padding = ctx.params
+        dy = dy.contiguous()
+        if dy.is_cuda:
+            ext = require_extension()
+            dx = ext.maxpool2d_bwd(dy, idx, ctx.x_shape[1], ctx.x_shape[2])
+        else:
+            dyc = dy.permute(0, 3, 1, 2).float()
+            n, h, w, c = ctx.x_shape
+            # scatter-ADD (max_unpool2d overwrites duplicate indices; gradients
+            # must accumulate when windows share an argmax)
+            dx_flat = torch.zeros(n, c, h * w, dtype=dyc.dtype)
+            dx_flat.scatter_add_(2, idx.reshape(n, c, -1), dyc.reshape(n, c, -1))
+            dx = dx_flat.reshape(n, c, h, w).to(ctx.x_dtype)
+            dx = dx.permute(0, 2, 3, 1).contiguous()
+        return dx, None, None, None
+
+
+def max_pool2d(x, kernel=3, stride=2, padding=1):
+    return MaxPool2dNHWC.apply(x, kernel, stride, padding)
+
+
+class GlobalAvgPoolNHWC(Function):
+    """(N,H,W,C) -> (N,C) mean over H,W (ResNet head)."""
+
+    @staticmethod
+    def forward(ctx, x):
+        ctx.x_shape = x.shape
+        ctx.x_dtype = x.dtype
+        if x.is_cuda:
+            ext = require_extension()
+            return ext.global_avg_pool(x)
+        return x.float().mean(dim=(1, 2)).to(x.dtype)
+
+    @staticmethod
+    def backward(ctx, dy):
+        n, h, w, c = ctx.x_shape
+        scale = 1.0 / (h * w)
+        dx = (dy.float() * scale)[:, None, None, :].expand(n, h, w, c)
+        return dx.to(ctx.x_dtype).contiguous()
+
+
+def global_avg_pool(x):
+    return GlobalAvgPoolNHWC.apply(x)

@@ -1,0 +1,219 @@
+"""Query-scoring math: fused softmax scores, MASE margins, BADGE Gram
+factorization, pairwise distances and the greedy k-center loop.
+
+Reference counterparts:
+  confidence/margin: softmax + top-k over logits (confidence_sampler.py:27-36,
+    margin_sampler.py:29-39) — here a single fused kernel pass on GPU.
+  MASE (mase_sampler.py:59-79): the reference materializes weight_delta
+    (B,C,M) and epsilon (B,C,M). Closed form used here: the L2 distance from
+    embedding e to the decision boundary between predicted class p and class c
+    is |logit_p - logit_c| / ||w_p - w_c|| — needs only the logits and the
+    C x C weight Gram matrix; no (B,C,M) intermediate exists.
+  BADGE (badge_sampler.py:36-48): gradient embedding g_i = a_i (x) e_i with
+    a_i = softmax(l_i) - onehot(argmax l_i). Since <a (x) e, a' (x) e'> =
+    (a.a')(e.e'), the N x N Gram of gradient embeddings factorizes into the
+    elementwise product of two small Grams; the (B, C*M) embedding is never
+    materialized. The adaptive-avg-pool variant (badge_sampler.py:41-44)
+    pools the outer product over a (16, 32) grid, which equals the outer
+    product of the pooled vectors, so the factorization survives pooling.
+  Coreset greedy k-center (coreset_sampler.py:66-105): the reference keeps the
+    N x N matrix on GPU but runs the selection loop on the host with numpy.
+    Here the distance matrix, the running min-distance vector and the
+    argmax/sampling all stay on device (HBM-resident per BASELINE.json).
+"""
+
+import numpy as np
+import torch
+import torch.nn.functional as F
+
+from .extension import require_extension
+
+
+# --------------------------------------------------------------------------- #
+# softmax-based uncertainty scores
+# --------------------------------------------------------------------------- #
+
+def softmax_scores(logits: torch.Tensor):
+    """Return (top1_prob, margin = p1 - p2, entropy) per row, one fused pass
+    on GPU."""
+    if logits.is_cuda:
+        ext = require_extension()
+        out = ext.softmax_scores(logits)
+        return out[0], out[1], out[2]
+    p = F.softmax(logits.float(), dim=1)
+    top2 = torch.topk(p, k=min(2, p.shape[1]), dim=1).values
+    ent = -(p * torch.log(p.clamp_min(1e-12))).sum(dim=1)
+    margin = top2[:, 0] - (top2[:, 1] if top2.shape[1] > 1 else torch.zeros_like(top2[:, 0]))
+    return top2[:, 0], margin, ent
+
+
+def confidence_scores(logits):
+    return softmax_scores(logits)[0]
+
+
+def margin_scores(logits):
+    return softmax_scores(logits)[1]
+
+
+def entropy_scores(logits):
+    return softmax_scores(logits)[2]
+
+
+# --------------------------------------------------------------------------- #
+# MASE margins
+# --------------------------------------------------------------------------- #
+
+def mase_margins(logits: torch.Tensor, weight: torch.Tensor):
+    """Distances to pairwise decision boundaries in feature space.
+
+    radius[i, c] = |logit_i[pred_i] - logit_i[c]| / ||w_pred_i - w_c||,
+    radius[i, pred_i] = +inf (reference gets inf from 0/0 -> nan -> inf,
+    mase_sampler.py:76-78). Returns (min_margins (B,), radius (B,C), pred (B,)).
+
+    Bias contributes through the logits (logit = e.w + b), identical to the
+    reference's lam formula with bias_delta folded in (mase_sampler.py:66-79).
+    """
+    logits = logits.float()
+    w = weight.float()
+    pred = logits.argmax(dim=1)
+    gram = w @ w.t()  # (C, C)
+    sq_norms = gram.diagonal()
+    # ||w_p - w_c||^2 for every (p, c)
+    d2 = (sq_norms[:, None] + sq_norms[None, :] - 2 * gram).clamp_min_(0)
+    denom = d2[pred].sqrt()  # (B, C)
+    num = (logits.gather(1, pred[:, None]) - logits).abs()
+    radius = num / denom
+    radius[torch.arange(len(pred), device=pred.device), pred] = float("inf")
+    radius = torch.nan_to_num(radius, nan=float("inf"))
+    min_margins = radius.min(dim=1).values
+    return min_margins, radius, pred
+
+
+# --------------------------------------------------------------------------- #
+# pairwise distances / BADGE Gram factorization
+# --------------------------------------------------------------------------- #
+
+def pairwise_sqdist(features: torch.Tensor, chunk=8192) -> torch.Tensor:
+    """||x_i - x_j||^2 as norms + GEMM (coreset_sampler.py:59-64), chunked so
+    the N x M read stays streaming while the N x N output is materialized
+    (fp32; N=130k -> 68 GB, resident in the 288 GB HBM)."""
+    f = features.float()
+    n = f.shape[0]
+    sq = (f * f).sum(dim=1)
+    out = torch.empty((n, n), dtype=torch.float32, device=f.device)
+    for i0 in range(0, n, chunk):
+        i1 = min(i0 + chunk, n)
+        dp = f[i0:i1] @ f.t()
+        out[i0:i1] = sq[i0:i1, None] + sq[None, :] - 2 * dp
+    return out
+
+
+def badge_pairwise_sqdist(a_vec: torch.Tensor, e_vec: torch.Tensor) -> torch.Tensor:
+    """N x N squared distances between gradient embeddings g_i = a_i (x) e_i
+    without materializing them: <g_i, g_j> = (a_i.a_j)(e_i.e_j)."""
+    a = a_vec.float()
+    e = e_vec.float()
+    ga = a @ a.t()
+    ge = e @ e.t()
+    g = ga * ge
+    d = g.diagonal()
+    return d[:, None] + d[None, :] - 2 * g
+
+
+def badge_vectors(logits: torch.Tensor, embedding: torch.Tensor, pool=None):
+    """(a_i, e_i) pair defining the BADGE gradient embedding; with pool=(Ph,Pw)
+    both vectors are adaptive-avg-pooled (outer product of pooled vectors ==
+    pooled outer product, badge_sampler.py:41-44)."""
+    p = F.softmax(logits.float(), dim=1)
+    yhat = logits.argmax(dim=1)
+    a = p.clone()
+    a[torch.arange(len(yhat), device=logits.device), yhat] -= 1.0
+    e = embedding.float()
+    if pool is not None:
+        ph, pw = pool
+        a = F.adaptive_avg_pool1d(a[:, None, :], ph)[:, 0, :]
+        e = F.adaptive_avg_pool1d(e[:, None, :], pw)[:, 0, :]
+    return a, e
+
+
+# --------------------------------------------------------------------------- #
+# greedy k-center (device-resident)
+# --------------------------------------------------------------------------- #
+
+def kcenter_greedy(dist: torch.Tensor, labeled: torch.Tensor, budget: int,
+                   randomize=False, generator=None):
+    """Greedy farthest-point selection over a pairwise-distance matrix.
+
+    dist: (N, N) on device; labeled: (N,) bool on device; returns list of N-space
+    indices. Semantics mirror coreset_sampler.py:66-105, including:
+      * no labeled points yet: deterministic -> argmin of row-max
+        (coreset_sampler.py:100), randomized -> uniform choice (:97);
+      * randomize=True: probability proportional to clamp(min_dist, 0) with
+        already-selected masked to 0 (k-means++ seeding, :81-92).
+    Everything stays on device; per-iteration work is one column min-update
+    plus one argmax/multinomial.
+    """
+    n = dist.shape[0]
+    labeled = labeled.clone()
+    selected = []
+    neg_inf = torch.tensor(float("-inf"), device=dist.device)
+
+    if labeled.any():
+        min_dist = _masked_col_min(dist, labeled)
+    else:
+        if randomize:
+            j = int(torch.randint(n, (1,)).item())
+        else:
+            j = int(dist.max(dim=1).values.argmin().item())
+        selected.append(j)
+        labeled[j] = True
+        min_dist = dist[:, j].clone()
+
+    while len(selected) < budget:
+        if randomize:
+            probs = min_dist.clamp_min(0.0)
+            probs = torch.where(labeled, torch.zeros_like(probs), probs)
+            total = probs.sum()
+            if total <= 0 or not torch.isfinite(total):
+                # reference jitters by +1e-5 until valid (coreset_sampler.py:85-92);
+                # equivalent: fall back to uniform over unlabeled
+                probs = (~labeled).float()
+                total = probs.sum()
+            j = int(torch.multinomial(probs / total, 1).item())
+        else:
+            scores = torch.where(labeled, neg_inf, min_dist)
+            j = int(scores.argmax().item())
+        selected.append(j)
+        labeled[j] = True
+        min_dist = torch.minimum(min_dist, dist[:, j])
+    return selected
+
+
+def _masked_col_min(dist, labeled_mask, chunk=16384):
+    """min over labeled columns, chunked (avoids a N x N_l gather copy)."""
+    cols = labeled_mask.nonzero(as_tuple=True)[0]
+    out = None
+    for c0 in range(0, len(cols), chunk):
+        part = dist[:, cols[c0:c0 + chunk]].min(dim=1).values
+        out = part if out is None else torch.minimum(out, part)
+    return out
+
+
+# --------------------------------------------------------------------------- #
+# class centroids (BalancingSampler)
+# --------------------------------------------------------------------------- #
+
+def class_centroids(embeddings: torch.Tensor, labels: torch.Tensor, num_classes: int):
+    """Per-class mean embedding via scatter-add; classes with no samples get a
+    zero centroid (balancing_sampler.py:87-96)."""
+    e = embeddings.float()
+    out = torch.zeros(num_classes, e.shape[1], device=e.device)
+    out.index_add_(0, labels, e)
+    counts = torch.bincount(labels, minlength=num_classes).float()
+    return out / (counts[:, None] + 1e-5)
+
+
+def sqdist_to_centers(x: torch.Tensor, centers: torch.Tensor):
+    x = x.float()
+    c = centers.float()
+    return ((x * x).sum(1)[:, None] + (c * c).sum(1)[None, :] - 2 * x @ c.t())

@@ -1,0 +1,195 @@
+"""Multi-process CPU tests of the distributed path (gloo backend, world=2):
+BucketedDDP gradient averaging, unused-param handling, SyncBN statistics,
+and packed eval gathering. The same code paths run over RCCL on MI355X."""
+
+import os
+
+import numpy as np
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+from active_learning_amd.parallel import get_free_tcp_port
+
+WORLD = 2
+
+
+def _run_dist(fn, world=WORLD, args=()):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(get_free_tcp_port())
+    ctx = mp.get_context("spawn")
+    procs = []
+    err_q = ctx.SimpleQueue()
+    for rank in range(world):
+        p = ctx.Process(target=_entry, args=(fn, rank, world, err_q, args))
+        p.start()
+        procs.append(p)
+    for p in procs:
+        p.join(timeout=240)
+    fails = [p.exitcode for p in procs if p.exitcode != 0]
+    if fails:
+        msg = err_q.get() if not err_q.empty() else f"exit codes {fails}"
+        raise AssertionError(f"distributed test failed: {msg}")
+
+
+def _entry(fn, rank, world, err_q, args):
+    try:
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+        torch.manual_seed(1234)  # same model init everywhere
+        fn(rank, world, *args)
+    except Exception as e:  # pragma: no cover
+        import traceback
+        err_q.put(traceback.format_exc())
+        raise
+    finally:
+        if dist.is_initialized():
+            dist.destroy_process_group()
+
+
+# --------------------------------------------------------------------------- #
+
+def _ddp_grad_check(rank, world):
+    from active_learning_amd.parallel import BucketedDDP
+    torch.manual_seed(7)
+    model = torch.nn.Sequential(torch.nn.Linear(10, 32), torch.nn.ReLU(),
+                                torch.nn.Linear(32, 4))
+    ddp = BucketedDDP(model, bucket_cap_mb=0.0001)  # force several buckets
+    torch.manual_seed(100 + rank)
+    x = torch.randn(8, 10)
+    y = torch.randint(0, 4, (8,))
+    out = ddp(x)
+    loss = torch.nn.functional.cross_entropy(out, y)
+    loss.backward()
+    ddp.finalize_grads()
+
+    # reference: average of per-rank grads on a replica
+    torch.manual_seed(7)
+    ref = torch.nn.Sequential(torch.nn.Linear(10, 32), torch.nn.ReLU(),
+                              torch.nn.Linear(32, 4))
+    grads_accum = [torch.zeros_like(p) for p in ref.parameters()]
+    for r in range(world):
+        torch.manual_seed(100 + r)
+        xr = torch.randn(8, 10)
+        yr = torch.randint(0, 4, (8,))
+        for p in ref.parameters():
+            p.grad = None
+        lr_ = torch.nn.functional.cross_entropy(ref(xr), yr)
+        lr_.backward()
+        for g, p in zip(grads_accum, ref.parameters()):
+            g += p.grad / world
+    for p, g in zip(ddp.module.parameters(), grads_accum):
+        assert torch.allclose(p.grad, g, atol=1e-6), "DDP grad mismatch"
+
+
+def test_bucketed_ddp_grad_averaging():
+    _run_dist(_ddp_grad_check)
+
+
+def _ddp_unused_param_check(rank, world):
+    from active_learning_amd.parallel import BucketedDDP
+
+    class Partial(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.used = torch.nn.Linear(5, 5)
+            self.unused = torch.nn.Linear(5, 5)
+
+        def forward(self, x):
+            return self.used(x)
+
+    torch.manual_seed(3)
+    m = Partial()
+    ddp = BucketedDDP(m, bucket_cap_mb=0.0001)
+    x = torch.randn(4, 5)
+    ddp(x).sum().backward()
+    ddp.finalize_grads()  # must not hang; unused slots zero-filled
+    assert m.used.weight.grad is not None
+
+
+def test_bucketed_ddp_unused_params():
+    _run_dist(_ddp_unused_param_check)
+
+
+def _syncbn_check(rank, world):
+    from active_learning_amd.models.layers import BatchNormAct2d
+    from active_learning_amd.parallel import convert_sync_batchnorm
+
+    torch.manual_seed(5)
+    full = torch.randn(8, 4, 4, 3)  # global batch in NHWC
+    shard = full[rank * 4:(rank + 1) * 4]
+
+    bn = BatchNormAct2d(3, relu=False)
+    convert_sync_batchnorm(bn)
+    bn.train()
+    shard = shard.clone().requires_grad_(True)
+    y = bn(shard)
+
+    # reference: full-batch BN on one process
+    ref_bn = torch.nn.BatchNorm2d(3)
+    ref_in = full.permute(0, 3, 1, 2).clone().requires_grad_(True)
+    ref_y = ref_bn(ref_in)
+    ref_y_nhwc = ref_y.permute(0, 2, 3, 1)
+    assert torch.allclose(y, ref_y_nhwc[rank * 4:(rank + 1) * 4], atol=1e-5), \
+        "SyncBN forward mismatch"
+    assert torch.allclose(bn.running_mean, ref_bn.running_mean, atol=1e-6)
+    assert torch.allclose(bn.running_var, ref_bn.running_var, atol=1e-5)
+
+    # backward: per-rank loss = mean over LOCAL shard, matching DDP convention
+    dy = torch.ones_like(y)
+    y.backward(dy)
+    ref_y_nhwc.backward(torch.ones_like(ref_y_nhwc))
+    ref_dx = ref_in.grad.permute(0, 2, 3, 1)[rank * 4:(rank + 1) * 4]
+    assert torch.allclose(shard.grad, ref_dx, atol=1e-5), "SyncBN backward mismatch"
+
+
+def test_syncbn_matches_fullbatch():
+    _run_dist(_syncbn_check)
+
+
+def _eval_gather_check(rank, world):
+    from active_learning_amd.utils.evaluation import gather_parallel_eval
+    d = {"count": 10.0 + rank, "top_1_correct_count": 5.0, "top_5_correct_count": 8.0,
+         "count_byclass": torch.tensor([5.0, 5.0 + rank]),
+         "corrects_byclass": torch.tensor([2.0, 3.0])}
+    top1, top5, byclass = gather_parallel_eval(d, world, torch.device("cpu"))
+    assert abs(top1.item() - 10.0 / 21.0) < 1e-6
+    assert abs(top5.item() - 16.0 / 21.0) < 1e-6
+    assert torch.allclose(byclass, torch.tensor([4.0 / 10.0, 6.0 / 11.0]))
+
+
+def test_gather_parallel_eval():
+    _run_dist(_eval_gather_check)
+
+
+def _strategy_train_check(rank, world):
+    """Full Strategy.parallel_train_fn over gloo with world=2."""
+    from active_learning_amd.strategies import RandomSampler
+    import helpers
+    s = helpers.make_strategy(RandomSampler)
+    s.world_size = world
+    s.backend = "gloo"
+    s.update(np.arange(20), 20)
+    # emulate mp.spawn path: pg already initialized by _entry
+    s.parallel_train_fn(rank)
+    assert os.path.exists(s.generate_weight_paths()["best_ckpt"])
+
+
+def test_strategy_parallel_train_gloo():
+    # parallel_train_fn calls _init_distributed which would re-init; run the
+    # whole thing through Strategy.train-like spawn instead
+    _run_dist(_strategy_train_spawned)
+
+
+def _strategy_train_spawned(rank, world):
+    from active_learning_amd.strategies import RandomSampler
+    import helpers
+    torch.manual_seed(42)
+    s = helpers.make_strategy(RandomSampler)
+    s.world_size = world
+    s.backend = "gloo"
+    s.update(np.arange(20), 20)
+    s._init_distributed = lambda r: None  # group already up via _entry
+    s.parallel_train_fn(rank)
+    if rank == 0:
+        assert os.path.exists(s.generate_weight_paths()["best_ckpt"])
