@@ -18,6 +18,7 @@ def load_extension():
     if _C is not None:
         return _C
     try:
+        import torch  # noqa: F401 — brings libtorch/libc10 into the process
         _C = importlib.import_module("active_learning_amd._C")
     except ImportError as e:  # keep the error for diagnostics
         _LOAD_ERROR = e

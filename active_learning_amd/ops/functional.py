@@ -224,7 +224,8 @@ class MaxPool2dNHWC(Function):
         dy = dy.contiguous()
         if dy.is_cuda:
             ext = require_extension()
-            dx = ext.maxpool2d_bwd(dy, idx, ctx.x_shape[1], ctx.x_shape[2])
+            dx = ext.maxpool2d_bwd(dy, idx, ctx.x_shape[1], ctx.x_shape[2],
+                                   kernel, stride, padding)
         else:
             dyc = dy.permute(0, 3, 1, 2).float()
             n, h, w, c = ctx.x_shape

@@ -1,0 +1,293 @@
+// Implicit-GEMM convolution for NHWC bf16 on gfx950 (CDNA4 MFMA).
+//
+// Replaces the cuDNN conv kernels the reference reaches through
+// torchvision/resnet forward+backward.
+//
+// Design (MI355X-first):
+//  * forward:   y[M=N*P*Q, K] = im2col(x)[M, RSC] @ W^T      (W is (K,R,S,C))
+//  * bwd-data:  dx[M=N*H*W, C] = scat(dY)[M, RSK] @ Wt       (Wt is (C,R,S,K),
+//               k fastest in the contraction, pre-permuted host-side — weights
+//               are tiny next to activations)
+//    Both are the same GEMM with different per-row gather rules, so one
+//    templated kernel (MODE) serves both; a transposed conv's forward IS the
+//    bwd-data computation (models/layers.py).
+//  * 128x128x64 tile, 4 waves (2x2), 64x64 per wave, mfma_f32_16x16x32_bf16,
+//    fp32 accumulators.
+//  * global->LDS staging via __builtin_amdgcn_global_load_lds width 16
+//    (async, no VGPR round trip); LDS image is lane-linear so the
+//    bank-conflict XOR swizzle (slot ^= row&7) is applied to the per-lane
+//    SOURCE address and re-applied on the ds_read_b128 side (guide §5.4
+//    rule 21: both-sides-or-neither).
+//  * out-of-bounds / padding gathers read a device zero page: no branches in
+//    the staging inner loop, no predication on the DMA.
+//  * requires C % 8 == 0 and contraction-dim % 64 == 0 (all non-stem ResNet
+//    and VAE shapes); other shapes take the direct fallback kernels below.
+//
+// Fragment layouts (verified by tests/test_kernels_gpu.py layout probe):
+//   A: lane l holds A[row = l&15][k = (l>>4)*8 + j], j=0..7
+//   B: lane l holds B[k = (l>>4)*8 + j][col = l&15]
+//   C/D: lane l, reg r -> row = (l>>4)*4 + r, col = l&15
+
+#include "al_common.h"
+
+typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
+
+#define MODE_FWD 0
+#define MODE_BWD_DATA 1
+
+struct ConvShape {
+  int N, H, W, C;       // input (or dx) spatial shape
+  int K;                // output channels (fwd) / dY channels (bwd-data)
+  int R, S;             // filter
+  int P, Q;             // output spatial (fwd) / dY spatial (bwd-data)
+  int stride, pad;
+  long M;               // GEMM rows: N*P*Q (fwd) or N*H*W (bwd)
+  int Nout;             // GEMM cols: K (fwd) or C (bwd)
+  int KD;               // contraction: R*S*C (fwd) or R*S*K (bwd)
+};
+
+// ---------------------------------------------------------------------------
+// per-mode gather: global source pointer for a 16-byte (8 bf16) chunk of the
+// A matrix at (row m, kd..kd+7). Returns the zero page when out of bounds.
+// ---------------------------------------------------------------------------
+
+template <int MODE>
+AL_DEV const bf16* a_chunk_ptr(const bf16* __restrict__ a, const bf16* zero,
+                               const ConvShape& sh, long m, int kd) {
+  if (m >= sh.M || kd >= sh.KD) return zero;
+  if (MODE == MODE_FWD) {
+    // m -> (n,p,q); kd = (r*S + s)*C + c
+    const int q = (int)(m % sh.Q);
+    long t = m / sh.Q;
+    const int p = (int)(t % sh.P);
+    const int n = (int)(t / sh.P);
+    const int c = kd % sh.C;
+    const int rs = kd / sh.C;
+    const int s = rs % sh.S;
+    const int r = rs / sh.S;
+    const int h = p * sh.stride + r - sh.pad;
+    const int w = q * sh.stride + s - sh.pad;
+    if (h < 0 || h >= sh.H || w < 0 || w >= sh.W) return zero;
+    return a + (((long)n * sh.H + h) * sh.W + w) * sh.C + c;
+  } else {
+    // m -> (n,h,w); kd = (r*S + s)*K + k; valid iff stride divides
+    const int w = (int)(m % sh.W);
+    long t = m / sh.W;
+    const int h = (int)(t % sh.H);
+    const int n = (int)(t / sh.H);
+    const int k = kd % sh.K;
+    const int rs = kd / sh.K;
+    const int s = rs % sh.S;
+    const int r = rs / sh.S;
+    const int hp = h + sh.pad - r;
+    const int wp = w + sh.pad - s;
+    if (hp < 0 || wp < 0) return zero;
+    if (hp % sh.stride || wp % sh.stride) return zero;
+    const int p = hp / sh.stride, q = wp / sh.stride;
+    if (p >= sh.P || q >= sh.Q) return zero;
+    return a + (((long)n * sh.P + p) * sh.Q + q) * sh.K + k;
+  }
+}
+
+// B matrix chunk: row j (output col), kd..kd+7 contiguous. B storage is
+// (Nout, KD) row-major: W (K,RSC) for fwd, Wt (C,RSK) for bwd-data.
+AL_DEV const bf16* b_chunk_ptr(const bf16* __restrict__ b, const bf16* zero,
+                               const ConvShape& sh, int j, int kd) {
+  if (j >= sh.Nout) return zero;
+  return b + (long)j * sh.KD + kd;
+}
+
+// ---------------------------------------------------------------------------
+// the tiled kernel
+// ---------------------------------------------------------------------------
+
+template <int MODE>
+__launch_bounds__(256)
+__global__ void igemm_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
+                             bf16* __restrict__ out, const bf16* __restrict__ zero,
+                             ConvShape sh, int grid_m) {
+  constexpr int BM = 128, BN = 128, BK = 64;
+  // XCD-aware block remap (T1): contiguous output tiles on one XCD share B
+  // panels in its L2. Bijective variant for any grid size.
+  const int nwg = gridDim.x;
+  int bid = blockIdx.x;
+  {
+    const int nx = 8;
+    const int q = nwg / nx, r = nwg % nx;
+    const int xcd = bid % nx, pos = bid / nx;
+    if (pos < (xcd < r ? q + 1 : q))
+      bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + pos;
+  }
+  const int bm = bid % grid_m, bn = bid / grid_m;
+  const long m0 = (long)bm * BM;
+  const int n0 = bn * BN;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  bf16* As = (bf16*)smem;                    // [2][BM][BK]
+  bf16* Bs = As + 2 * BM * BK;               // [2][BN][BK]
+
+  const int tid = threadIdx.x;
+  const int wid = tid >> 6, lane = tid & 63;
+  const int wr = wid >> 1, wc = wid & 1;     // 2x2 wave grid, 64x64 each
+
+  const int KT = sh.KD / BK;
+
+  // stage one K-tile (both A and B) into buffer `buf` via glds.
+  // chunk t in [0,1024): row = t>>3, slot u = t&7 (16B each); the data for
+  // swizzled slot u comes from source slot u ^ (row&7).
+  auto stage = [&](int buf, int kt) {
+    const int k0 = kt * BK;
+    bf16* abase = As + buf * BM * BK;
+    bf16* bbase = Bs + buf * BN * BK;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const int t = (wid * 4 + i) * 64 + lane;
+      const int row = t >> 3, u = t & 7;
+      const int usw = u ^ (row & 7);
+      const bf16* src = a_chunk_ptr<MODE>(A, zero, sh, m0 + row, k0 + usw * 8);
+      __builtin_amdgcn_global_load_lds((const __attribute__((address_space(1))) void*)src,
+                                       (__attribute__((address_space(3))) void*)(abase + (wid * 4 + i) * 512),
+                                       16, 0, 0);
+    }
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const int t = (wid * 4 + i) * 64 + lane;
+      const int row = t >> 3, u = t & 7;
+      const int usw = u ^ (row & 7);
+      const bf16* src = b_chunk_ptr(B, zero, sh, n0 + row, k0 + usw * 8);
+      __builtin_amdgcn_global_load_lds((const __attribute__((address_space(1))) void*)src,
+                                       (__attribute__((address_space(3))) void*)(bbase + (wid * 4 + i) * 512),
+                                       16, 0, 0);
+    }
+  };
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int l15 = lane & 15, l4 = lane >> 4;
+
+  stage(0, 0);
+  int buf = 0;
+  for (int kt = 0; kt < KT; ++kt) {
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+    if (kt + 1 < KT) stage(buf ^ 1, kt + 1);
+    const bf16* abase = As + buf * BM * BK;
+    const bf16* bbase = Bs + buf * BN * BK;
+#pragma unroll
+    for (int kc = 0; kc < 2; ++kc) {  // two 32-deep chunks per K-tile
+      bf16x8 afrag[4], bfrag[4];
+#pragma unroll
+      for (int fi = 0; fi < 4; ++fi) {
+        const int arow = wr * 64 + fi * 16 + l15;
+        const int aslot = (kc * 4 + l4) ^ (arow & 7);
+        afrag[fi] = *(const bf16x8*)(abase + arow * BK + aslot * 8);
+        const int brow = wc * 64 + fi * 16 + l15;
+        const int bslot = (kc * 4 + l4) ^ (brow & 7);
+        bfrag[fi] = *(const bf16x8*)(bbase + brow * BK + bslot * 8);
+      }
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag[mi], bfrag[ni], acc[mi][ni], 0, 0, 0);
+    }
+    __syncthreads();
+    buf ^= 1;
+  }
+
+  // epilogue: C/D layout row = (l>>4)*4 + r, col = l&15
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi) {
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni) {
+      const int col = n0 + wc * 64 + ni * 16 + l15;
+      if (col >= sh.Nout) continue;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const long row = m0 + wr * 64 + mi * 16 + l4 * 4 + r;
+        if (row < sh.M)
+          out[row * sh.Nout + col] = f2bf(acc[mi][ni][r]);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// direct fallback (any shape; used for the C=3 stems): one output element per
+// thread, fp32 accumulate. Slow path by design — stems are ~3% of FLOPs.
+// ---------------------------------------------------------------------------
+
+template <int MODE>
+__global__ void conv_direct_kernel(const bf16* __restrict__ A,
+                                   const bf16* __restrict__ B,
+                                   bf16* __restrict__ out, ConvShape sh) {
+  const long total = sh.M * sh.Nout;
+  for (long i = grid_stride_begin(); i < total; i += grid_stride_step()) {
+    const long m = i / sh.Nout;
+    const int j = (int)(i % sh.Nout);
+    float acc = 0.f;
+    for (int kd = 0; kd < sh.KD; ++kd) {
+      const bf16* pa = a_chunk_ptr<MODE>(A, nullptr, sh, m, kd);
+      if (pa == nullptr) continue;
+      acc += bf2f(*pa) * bf2f(B[(long)j * sh.KD + kd]);
+    }
+    out[i] = f2bf(acc);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// launchers
+// ---------------------------------------------------------------------------
+
+static inline bool igemm_ok(int mode, const ConvShape& sh) {
+  // 16B A-chunks must not straddle an (r,s) boundary: the fast gather dim
+  // (C for fwd, K for bwd) must be a multiple of 8; contraction tiles exact.
+  const int fast = (mode == MODE_FWD) ? sh.C : sh.K;
+  return (sh.KD % 64 == 0) && (fast % 8 == 0);
+}
+
+extern "C" void al_conv2d_mm(int mode, const void* A, const void* B, void* out,
+                             const void* zero_page, int N, int H, int W, int C,
+                             int K, int R, int S, int P, int Q, int stride, int pad,
+                             hipStream_t stream) {
+  ConvShape sh;
+  sh.N = N; sh.H = H; sh.W = W; sh.C = C; sh.K = K; sh.R = R; sh.S = S;
+  sh.P = P; sh.Q = Q; sh.stride = stride; sh.pad = pad;
+  if (mode == MODE_FWD) {
+    sh.M = (long)N * P * Q;
+    sh.Nout = K;
+    sh.KD = R * S * C;
+  } else {
+    sh.M = (long)N * H * W;
+    sh.Nout = C;
+    sh.KD = R * S * K;
+  }
+  if (igemm_ok(mode, sh)) {
+    const int grid_m = (int)((sh.M + 127) / 128);
+    const int grid_n = (sh.Nout + 127) / 128;
+    const size_t lds = 2 * (128 * 64 + 128 * 64) * sizeof(bf16);  // 64 KiB
+    dim3 grid(grid_m * grid_n), block(256);
+    if (mode == MODE_FWD)
+      hipLaunchKernelGGL((igemm_kernel<MODE_FWD>), grid, block, lds, stream,
+                         (const bf16*)A, (const bf16*)B, (bf16*)out,
+                         (const bf16*)zero_page, sh, grid_m);
+    else
+      hipLaunchKernelGGL((igemm_kernel<MODE_BWD_DATA>), grid, block, lds, stream,
+                         (const bf16*)A, (const bf16*)B, (bf16*)out,
+                         (const bf16*)zero_page, sh, grid_m);
+  } else {
+    long total = sh.M * sh.Nout;
+    int blocks = (int)min((total + 255) / 256, (long)8192);
+    if (mode == MODE_FWD)
+      hipLaunchKernelGGL((conv_direct_kernel<MODE_FWD>), dim3(blocks), dim3(256), 0,
+                         stream, (const bf16*)A, (const bf16*)B, (bf16*)out, sh);
+    else
+      hipLaunchKernelGGL((conv_direct_kernel<MODE_BWD_DATA>), dim3(blocks), dim3(256),
+                         0, stream, (const bf16*)A, (const bf16*)B, (bf16*)out, sh);
+  }
+}

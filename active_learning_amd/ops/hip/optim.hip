@@ -1,0 +1,49 @@
+// Fused optimizer updates (fp32 master params).
+// SGD: p -= lr * (momentum-buffered (g + wd*p)); one read-modify-write pass
+// (reference reaches torch.optim.SGD, arg_pools/default.py:39-40).
+// Adam for VAAL's VAE/discriminator (vaal_sampler.py:139-140).
+
+#include "al_common.h"
+
+__global__ void sgd_kernel(float* __restrict__ p, const float* __restrict__ g,
+                           float* __restrict__ buf, float lr, float momentum,
+                           float wd, long n) {
+  for (long i = grid_stride_begin(); i < n; i += grid_stride_step()) {
+    float grad = g[i] + wd * p[i];
+    if (momentum != 0.f) {
+      const float b = buf[i] * momentum + grad;
+      buf[i] = b;
+      grad = b;
+    }
+    p[i] -= lr * grad;
+  }
+}
+
+extern "C" void al_sgd_step(float* p, const float* g, float* buf, float lr,
+                            float momentum, float wd, long n, hipStream_t stream) {
+  int blocks = (int)min((n + 255) / 256, (long)2048);
+  hipLaunchKernelGGL(sgd_kernel, dim3(blocks), dim3(256), 0, stream, p, g, buf, lr,
+                     momentum, wd, n);
+}
+
+__global__ void adam_kernel(float* __restrict__ p, const float* __restrict__ g,
+                            float* __restrict__ m, float* __restrict__ v, float lr,
+                            float b1, float b2, float eps, float wd, float bc1,
+                            float bc2, long n) {
+  for (long i = grid_stride_begin(); i < n; i += grid_stride_step()) {
+    const float grad = g[i] + wd * p[i];
+    const float mi = m[i] * b1 + (1.f - b1) * grad;
+    const float vi = v[i] * b2 + (1.f - b2) * grad * grad;
+    m[i] = mi;
+    v[i] = vi;
+    p[i] -= lr * (mi / bc1) / (sqrtf(vi / bc2) + eps);
+  }
+}
+
+extern "C" void al_adam_step(float* p, const float* g, float* m, float* v, float lr,
+                             float b1, float b2, float eps, float wd, float bc1,
+                             float bc2, long n, hipStream_t stream) {
+  int blocks = (int)min((n + 255) / 256, (long)2048);
+  hipLaunchKernelGGL(adam_kernel, dim3(blocks), dim3(256), 0, stream, p, g, m, v, lr,
+                     b1, b2, eps, wd, bc1, bc2, n);
+}

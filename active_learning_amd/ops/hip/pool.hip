@@ -1,0 +1,110 @@
+// NHWC pooling kernels (ResNet stem maxpool 3x3 s2 + global average pool).
+// Reference counterpart: torchvision resnet maxpool/avgpool.
+
+#include "al_common.h"
+
+// ---------------------------------------------------------------------------
+// maxpool fwd: y[n,p,q,c] = max window; idx = flat h*w argmax (torch format,
+// so the CPU fallback and the GPU kernel share the backward contract)
+// ---------------------------------------------------------------------------
+
+__global__ void maxpool_fwd_kernel(const bf16* __restrict__ x, bf16* __restrict__ y,
+                                   long* __restrict__ idx, int N, int H, int W, int C,
+                                   int P, int Q, int kernel, int stride, int pad) {
+  const long total = (long)N * P * Q * C;
+  for (long i = grid_stride_begin(); i < total; i += grid_stride_step()) {
+    const int c = (int)(i % C);
+    long t = i / C;
+    const int q = (int)(t % Q);
+    t /= Q;
+    const int p = (int)(t % P);
+    const int n = (int)(t / P);
+    float best = -3.0e38f;
+    int best_hw = 0;
+    const int h0 = p * stride - pad, w0 = q * stride - pad;
+    for (int r = 0; r < kernel; ++r) {
+      const int h = h0 + r;
+      if (h < 0 || h >= H) continue;
+      for (int s = 0; s < kernel; ++s) {
+        const int w = w0 + s;
+        if (w < 0 || w >= W) continue;
+        const float v = bf2f(x[(((long)n * H + h) * W + w) * C + c]);
+        if (v > best) { best = v; best_hw = h * W + w; }
+      }
+    }
+    y[i] = f2bf(best);
+    idx[i] = best_hw;
+  }
+}
+
+extern "C" void al_maxpool_fwd(const void* x, void* y, long* idx, int N, int H, int W,
+                               int C, int P, int Q, int kernel, int stride, int pad,
+                               hipStream_t stream) {
+  long total = (long)N * P * Q * C;
+  int blocks = (int)min((total + 255) / 256, (long)4096);
+  hipLaunchKernelGGL(maxpool_fwd_kernel, dim3(blocks), dim3(256), 0, stream,
+                     (const bf16*)x, (bf16*)y, idx, N, H, W, C, P, Q, kernel, stride,
+                     pad);
+}
+
+// backward: per INPUT pixel, sum dy over covering windows whose argmax == me
+// (deterministic, no atomics; each input is covered by <= ceil(k/s)^2 windows)
+__global__ void maxpool_bwd_kernel(const bf16* __restrict__ dy,
+                                   const long* __restrict__ idx,
+                                   bf16* __restrict__ dx, int N, int H, int W, int C,
+                                   int P, int Q, int kernel, int stride, int pad) {
+  const long total = (long)N * H * W * C;
+  for (long i = grid_stride_begin(); i < total; i += grid_stride_step()) {
+    const int c = (int)(i % C);
+    long t = i / C;
+    const int w = (int)(t % W);
+    t /= W;
+    const int h = (int)(t % H);
+    const int n = (int)(t / H);
+    const int my_hw = h * W + w;
+    float acc = 0.f;
+    const int pmin = max(0, (h + pad - kernel + stride) / stride);
+    const int pmax = min(P - 1, (h + pad) / stride);
+    const int qmin = max(0, (w + pad - kernel + stride) / stride);
+    const int qmax = min(Q - 1, (w + pad) / stride);
+    for (int p = pmin; p <= pmax; ++p)
+      for (int q = qmin; q <= qmax; ++q) {
+        const long o = (((long)n * P + p) * Q + q) * C + c;
+        if (idx[o] == my_hw) acc += bf2f(dy[o]);
+      }
+    dx[i] = f2bf(acc);
+  }
+}
+
+extern "C" void al_maxpool_bwd(const void* dy, const long* idx, void* dx, int N, int H,
+                               int W, int C, int P, int Q, int kernel, int stride,
+                               int pad, hipStream_t stream) {
+  long total = (long)N * H * W * C;
+  int blocks = (int)min((total + 255) / 256, (long)4096);
+  hipLaunchKernelGGL(maxpool_bwd_kernel, dim3(blocks), dim3(256), 0, stream,
+                     (const bf16*)dy, idx, (bf16*)dx, N, H, W, C, P, Q, kernel, stride,
+                     pad);
+}
+
+// ---------------------------------------------------------------------------
+// global average pool: (N,H,W,C) -> (N,C) fp32-accumulated, bf16-out
+// block per image; lanes stride channels; rows streamed
+// ---------------------------------------------------------------------------
+
+__global__ void gap_kernel(const bf16* __restrict__ x, bf16* __restrict__ y, int HW,
+                           int C) {
+  const int n = blockIdx.x;
+  const float inv = 1.0f / HW;
+  for (int c = threadIdx.x; c < C; c += blockDim.x) {
+    float acc = 0.f;
+    const bf16* base = x + (long)n * HW * C + c;
+    for (int r = 0; r < HW; ++r) acc += bf2f(base[(long)r * C]);
+    y[(long)n * C + c] = f2bf(acc * inv);
+  }
+}
+
+extern "C" void al_global_avg_pool(const void* x, void* y, int N, int HW, int C,
+                                   hipStream_t stream) {
+  hipLaunchKernelGGL(gap_kernel, dim3(N), dim3(256), 0, stream, (const bf16*)x,
+                     (bf16*)y, HW, C);
+}

@@ -1,0 +1,133 @@
+#!/usr/bin/env python3
+"""Flagship training-step benchmark: ResNet-50 / synthetic ImageNet, bf16,
+NHWC, hand-written CDNA4 kernels, RCCL DDP (BASELINE.json metric:
+images/sec (train), weak scaling over 1/2/4/8 MI355X GPUs).
+
+Single GPU:      python bench.py --gpus 1 --steps 30 --warmup 10
+Multi GPU (driver): python -m torch.distributed.run --nnodes=1
+                 --nproc-per-node N --master-addr 127.0.0.1 bench.py --gpus N ...
+
+Each timed step = forward + CE loss + backward (+ bucketed gradient
+all-reduce overlapped with backward when N > 1) + fused SGD update. Synthetic
+ImageNet-shaped data (3x224x224, random labels), random-init weights.
+"""
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+
+def get_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=30)
+    p.add_argument("--warmup", type=int, default=10)
+    p.add_argument("--batch", type=int, default=128, help="per-GPU batch size")
+    p.add_argument("--model", default="resnet50", choices=["resnet50", "resnet18"])
+    p.add_argument("--img", type=int, default=224)
+    p.add_argument("--classes", type=int, default=1000)
+    p.add_argument("--bucket-mb", type=float, default=16.0)
+    return p.parse_args()
+
+
+def main():
+    args = get_args()
+    assert torch.cuda.is_available(), "bench.py requires a GPU"
+
+    rank = int(os.environ.get("RANK", 0))
+    world = int(os.environ.get("WORLD_SIZE", 1))
+    local_rank = int(os.environ.get("LOCAL_RANK", 0))
+    torch.cuda.set_device(local_rank)
+    device = torch.device("cuda", local_rank)
+
+    dist = None
+    if world > 1:
+        import torch.distributed as dist_mod
+        dist = dist_mod
+        dist.init_process_group("nccl")  # RCCL over xGMI
+
+    from active_learning_amd.models.ssl_resnet import ResNetSimCLR
+    from active_learning_amd.ops.extension import require_extension
+    from active_learning_amd.ops.loss import cross_entropy
+    from active_learning_amd.ops.optim import FusedSGD
+    require_extension()
+
+    torch.manual_seed(1234 + rank)
+    base = "resnet50" if args.model == "resnet50" else "resnet18"
+    net = ResNetSimCLR(base, num_classes=args.classes).to(device)
+    if world > 1:
+        from active_learning_amd.parallel import BucketedDDP, convert_sync_batchnorm
+        net = BucketedDDP(net, bucket_cap_mb=args.bucket_mb)
+    opt = FusedSGD(net.parameters(), lr=0.1, momentum=0.9, weight_decay=1e-4)
+
+    # synthetic device-resident batches (data=synthetic per BASELINE.json)
+    n_buf = 4
+    xs = [torch.randn(args.batch, 3, args.img, args.img, device=device)
+          for _ in range(n_buf)]
+    ys = [torch.randint(0, args.classes, (args.batch,), device=device)
+          for _ in range(n_buf)]
+
+    def step(i):
+        x, y = xs[i % n_buf], ys[i % n_buf]
+        opt.zero_grad(set_to_none=True)
+        out = net(x)
+        loss = cross_entropy(out, y)
+        loss.backward()
+        if world > 1:
+            net.finalize_grads()
+        opt.step()
+        return loss
+
+    net.train()
+    for i in range(args.warmup):
+        step(i)
+    torch.cuda.synchronize()
+    if dist:
+        dist.barrier()
+        torch.cuda.synchronize()
+
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        step(i)
+    torch.cuda.synchronize()
+    if dist:
+        dist.barrier()
+        torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+
+    if dist:
+        t = torch.tensor([elapsed], device=device)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = t.item()
+
+    n_gpus = world if world > 1 else 1
+    ms_per_step = elapsed / args.steps * 1000.0
+    images_per_sec = args.batch * n_gpus * args.steps / elapsed
+
+    if rank == 0:
+        print(json.dumps({
+            "metric": "images/sec (train)",
+            "value": round(images_per_sec, 2),
+            "unit": "images/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {"model": args.model, "global_batch": args.batch * n_gpus,
+                       "seq_len": None, "img": args.img,
+                       "parallelism": f"dp{n_gpus}"},
+        }))
+    if dist:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()
