@@ -29,6 +29,33 @@ def _cpu_conv_args(x_nhwc, w_krsc):
     return x_nhwc.permute(0, 3, 1, 2), w_krsc.permute(0, 3, 1, 2)
 
 
+def _igemm_eligible(C, KD):
+    """Shapes the MFMA igemm gathers directly (C%8 for 16B chunks, KD%64 for
+    exact contraction tiles). Others — the C=3 stems — go through the packed
+    im2col path so they still run on MFMA instead of the slow direct kernel."""
+    return C % 8 == 0 and KD % 64 == 0
+
+
+def _gpu_conv_fwd_packed(ext, x, w_c, stride, padding):
+    """Stem path: A = im2col(x) zero-padded to kdpad cols, conv as 1x1 igemm."""
+    K, R, S, C = w_c.shape
+    KD = R * S * C
+    kdpad = ((KD + 63) // 64) * 64
+    apack = ext.im2col_pack(x, R, S, stride, padding, kdpad)
+    wpad = torch.zeros(K, 1, 1, kdpad, dtype=w_c.dtype, device=w_c.device)
+    wpad.view(K, kdpad)[:, :KD] = w_c.reshape(K, KD)
+    return ext.conv2d_fwd(apack, wpad, 1, 0)
+
+
+def _gpu_conv_wgrad_packed(ext, dy, x, w_shape, stride, padding):
+    K, R, S, C = w_shape
+    KD = R * S * C
+    kdpad = ((KD + 63) // 64) * 64
+    apack = ext.im2col_pack(x, R, S, stride, padding, kdpad)
+    dwpad = ext.conv2d_bwd_weight(dy, apack, 1, 1, 1, 0)  # (K,1,1,kdpad) fp32
+    return dwpad.view(K, kdpad)[:, :KD].reshape(K, R, S, C).contiguous()
+
+
 class Conv2dNHWC(Function):
     """y[N,P,Q,K] = conv(x[N,H,W,C], w[K,R,S,C]; stride, pad), no bias
     (ResNet convs carry no bias; BN follows)."""
@@ -41,7 +68,10 @@ class Conv2dNHWC(Function):
         ctx.weight_dtype = weight.dtype
         if x.is_cuda:
             ext = require_extension()
-            return ext.conv2d_fwd(x, w_c, stride, padding)
+            K, R, S, C = w_c.shape
+            if _igemm_eligible(C, R * S * C):
+                return ext.conv2d_fwd(x, w_c, stride, padding)
+            return _gpu_conv_fwd_packed(ext, x, w_c, stride, padding)
         xc, wc = _cpu_conv_args(x, w_c)
         y = F.conv2d(xc.float(), wc.float(), stride=stride, padding=padding)
         return y.to(x.dtype).permute(0, 2, 3, 1).contiguous()
@@ -53,12 +83,16 @@ class Conv2dNHWC(Function):
         dx = dw = None
         if x.is_cuda:
             ext = require_extension()
+            K, R, S, C = w_c.shape
             if ctx.needs_input_grad[0]:
                 dx = ext.conv2d_bwd_data(dy, w_c, ctx.stride, ctx.padding,
                                          x.shape[1], x.shape[2])
             if ctx.needs_input_grad[1]:
-                dw = ext.conv2d_bwd_weight(dy, x, w_c.shape[1], w_c.shape[2],
-                                           ctx.stride, ctx.padding)
+                if _igemm_eligible(C, R * S * C):
+                    dw = ext.conv2d_bwd_weight(dy, x, R, S, ctx.stride, ctx.padding)
+                else:
+                    dw = _gpu_conv_wgrad_packed(ext, dy, x, w_c.shape, ctx.stride,
+                                                ctx.padding)
         else:
             xc, wc = _cpu_conv_args(x, w_c)
             dyc = dy.permute(0, 3, 1, 2).float()

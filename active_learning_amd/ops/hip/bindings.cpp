@@ -41,6 +41,9 @@ void al_ce_fwd(const float* logits, const long* targets, float* losses, float* p
                int B, int C, void* stream);
 void al_ce_bwd(const float* probs, const long* targets, const float* scale,
                float* dlogits, int B, int C, void* stream);
+void al_im2col_pack(const void* x, void* out, int N, int H, int W, int C, int R,
+                    int S, int P, int Q, int stride, int pad, int kdpad,
+                    void* stream);
 void al_sgd_step(float* p, const float* g, float* buf, float lr, float momentum,
                  float wd, long n, void* stream);
 void al_adam_step(float* p, const float* g, float* m, float* v, float lr, float b1,
@@ -111,6 +114,17 @@ Tensor conv2d_bwd_weight(const Tensor& dy, const Tensor& x, long R, long S,
   al_conv2d_wgrad(dy.data_ptr(), x.data_ptr(), dw.data_ptr<float>(), N, H, W, C, K,
                   (int)R, (int)S, P, Q, (int)stride, (int)pad, cur_stream());
   return dw;
+}
+
+Tensor im2col_pack(const Tensor& x, long R, long S, long stride, long pad,
+                   long kdpad) {
+  check_bf16_contig(x, "x");
+  const int N = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
+  const int P = out_dim(H, R, stride, pad), Q = out_dim(W, S, stride, pad);
+  auto out = torch::empty({N, P, Q, kdpad}, x.options());
+  al_im2col_pack(x.data_ptr(), out.data_ptr(), N, H, W, C, (int)R, (int)S, P, Q,
+                 (int)stride, (int)pad, (int)kdpad, cur_stream());
+  return out;
 }
 
 std::vector<Tensor> bn_stats(const Tensor& x) {
@@ -274,6 +288,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_norm_fwd", &bn_norm_fwd);
   m.def("bn_bwd_reduce", &bn_bwd_reduce);
   m.def("bn_bwd", &bn_bwd);
+  m.def("im2col_pack", &im2col_pack);
   m.def("maxpool2d_fwd", &maxpool2d_fwd);
   m.def("maxpool2d_bwd", &maxpool2d_bwd);
   m.def("global_avg_pool", &global_avg_pool);

@@ -19,44 +19,67 @@
 // ---------------------------------------------------------------------------
 
 __global__ void bn_stats_kernel(const bf16* __restrict__ x, float* __restrict__ sum,
-                                float* __restrict__ sumsq, long rows, int C) {
-  // blockIdx.y selects a 64-channel group; 256 threads = 64 channels x 4 row lanes
-  const int cg = blockIdx.y * 64;
-  const int c = cg + (threadIdx.x & 63);
-  if (c >= C) return;
-  const int row_lane = threadIdx.x >> 6;          // 0..3
-  const long row0 = (long)blockIdx.x * 4 + row_lane;
-  const long row_step = (long)gridDim.x * 4;
-  float s = 0.f, ss = 0.f;
-  for (long r = row0; r < rows; r += row_step) {
-    float v = bf2f(x[r * C + c]);
-    s += v;
-    ss += v * v;
-  }
-  // reduce across the 4 row lanes holding the same channel via LDS
-  __shared__ float red[2][256];
-  red[0][threadIdx.x] = s;
-  red[1][threadIdx.x] = ss;
-  __syncthreads();
-  if (threadIdx.x < 64) {
-    float ts = 0.f, tss = 0.f;
+                                float* __restrict__ sumsq, long rows, int C,
+                                int cg_per_block) {
+  // thread -> (c8 group, row lane); consecutive lanes read consecutive 16B
+  // chunks (s16x8), so a 32-lane group streams 512 contiguous bytes.
+  const int cg_local = threadIdx.x % cg_per_block;
+  const int row_lane = threadIdx.x / cg_per_block;
+  const int rows_per_block = 256 / cg_per_block;
+  const int c8 = blockIdx.y * cg_per_block + cg_local;
+  if (c8 * 8 >= C) return;
+  const int C8 = C / 8;
+  float s[8] = {0, 0, 0, 0, 0, 0, 0, 0}, ss[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  const long row0 = (long)blockIdx.x * rows_per_block + row_lane;
+  const long step = (long)gridDim.x * rows_per_block;
+  for (long r = row0; r < rows; r += step) {
+    s16x8 v = ((const s16x8*)x)[r * C8 + c8];
 #pragma unroll
-    for (int i = 0; i < 4; ++i) {
-      ts += red[0][threadIdx.x + 64 * i];
-      tss += red[1][threadIdx.x + 64 * i];
+    for (int j = 0; j < 8; ++j) {
+      const float f = bits2f(v[j]);
+      s[j] += f;
+      ss[j] += f * f;
     }
-    atomicAdd(&sum[c], ts);
-    atomicAdd(&sumsq[c], tss);
   }
+  // reduce across row lanes sharing a c8 group through LDS
+  __shared__ float red[2][256][8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    red[0][threadIdx.x][j] = s[j];
+    red[1][threadIdx.x][j] = ss[j];
+  }
+  __syncthreads();
+  if (row_lane == 0) {
+    for (int rl = 1; rl < rows_per_block; ++rl)
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        s[j] += red[0][rl * cg_per_block + cg_local][j];
+        ss[j] += red[1][rl * cg_per_block + cg_local][j];
+      }
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      atomicAdd(&sum[c8 * 8 + j], s[j]);
+      atomicAdd(&sumsq[c8 * 8 + j], ss[j]);
+    }
+  }
+}
+
+static inline int bn_cg_per_block(int C) {
+  int c8 = C / 8;
+  int cg = 1;
+  while (cg < 32 && cg * 2 <= c8 && (c8 % (cg * 2)) == 0) cg *= 2;
+  return cg;  // largest power-of-two divisor of C/8, capped at 32
 }
 
 extern "C" void al_bn_stats(const void* x, float* sum, float* sumsq, long rows, int C,
                             hipStream_t stream) {
   dim3 block(256);
-  int row_blocks = (int)min((rows + 3) / 4, (long)1024);
-  dim3 grid(row_blocks, (C + 63) / 64);
+  const int cg = bn_cg_per_block(C);
+  const int rpb = 256 / cg;
+  int row_blocks = (int)min((rows + rpb - 1) / rpb, (long)512);
+  dim3 grid(row_blocks, (C / 8 + cg - 1) / cg);
   hipLaunchKernelGGL(bn_stats_kernel, grid, block, 0, stream, (const bf16*)x, sum,
-                     sumsq, rows, C);
+                     sumsq, rows, C, cg);
 }
 
 // ---------------------------------------------------------------------------
@@ -118,35 +141,55 @@ __global__ void bn_bwd_reduce_kernel(const bf16* __restrict__ dy,
                                      const float* __restrict__ invstd,
                                      float* __restrict__ sum_dy,
                                      float* __restrict__ sum_dy_xhat, long rows,
-                                     int C) {
-  const int cg = blockIdx.y * 64;
-  const int c = cg + (threadIdx.x & 63);
-  if (c >= C) return;
-  const int row_lane = threadIdx.x >> 6;
-  const long row0 = (long)blockIdx.x * 4 + row_lane;
-  const long row_step = (long)gridDim.x * 4;
-  const float m = mean[c], is = invstd[c];
-  float s = 0.f, sx = 0.f;
-  for (long r = row0; r < rows; r += row_step) {
-    const long off = r * C + c;
-    float g = bf2f(dy[off]);
-    if (RELU) g = bf2f(y[off]) > 0.f ? g : 0.f;
-    s += g;
-    sx += g * (bf2f(x[off]) - m) * is;
-  }
-  __shared__ float red[2][256];
-  red[0][threadIdx.x] = s;
-  red[1][threadIdx.x] = sx;
-  __syncthreads();
-  if (threadIdx.x < 64) {
-    float ts = 0.f, tsx = 0.f;
+                                     int C, int cg_per_block) {
+  const int cg_local = threadIdx.x % cg_per_block;
+  const int row_lane = threadIdx.x / cg_per_block;
+  const int rows_per_block = 256 / cg_per_block;
+  const int c8 = blockIdx.y * cg_per_block + cg_local;
+  if (c8 * 8 >= C) return;
+  const int C8 = C / 8;
+  float m[8], is[8];
 #pragma unroll
-    for (int i = 0; i < 4; ++i) {
-      ts += red[0][threadIdx.x + 64 * i];
-      tsx += red[1][threadIdx.x + 64 * i];
+  for (int j = 0; j < 8; ++j) {
+    m[j] = mean[c8 * 8 + j];
+    is[j] = invstd[c8 * 8 + j];
+  }
+  float s[8] = {0, 0, 0, 0, 0, 0, 0, 0}, sx[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  const long row0 = (long)blockIdx.x * rows_per_block + row_lane;
+  const long step = (long)gridDim.x * rows_per_block;
+  for (long r = row0; r < rows; r += step) {
+    const long off = r * C8 + c8;
+    s16x8 gv = ((const s16x8*)dy)[off];
+    s16x8 xv = ((const s16x8*)x)[off];
+    s16x8 yv;
+    if (RELU) yv = ((const s16x8*)y)[off];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float g = bits2f(gv[j]);
+      if (RELU) g = bits2f(yv[j]) > 0.f ? g : 0.f;
+      s[j] += g;
+      sx[j] += g * (bits2f(xv[j]) - m[j]) * is[j];
     }
-    atomicAdd(&sum_dy[c], ts);
-    atomicAdd(&sum_dy_xhat[c], tsx);
+  }
+  __shared__ float red[2][256][8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    red[0][threadIdx.x][j] = s[j];
+    red[1][threadIdx.x][j] = sx[j];
+  }
+  __syncthreads();
+  if (row_lane == 0) {
+    for (int rl = 1; rl < rows_per_block; ++rl)
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        s[j] += red[0][rl * cg_per_block + cg_local][j];
+        sx[j] += red[1][rl * cg_per_block + cg_local][j];
+      }
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      atomicAdd(&sum_dy[c8 * 8 + j], s[j]);
+      atomicAdd(&sum_dy_xhat[c8 * 8 + j], sx[j]);
+    }
   }
 }
 
@@ -155,16 +198,18 @@ extern "C" void al_bn_bwd_reduce(const void* dy, const void* x, const void* y,
                                  float* sum_dy, float* sum_dy_xhat, int relu,
                                  long rows, int C, hipStream_t stream) {
   dim3 block(256);
-  int row_blocks = (int)min((rows + 3) / 4, (long)1024);
-  dim3 grid(row_blocks, (C + 63) / 64);
+  const int cg = bn_cg_per_block(C);
+  const int rpb = 256 / cg;
+  int row_blocks = (int)min((rows + rpb - 1) / rpb, (long)512);
+  dim3 grid(row_blocks, (C / 8 + cg - 1) / cg);
   if (relu)
     hipLaunchKernelGGL((bn_bwd_reduce_kernel<true>), grid, block, 0, stream,
                        (const bf16*)dy, (const bf16*)x, (const bf16*)y, mean, invstd,
-                       sum_dy, sum_dy_xhat, rows, C);
+                       sum_dy, sum_dy_xhat, rows, C, cg);
   else
     hipLaunchKernelGGL((bn_bwd_reduce_kernel<false>), grid, block, 0, stream,
                        (const bf16*)dy, (const bf16*)x, (const bf16*)y, mean, invstd,
-                       sum_dy, sum_dy_xhat, rows, C);
+                       sum_dy, sum_dy_xhat, rows, C, cg);
 }
 
 // ---------------------------------------------------------------------------

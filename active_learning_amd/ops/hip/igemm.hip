@@ -291,3 +291,41 @@ extern "C" void al_conv2d_mm(int mode, const void* A, const void* B, void* out,
                          0, stream, (const bf16*)A, (const bf16*)B, (bf16*)out, sh);
   }
 }
+
+// ---------------------------------------------------------------------------
+// im2col packing for shapes the MFMA kernel cannot gather directly (the C=3
+// stems): materialize A = im2col(x) zero-padded to KDpad columns (KDpad % 64
+// == 0); the conv then runs as a 1x1 MFMA igemm over the packed buffer.
+// Memory-bound; output writes are fully coalesced (consecutive threads ->
+// consecutive kd).
+// ---------------------------------------------------------------------------
+
+__global__ void im2col_pack_kernel(const bf16* __restrict__ x, bf16* __restrict__ out,
+                                   ConvShape sh, int kdpad) {
+  const long total = sh.M * kdpad;
+  for (long i = grid_stride_begin(); i < total; i += grid_stride_step()) {
+    const long m = i / kdpad;
+    const int kd = (int)(i % kdpad);
+    float v = 0.f;
+    if (kd < sh.KD) {
+      const bf16* p = a_chunk_ptr<MODE_FWD>(x, nullptr, sh, m, kd);
+      if (p) v = bf2f(*p);
+    }
+    out[i] = f2bf(v);
+  }
+}
+
+extern "C" void al_im2col_pack(const void* x, void* out, int N, int H, int W, int C,
+                               int R, int S, int P, int Q, int stride, int pad,
+                               int kdpad, hipStream_t stream) {
+  ConvShape sh;
+  sh.N = N; sh.H = H; sh.W = W; sh.C = C; sh.K = 0; sh.R = R; sh.S = S;
+  sh.P = P; sh.Q = Q; sh.stride = stride; sh.pad = pad;
+  sh.M = (long)N * P * Q;
+  sh.Nout = 0;
+  sh.KD = R * S * C;
+  long total = sh.M * kdpad;
+  int blocks = (int)min((total + 255) / 256, (long)8192);
+  hipLaunchKernelGGL(im2col_pack_kernel, dim3(blocks), dim3(256), 0, stream,
+                     (const bf16*)x, (bf16*)out, sh, kdpad);
+}
