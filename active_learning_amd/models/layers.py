@@ -110,8 +110,10 @@ class ConvTranspose2dNHWC(nn.Module):
 class _TransposedConv2d(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, stride, padding):
-        w_c = weight.to(x.dtype)
-        ctx.save_for_backward(x, w_c)
+        from ..ops.functional import cast_cached, wt_cached
+        w_c = cast_cached(weight, x.dtype)
+        ctx.save_for_backward(x)
+        ctx.w_c = w_c
         ctx.stride, ctx.padding = stride, padding
         ctx.weight_dtype = weight.dtype
         k = w_c.shape[1]
@@ -119,7 +121,9 @@ class _TransposedConv2d(torch.autograd.Function):
         w_out = (x.shape[2] - 1) * stride - 2 * padding + k
         if x.is_cuda:
             from ..ops.extension import require_extension
-            return require_extension().conv2d_bwd_data(x, w_c, stride, padding, h_out, w_out)
+            # convT fwd == conv bwd-data; its "wt" is (Cout,R,S,Cin)
+            return require_extension().conv2d_bwd_data(x, wt_cached(w_c), stride,
+                                                       padding, h_out, w_out)
         xc = x.permute(0, 3, 1, 2).float()
         wc = w_c.permute(0, 3, 1, 2).float()  # (C_in, C_out, R, S)
         y = torch.nn.functional.conv_transpose2d(xc, wc, stride=stride, padding=padding)
@@ -127,7 +131,8 @@ class _TransposedConv2d(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, dy):
-        x, w_c = ctx.saved_tensors
+        (x,) = ctx.saved_tensors
+        w_c = ctx.w_c
         dy = dy.contiguous()
         dx = dw = None
         if x.is_cuda:

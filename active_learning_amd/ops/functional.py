@@ -29,6 +29,42 @@ def _cpu_conv_args(x_nhwc, w_krsc):
     return x_nhwc.permute(0, 3, 1, 2), w_krsc.permute(0, 3, 1, 2)
 
 
+def cast_cached(weight, dtype):
+    """Per-version cached bf16 copy of an fp32 master weight (saves a cast +
+    copy per layer per step; invalidated by optimizers via
+    invalidate_param_cache)."""
+    if weight.dtype == dtype:
+        return weight
+    cache = getattr(weight, "_al_cast", None)
+    ver = weight._version
+    if cache is not None and cache[0] == ver and cache[1].dtype == dtype:
+        return cache[1]
+    w = weight.detach().to(dtype)
+    try:
+        weight._al_cast = (ver, w)
+    except Exception:
+        pass
+    return w
+
+
+def invalidate_param_cache(p):
+    if getattr(p, "_al_cast", None) is not None:
+        p._al_cast = None
+
+
+def wt_cached(w_c):
+    """(K,R,S,C) -> (C,R,S,K) permutation for the bwd-data contraction,
+    cached on the bf16 weight copy (which is itself version-cached)."""
+    wt = getattr(w_c, "_al_wt", None)
+    if wt is None:
+        wt = w_c.permute(3, 1, 2, 0).contiguous()
+        try:
+            w_c._al_wt = wt
+        except Exception:
+            pass
+    return wt
+
+
 def _igemm_eligible(C, KD):
     """Shapes the MFMA igemm gathers directly (C%8 for 16B chunks, KD%64 for
     exact contraction tiles). Others — the C=3 stems — go through the packed
@@ -62,8 +98,9 @@ class Conv2dNHWC(Function):
 
     @staticmethod
     def forward(ctx, x, weight, stride, padding):
-        w_c = weight.to(x.dtype)  # fp32 master -> compute dtype
-        ctx.save_for_backward(x, w_c)
+        w_c = cast_cached(weight, x.dtype)  # fp32 master -> compute dtype
+        ctx.save_for_backward(x)
+        ctx.w_c = w_c
         ctx.stride, ctx.padding = stride, padding
         ctx.weight_dtype = weight.dtype
         if x.is_cuda:
@@ -78,14 +115,15 @@ class Conv2dNHWC(Function):
 
     @staticmethod
     def backward(ctx, dy):
-        x, w_c = ctx.saved_tensors
+        (x,) = ctx.saved_tensors
+        w_c = ctx.w_c
         dy = dy.contiguous()
         dx = dw = None
         if x.is_cuda:
             ext = require_extension()
             K, R, S, C = w_c.shape
             if ctx.needs_input_grad[0]:
-                dx = ext.conv2d_bwd_data(dy, w_c, ctx.stride, ctx.padding,
+                dx = ext.conv2d_bwd_data(dy, wt_cached(w_c), ctx.stride, ctx.padding,
                                          x.shape[1], x.shape[2])
             if ctx.needs_input_grad[1]:
                 if _igemm_eligible(C, R * S * C):

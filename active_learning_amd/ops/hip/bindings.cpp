@@ -87,15 +87,15 @@ Tensor conv2d_fwd(const Tensor& x, const Tensor& w, long stride, long pad) {
   return y;
 }
 
-Tensor conv2d_bwd_data(const Tensor& dy, const Tensor& w, long stride, long pad,
+Tensor conv2d_bwd_data(const Tensor& dy, const Tensor& wt, long stride, long pad,
                        long H, long W) {
+  // wt is the weight PRE-PERMUTED to (C,R,S,K) — k fastest in the contraction;
+  // the Python layer caches this permutation per weight version.
   check_bf16_contig(dy, "dy");
-  check_bf16_contig(w, "w");
+  check_bf16_contig(wt, "wt");
   const int N = dy.size(0), P = dy.size(1), Q = dy.size(2), K = dy.size(3);
-  const int R = w.size(1), S = w.size(2), C = w.size(3);
-  TORCH_CHECK(w.size(0) == K, "conv2d_bwd_data: channel mismatch");
-  // contraction runs (r,s,k) with k fastest: B matrix = W permuted to (C,R,S,K)
-  auto wt = w.permute({3, 1, 2, 0}).contiguous();
+  const int C = wt.size(0), R = wt.size(1), S = wt.size(2);
+  TORCH_CHECK(wt.size(3) == K, "conv2d_bwd_data: channel mismatch (wt must be CRSK)");
   auto dx = torch::empty({N, (long)H, (long)W, C}, dy.options());
   al_conv2d_mm(1, dy.data_ptr(), wt.data_ptr(), dx.data_ptr(),
                zero_page(dy).data_ptr(), N, (int)H, (int)W, C, K, R, S, P, Q,

@@ -76,7 +76,7 @@ extern "C" void al_bn_stats(const void* x, float* sum, float* sumsq, long rows, 
   dim3 block(256);
   const int cg = bn_cg_per_block(C);
   const int rpb = 256 / cg;
-  int row_blocks = (int)min((rows + rpb - 1) / rpb, (long)512);
+  int row_blocks = (int)min((rows + rpb - 1) / rpb, (long)2048);
   dim3 grid(row_blocks, (C / 8 + cg - 1) / cg);
   hipLaunchKernelGGL(bn_stats_kernel, grid, block, 0, stream, (const bf16*)x, sum,
                      sumsq, rows, C, cg);
@@ -200,7 +200,7 @@ extern "C" void al_bn_bwd_reduce(const void* dy, const void* x, const void* y,
   dim3 block(256);
   const int cg = bn_cg_per_block(C);
   const int rpb = 256 / cg;
-  int row_blocks = (int)min((rows + rpb - 1) / rpb, (long)512);
+  int row_blocks = (int)min((rows + rpb - 1) / rpb, (long)2048);
   dim3 grid(row_blocks, (C / 8 + cg - 1) / cg);
   if (relu)
     hipLaunchKernelGGL((bn_bwd_reduce_kernel<true>), grid, block, 0, stream,

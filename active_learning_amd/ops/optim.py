@@ -18,6 +18,13 @@ import torch
 from .extension import extension_available, load_extension
 
 
+def _invalidate(p):
+    """Native updates mutate storage without bumping torch's version counter;
+    drop the per-version bf16 weight cache (ops/functional.cast_cached)."""
+    if getattr(p, "_al_cast", None) is not None:
+        p._al_cast = None
+
+
 class FusedSGD(torch.optim.Optimizer):
     def __init__(self, params, lr, momentum=0.0, weight_decay=0.0, dampening=0.0,
                  nesterov=False):
@@ -46,6 +53,7 @@ class FusedSGD(torch.optim.Optimizer):
                 if p.is_cuda and ext is not None:
                     ext.sgd_step(p, g.to(torch.float32), buf if buf is not None else p.new_empty(0),
                                  lr, momentum, wd)
+                    _invalidate(p)
                 else:
                     gf = g.float()
                     if wd != 0:
@@ -86,6 +94,7 @@ class FusedAdam(torch.optim.Optimizer):
                 if p.is_cuda and ext is not None:
                     ext.adam_step(p, p.grad.to(torch.float32), state["exp_avg"],
                                   state["exp_avg_sq"], lr, beta1, beta2, eps, wd, bc1, bc2)
+                    _invalidate(p)
                 else:
                     gf = p.grad.float()
                     if wd != 0:
