@@ -21,6 +21,7 @@ void al_conv2d_wgrad(const void* dy, const void* x, float* dw, int N, int H, int
                      void* stream);
 void al_bn_stats(const void* x, float* sum, float* sumsq, long rows, int C,
                  void* stream);
+int al_bn_reduce_blocks(long rows, int C);
 void al_bn_norm_fwd(const void* x, void* y, const float* mean, const float* invstd,
                     const float* gamma, const float* beta, const void* res, int relu,
                     long rows, int C, void* stream);
@@ -132,11 +133,12 @@ std::vector<Tensor> bn_stats(const Tensor& x) {
   const int C = x.size(-1);
   const long rows = x.numel() / C;
   auto opts = x.options().dtype(torch::kFloat32);
-  auto sum = torch::zeros({C}, opts);
-  auto sumsq = torch::zeros({C}, opts);
-  al_bn_stats(x.data_ptr(), sum.data_ptr<float>(), sumsq.data_ptr<float>(), rows, C,
-              cur_stream());
-  return {sum, sumsq};
+  const int nb = al_bn_reduce_blocks(rows, C);
+  auto part_sum = torch::empty({nb, C}, opts);
+  auto part_sumsq = torch::empty({nb, C}, opts);
+  al_bn_stats(x.data_ptr(), part_sum.data_ptr<float>(), part_sumsq.data_ptr<float>(),
+              rows, C, cur_stream());
+  return {part_sum.sum(0), part_sumsq.sum(0)};
 }
 
 Tensor bn_norm_fwd(const Tensor& x, const Tensor& mean, const Tensor& invstd,
@@ -165,14 +167,15 @@ std::vector<Tensor> bn_bwd_reduce(const Tensor& dy, const Tensor& x, const Tenso
   const int C = x.size(-1);
   const long rows = x.numel() / C;
   auto opts = x.options().dtype(torch::kFloat32);
-  auto sum_dy = torch::zeros({C}, opts);
-  auto sum_dy_xhat = torch::zeros({C}, opts);
+  const int nb = al_bn_reduce_blocks(rows, C);
+  auto part_s = torch::empty({nb, C}, opts);
+  auto part_sx = torch::empty({nb, C}, opts);
   al_bn_bwd_reduce(dy.data_ptr(), x.data_ptr(), y.data_ptr(),
                    mean.contiguous().data_ptr<float>(),
-                   invstd.contiguous().data_ptr<float>(), sum_dy.data_ptr<float>(),
-                   sum_dy_xhat.data_ptr<float>(), relu ? 1 : 0, rows, C,
+                   invstd.contiguous().data_ptr<float>(), part_s.data_ptr<float>(),
+                   part_sx.data_ptr<float>(), relu ? 1 : 0, rows, C,
                    cur_stream());
-  return {sum_dy, sum_dy_xhat};
+  return {part_s.sum(0), part_sx.sum(0)};
 }
 
 std::vector<Tensor> bn_bwd(const Tensor& dy, const Tensor& x, const Tensor& y,

@@ -56,10 +56,12 @@ __global__ void bn_stats_kernel(const bf16* __restrict__ x, float* __restrict__ 
         s[j] += red[0][rl * cg_per_block + cg_local][j];
         ss[j] += red[1][rl * cg_per_block + cg_local][j];
       }
+    // per-block partials: sum[2][gridDim.x][C]; the caller reduces over
+    // blocks (atomics on few addresses serialize at this block count)
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      atomicAdd(&sum[c8 * 8 + j], s[j]);
-      atomicAdd(&sumsq[c8 * 8 + j], ss[j]);
+      sum[(long)blockIdx.x * C + c8 * 8 + j] = s[j];
+      sumsq[(long)blockIdx.x * C + c8 * 8 + j] = ss[j];
     }
   }
 }
@@ -80,6 +82,12 @@ extern "C" void al_bn_stats(const void* x, float* sum, float* sumsq, long rows, 
   dim3 grid(row_blocks, (C / 8 + cg - 1) / cg);
   hipLaunchKernelGGL(bn_stats_kernel, grid, block, 0, stream, (const bf16*)x, sum,
                      sumsq, rows, C, cg);
+}
+
+extern "C" int al_bn_reduce_blocks(long rows, int C) {
+  const int cg = bn_cg_per_block(C);
+  const int rpb = 256 / cg;
+  return (int)min((rows + rpb - 1) / rpb, (long)2048);
 }
 
 // ---------------------------------------------------------------------------
@@ -187,8 +195,8 @@ __global__ void bn_bwd_reduce_kernel(const bf16* __restrict__ dy,
       }
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      atomicAdd(&sum_dy[c8 * 8 + j], s[j]);
-      atomicAdd(&sum_dy_xhat[c8 * 8 + j], sx[j]);
+      sum_dy[(long)blockIdx.x * C + c8 * 8 + j] = s[j];
+      sum_dy_xhat[(long)blockIdx.x * C + c8 * 8 + j] = sx[j];
     }
   }
 }

@@ -63,8 +63,9 @@ def test_conv_bwd_data(case):
     ref = torch.nn.grad.conv2d_input([n, c, h, w], wt.permute(0, 3, 1, 2),
                                      dy.permute(0, 3, 1, 2), stride=stride,
                                      padding=pad).permute(0, 2, 3, 1)
+    wt_perm = wt.permute(3, 1, 2, 0).contiguous()  # binding ABI: (C,R,S,K)
     dx = EXT.conv2d_bwd_data(dy.cuda().to(torch.bfloat16),
-                             wt.cuda().to(torch.bfloat16), stride, pad, h, w)
+                             wt_perm.cuda().to(torch.bfloat16), stride, pad, h, w)
     assert relerr(dx, ref) < 0.02, f"conv bwd_data {case}: relerr {relerr(dx, ref)}"
 
 

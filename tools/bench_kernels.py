@@ -5,7 +5,11 @@ HIP kernel at representative ResNet-50 shapes. Run on a GPU box:
 """
 
 import argparse
+import os
+import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 
