@@ -132,29 +132,103 @@ __global__ void igemm_kernel(const bf16* __restrict__ A, const bf16* __restrict_
 
   const int KT = sh.KD / BK;
 
-  // stage one K-tile (both A and B) into buffer `buf` via glds.
-  // chunk t in [0,1024): row = t>>3, slot u = t&7 (16B each); the data for
-  // swizzled slot u comes from source slot u ^ (row&7).
-  auto stage = [&](int buf, int kt) {
-    const int k0 = kt * BK;
+  // -------------------------------------------------------------------
+  // per-thread staging state, hoisted out of the K-loop: each thread owns
+  // 4 A chunks and 4 B chunks per tile (16 B each). The GEMM row (and its
+  // (n,p,q)/(n,h,w) decomposition) is FIXED per slot; only the contraction
+  // coordinate advances (+BK per tile) with cheap carry propagation — no
+  // divisions inside the loop.
+  // -------------------------------------------------------------------
+  int a_r[4], a_s[4], a_cf[4];          // filter pos + fast-channel coord
+  long a_pix[4];                        // (n, outer-spatial) base index
+  int a_p[4], a_q[4];                   // per-row spatial (fwd: p,q; bwd: h,w)
+  bool a_ok[4];
+  const bf16* b_ptr[4];
+  const int fastC = (MODE == MODE_FWD) ? sh.C : sh.K;
+
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const int t = (wid * 4 + i) * 64 + lane;
+    const int row = t >> 3, u = t & 7;
+    const int usw = u ^ (row & 7);
+    const int kd0 = usw * 8;
+    // contraction decomposition at kd0 (kd0 < 64 <= fastC except tails)
+    int cf = kd0 % fastC;
+    int rs = kd0 / fastC;
+    a_cf[i] = cf;
+    a_s[i] = rs % sh.S;
+    a_r[i] = rs / sh.S;
+    const long m = m0 + row;
+    a_ok[i] = m < sh.M;
+    if (a_ok[i]) {
+      if (MODE == MODE_FWD) {
+        const int q = (int)(m % sh.Q);
+        long tt = m / sh.Q;
+        const int p = (int)(tt % sh.P);
+        const int n = (int)(tt / sh.P);
+        a_pix[i] = (long)n * sh.H * sh.W;
+        a_p[i] = p * sh.stride - sh.pad;
+        a_q[i] = q * sh.stride - sh.pad;
+      } else {
+        const int w = (int)(m % sh.W);
+        long tt = m / sh.W;
+        const int h = (int)(tt % sh.H);
+        const int n = (int)(tt / sh.H);
+        a_pix[i] = (long)n * sh.P * sh.Q;
+        a_p[i] = h + sh.pad;
+        a_q[i] = w + sh.pad;
+      }
+    } else {
+      a_pix[i] = 0; a_p[i] = 0; a_q[i] = 0;
+    }
+    // B: row j fixed, contraction contiguous
+    const int j = row;
+    b_ptr[i] = (j < sh.Nout) ? (B + (long)j * sh.KD + kd0) : nullptr;
+  }
+
+  auto a_src = [&](int i) -> const bf16* {
+    if (!a_ok[i]) return zero;
+    if (MODE == MODE_FWD) {
+      const int h = a_p[i] + a_r[i];
+      const int w = a_q[i] + a_s[i];
+      if ((unsigned)h >= (unsigned)sh.H || (unsigned)w >= (unsigned)sh.W) return zero;
+      return A + (a_pix[i] + (long)h * sh.W + w) * sh.C + a_cf[i];
+    } else {
+      const int hp = a_p[i] - a_r[i];
+      const int wp = a_q[i] - a_s[i];
+      if (hp < 0 || wp < 0) return zero;
+      if (hp % sh.stride || wp % sh.stride) return zero;
+      const int p = hp / sh.stride, q = wp / sh.stride;
+      if (p >= sh.P || q >= sh.Q) return zero;
+      return A + (a_pix[i] + (long)p * sh.Q + q) * sh.K + a_cf[i];
+    }
+  };
+
+  auto advance = [&]() {
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      a_cf[i] += BK;
+      while (a_cf[i] >= fastC) {
+        a_cf[i] -= fastC;
+        if (++a_s[i] == sh.S) { a_s[i] = 0; ++a_r[i]; }
+      }
+      if (b_ptr[i]) b_ptr[i] += BK;
+    }
+  };
+
+  auto stage = [&](int buf) {
     bf16* abase = As + buf * BM * BK;
     bf16* bbase = Bs + buf * BN * BK;
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
-      const int t = (wid * 4 + i) * 64 + lane;
-      const int row = t >> 3, u = t & 7;
-      const int usw = u ^ (row & 7);
-      const bf16* src = a_chunk_ptr<MODE>(A, zero, sh, m0 + row, k0 + usw * 8);
+      const bf16* src = a_src(i);
       __builtin_amdgcn_global_load_lds((const __attribute__((address_space(1))) void*)src,
                                        (__attribute__((address_space(3))) void*)(abase + (wid * 4 + i) * 512),
                                        16, 0, 0);
     }
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
-      const int t = (wid * 4 + i) * 64 + lane;
-      const int row = t >> 3, u = t & 7;
-      const int usw = u ^ (row & 7);
-      const bf16* src = b_chunk_ptr(B, zero, sh, n0 + row, k0 + usw * 8);
+      const bf16* src = b_ptr[i] ? b_ptr[i] : zero;
       __builtin_amdgcn_global_load_lds((const __attribute__((address_space(1))) void*)src,
                                        (__attribute__((address_space(3))) void*)(bbase + (wid * 4 + i) * 512),
                                        16, 0, 0);
@@ -169,12 +243,15 @@ __global__ void igemm_kernel(const bf16* __restrict__ A, const bf16* __restrict_
 
   const int l15 = lane & 15, l4 = lane >> 4;
 
-  stage(0, 0);
+  stage(0);
   int buf = 0;
   for (int kt = 0; kt < KT; ++kt) {
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __syncthreads();
-    if (kt + 1 < KT) stage(buf ^ 1, kt + 1);
+    if (kt + 1 < KT) {
+      advance();
+      stage(buf ^ 1);
+    }
     const bf16* abase = As + buf * BM * BK;
     const bf16* bbase = Bs + buf * BN * BK;
 #pragma unroll

@@ -4,14 +4,17 @@
 //
 // Both operands have the contraction (m) as their SLOW dim, so tiles are
 // staged into LDS TRANSPOSED ([channel][m], +8-element pad keeping rows
-// 16B-aligned and banks spread) and fragments then load as contiguous-m
-// 16B ds_read_b128. Split-K over m with fp32 atomicAdd into the dW
-// accumulator (zeroed by the caller); output fp32 keeps the optimizer's
-// master-weight update exact (SURVEY.md §2.4 bwd-weight row).
+// 16B-aligned) and fragments then load as contiguous-m 16B ds_read_b128.
+// Staging is double-buffered with the async-STAGE split (guide T14 /
+// Guideline 15): the next m-tile's global loads are issued into registers
+// BEFORE the current tile's MFMAs, and the transposed ds_write happens after
+// the barrier — HBM latency hides under the MFMA phase.
+// Split-K over m with fp32 atomicAdd into the dW accumulator (zeroed by the
+// caller); fp32 output feeds the master-weight update exactly.
 //
-// Tile: 64(k) x 64(rsc) x 64(m), 4 waves (2x2), 32x32 per wave,
-// mfma_f32_16x16x32_bf16. Shapes with C % 8 != 0 or K % 8 != 0 (the stems)
-// take the direct fallback.
+// Tile: 64(k) x 64(rsc) x 64(m), 4 waves (2x2), 32x32 per wave. Shapes with
+// C % 8 != 0 or K % 8 != 0 (the stems) take the packed-im2col path in
+// ops/functional.py, so the direct fallback below is a safety net only.
 
 #include "al_common.h"
 
@@ -52,14 +55,20 @@ __global__ void wgrad_kernel(const bf16* __restrict__ dy, const bf16* __restrict
   const int n0 = bn * BNW;
   const long lz0 = (long)blockIdx.y * l_per_z;
   const long lz1 = min(sh.L, lz0 + l_per_z);
+  if (lz0 >= lz1) return;
 
-  __shared__ __attribute__((aligned(16))) bf16 At[BMK][LDT];  // [k][m]
-  __shared__ __attribute__((aligned(16))) bf16 Bt[BNW][LDT];  // [rsc][m]
+  __shared__ __attribute__((aligned(16))) bf16 At[2][BMK][LDT];  // [k][m]
+  __shared__ __attribute__((aligned(16))) bf16 Bt[2][BNW][LDT];  // [rsc][m]
 
   const int tid = threadIdx.x;
   const int wid = tid >> 6, lane = tid & 63;
   const int wr = wid >> 1, wc = wid & 1;
   const int l15 = lane & 15, l4 = lane >> 4;
+
+  // each thread owns two 8-deep channel chunks per operand per tile:
+  //   chunk t in [0,512): ml = t>>3 (m within tile), u = t&7 (channel chunk)
+  const int ml0 = tid >> 3, u0 = tid & 7;           // chunk tid
+  const int ml1 = (tid + 256) >> 3, u1 = tid & 7;   // chunk tid+256
 
   f32x4 acc[2][2];
 #pragma unroll
@@ -67,40 +76,45 @@ __global__ void wgrad_kernel(const bf16* __restrict__ dy, const bf16* __restrict
 #pragma unroll
     for (int j = 0; j < 2; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  for (long m0 = lz0; m0 < lz1; m0 += BL) {
-    __syncthreads();
-    // stage dY^T: 64 m x 64 k -> At[k][m]; 512 chunks of 8 k over 256 threads
+  auto load_tile = [&](long m0, s16x8 va[2], s16x8 vb[2]) {
+    const int mls[2] = {ml0, ml1};
+    const int us[2] = {u0, u1};
 #pragma unroll
     for (int i = 0; i < 2; ++i) {
-      const int t = tid + 256 * i;
-      const int ml = t >> 3, u = t & 7;
-      const long m = m0 + ml;
-      s16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
-      if (m < lz1 && k0 + u * 8 < sh.K)
-        v = *(const s16x8*)(dy + m * sh.K + k0 + u * 8);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) { short sj = v[j]; At[u * 8 + j][ml] = *(bf16*)&sj; }
+      const long m = m0 + mls[i];
+      va[i] = s16x8{0, 0, 0, 0, 0, 0, 0, 0};
+      vb[i] = s16x8{0, 0, 0, 0, 0, 0, 0, 0};
+      if (m < lz1) {
+        if (k0 + us[i] * 8 < sh.K)
+          va[i] = *(const s16x8*)(dy + m * sh.K + k0 + us[i] * 8);
+        const bf16* src;
+        if (x_chunk(x, sh, m, n0 + us[i] * 8, &src)) vb[i] = *(const s16x8*)src;
+      }
     }
-    // stage im2col(x)^T: Bt[rsc][m]
+  };
+
+  auto write_tile = [&](int buf, const s16x8 va[2], const s16x8 vb[2]) {
+    const int mls[2] = {ml0, ml1};
+    const int us[2] = {u0, u1};
 #pragma unroll
     for (int i = 0; i < 2; ++i) {
-      const int t = tid + 256 * i;
-      const int ml = t >> 3, u = t & 7;
-      const long m = m0 + ml;
-      s16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
-      const bf16* src;
-      if (m < lz1 && x_chunk(x, sh, m, n0 + u * 8, &src)) v = *(const s16x8*)src;
 #pragma unroll
-      for (int j = 0; j < 8; ++j) { short sj = v[j]; Bt[u * 8 + j][ml] = *(bf16*)&sj; }
+      for (int j = 0; j < 8; ++j) {
+        short aj = va[i][j], bj = vb[i][j];
+        At[buf][us[i] * 8 + j][mls[i]] = *(bf16*)&aj;
+        Bt[buf][us[i] * 8 + j][mls[i]] = *(bf16*)&bj;
+      }
     }
-    __syncthreads();
+  };
+
+  auto compute = [&](int buf) {
 #pragma unroll
-    for (int mc = 0; mc < 2; ++mc) {  // two 32-deep m chunks
+    for (int mc = 0; mc < 2; ++mc) {
       bf16x8 afrag[2], bfrag[2];
 #pragma unroll
       for (int f = 0; f < 2; ++f) {
-        afrag[f] = *(const bf16x8*)(&At[wr * 32 + f * 16 + l15][mc * 32 + l4 * 8]);
-        bfrag[f] = *(const bf16x8*)(&Bt[wc * 32 + f * 16 + l15][mc * 32 + l4 * 8]);
+        afrag[f] = *(const bf16x8*)(&At[buf][wr * 32 + f * 16 + l15][mc * 32 + l4 * 8]);
+        bfrag[f] = *(const bf16x8*)(&Bt[buf][wc * 32 + f * 16 + l15][mc * 32 + l4 * 8]);
       }
 #pragma unroll
       for (int mi = 0; mi < 2; ++mi)
@@ -109,6 +123,19 @@ __global__ void wgrad_kernel(const bf16* __restrict__ dy, const bf16* __restrict
           acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               afrag[mi], bfrag[ni], acc[mi][ni], 0, 0, 0);
     }
+  };
+
+  s16x8 va[2], vb[2];
+  load_tile(lz0, va, vb);
+  write_tile(0, va, vb);
+  int buf = 0;
+  for (long m0 = lz0; m0 < lz1; m0 += BL) {
+    __syncthreads();
+    if (m0 + BL < lz1) load_tile(m0 + BL, va, vb);  // overlap with MFMAs below
+    compute(buf);
+    __syncthreads();
+    if (m0 + BL < lz1) write_tile(buf ^ 1, va, vb);
+    buf ^= 1;
   }
 
   // accumulate into global dW (fp32): D row = k, col = rsc
@@ -127,12 +154,12 @@ __global__ void wgrad_kernel(const bf16* __restrict__ dy, const bf16* __restrict
   }
 }
 
-// direct fallback: one dW element per thread block-row, strided over L
+// direct fallback: one dW element per thread, strided over L (safety net;
+// normal stems go through the packed-im2col MFMA path)
 __global__ void wgrad_direct_kernel(const bf16* __restrict__ dy,
                                     const bf16* __restrict__ x,
                                     float* __restrict__ dw, WgradShape sh) {
   const long total = (long)sh.K * sh.Nw;
-  // grid.x strides dW elements; grid.y splits L
   const long lz = (sh.L + gridDim.y - 1) / gridDim.y;
   const long m0 = (long)blockIdx.y * lz;
   const long m1 = min(sh.L, m0 + lz);
