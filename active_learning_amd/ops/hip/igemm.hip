@@ -181,8 +181,8 @@ __global__ void igemm_kernel(const bf16* __restrict__ A, const bf16* __restrict_
     } else {
       a_pix[i] = 0; a_p[i] = 0; a_q[i] = 0;
     }
-    // B: row j fixed, contraction contiguous
-    const int j = row;
+    // B: row n0 + row fixed, contraction contiguous
+    const int j = n0 + row;
     b_ptr[i] = (j < sh.Nout) ? (B + (long)j * sh.KD + kd0) : nullptr;
   }
 
