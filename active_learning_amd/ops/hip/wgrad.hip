@@ -44,11 +44,58 @@ AL_DEV bool x_chunk(const bf16* __restrict__ x, const WgradShape& sh, long m, in
   return true;
 }
 
+// ---------------------------------------------------------------------------
+// 8x8 in-register bf16 transpose across a contiguous 8-lane group.
+// Input:  lane l (within its group) holds row m = (l&7): 8 channel values
+// Output: lane l holds channel (l&7): 8 m values
+// 3 butterfly stages exchanging lane-bit k with element-bit k; the
+// element-bit-0 stage mixes halves of dwords via v_perm_b32.
+// ---------------------------------------------------------------------------
+typedef unsigned int u32;
+
+AL_DEV void xpose8x8(u32 d[4], int lane) {
+  // stage 1: lane bit0 <-> element bit0 (within-dword halves)
+  {
+    u32 x[4];
+#pragma unroll
+    for (int k = 0; k < 4; ++k) x[k] = __shfl_xor((int)d[k], 1, 64);
+    const bool hi = lane & 1;
+#pragma unroll
+    for (int k = 0; k < 4; ++k)
+      d[k] = hi ? __builtin_amdgcn_perm(d[k], x[k], 0x07060302)
+                : __builtin_amdgcn_perm(x[k], d[k], 0x05040100);
+  }
+  // stage 2: lane bit1 <-> element bit1 (dword pairs 0<->1, 2<->3)
+  {
+    u32 x[4];
+#pragma unroll
+    for (int k = 0; k < 4; ++k) x[k] = __shfl_xor((int)d[k], 2, 64);
+    const bool hi = lane & 2;
+    u32 n0 = hi ? x[1] : d[0];
+    u32 n1 = hi ? d[1] : x[0];
+    u32 n2 = hi ? x[3] : d[2];
+    u32 n3 = hi ? d[3] : x[2];
+    d[0] = n0; d[1] = n1; d[2] = n2; d[3] = n3;
+  }
+  // stage 3: lane bit2 <-> element bit2 (dword pairs 0<->2, 1<->3)
+  {
+    u32 x[4];
+#pragma unroll
+    for (int k = 0; k < 4; ++k) x[k] = __shfl_xor((int)d[k], 4, 64);
+    const bool hi = lane & 4;
+    u32 n0 = hi ? x[2] : d[0];
+    u32 n1 = hi ? x[3] : d[1];
+    u32 n2 = hi ? d[2] : x[0];
+    u32 n3 = hi ? d[3] : x[1];
+    d[0] = n0; d[1] = n1; d[2] = n2; d[3] = n3;
+  }
+}
+
 __launch_bounds__(256)
 __global__ void wgrad_kernel(const bf16* __restrict__ dy, const bf16* __restrict__ x,
                              float* __restrict__ dw, WgradShape sh, int grid_k,
                              long l_per_z) {
-  constexpr int BMK = 64, BNW = 64, BL = 64, PAD = 8, LDT = BL + PAD;  // 72
+  constexpr int BMK = 64, BNW = 64, BL = 64;
   const int bk = blockIdx.x % grid_k;
   const int bn = blockIdx.x / grid_k;
   const int k0 = bk * BMK;
@@ -57,18 +104,17 @@ __global__ void wgrad_kernel(const bf16* __restrict__ dy, const bf16* __restrict
   const long lz1 = min(sh.L, lz0 + l_per_z);
   if (lz0 >= lz1) return;
 
-  __shared__ __attribute__((aligned(16))) bf16 At[2][BMK][LDT];  // [k][m]
-  __shared__ __attribute__((aligned(16))) bf16 Bt[2][BNW][LDT];  // [rsc][m]
+  // [ch][m] tiles, XOR-swizzled 16B slots (slot ^= ch&7): conflict-free
+  // ds_write_b128 after the register transpose, igemm-style ds_read_b128.
+  __shared__ __attribute__((aligned(16))) bf16 At[2][BMK][BL];
+  __shared__ __attribute__((aligned(16))) bf16 Bt[2][BNW][BL];
 
   const int tid = threadIdx.x;
   const int wid = tid >> 6, lane = tid & 63;
   const int wr = wid >> 1, wc = wid & 1;
   const int l15 = lane & 15, l4 = lane >> 4;
-
-  // each thread owns two 8-deep channel chunks per operand per tile:
-  //   chunk t in [0,512): ml = t>>3 (m within tile), u = t&7 (channel chunk)
-  const int ml0 = tid >> 3, u0 = tid & 7;           // chunk tid
-  const int ml1 = (tid + 256) >> 3, u1 = tid & 7;   // chunk tid+256
+  const int u = lane >> 3;       // channel chunk (8 ch)
+  const int mo = lane & 7;       // m offset within the wave's 8-m slice
 
   f32x4 acc[2][2];
 #pragma unroll
@@ -76,33 +122,35 @@ __global__ void wgrad_kernel(const bf16* __restrict__ dy, const bf16* __restrict
 #pragma unroll
     for (int j = 0; j < 2; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
 
+  // each (wave, i) covers m-slice m0 + wid*8 + 32*i (+mo), all 64 channels
   auto load_tile = [&](long m0, s16x8 va[2], s16x8 vb[2]) {
-    const int mls[2] = {ml0, ml1};
-    const int us[2] = {u0, u1};
 #pragma unroll
     for (int i = 0; i < 2; ++i) {
-      const long m = m0 + mls[i];
+      const long m = m0 + wid * 8 + 32 * i + mo;
       va[i] = s16x8{0, 0, 0, 0, 0, 0, 0, 0};
       vb[i] = s16x8{0, 0, 0, 0, 0, 0, 0, 0};
       if (m < lz1) {
-        if (k0 + us[i] * 8 < sh.K)
-          va[i] = *(const s16x8*)(dy + m * sh.K + k0 + us[i] * 8);
+        if (k0 + u * 8 < sh.K)
+          va[i] = *(const s16x8*)(dy + m * sh.K + k0 + u * 8);
         const bf16* src;
-        if (x_chunk(x, sh, m, n0 + us[i] * 8, &src)) vb[i] = *(const s16x8*)src;
+        if (x_chunk(x, sh, m, n0 + u * 8, &src)) vb[i] = *(const s16x8*)src;
       }
     }
   };
 
-  auto write_tile = [&](int buf, const s16x8 va[2], const s16x8 vb[2]) {
-    const int mls[2] = {ml0, ml1};
-    const int us[2] = {u0, u1};
+  auto write_tile = [&](int buf, s16x8 va[2], s16x8 vb[2]) {
 #pragma unroll
     for (int i = 0; i < 2; ++i) {
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        short aj = va[i][j], bj = vb[i][j];
-        At[buf][us[i] * 8 + j][mls[i]] = *(bf16*)&aj;
-        Bt[buf][us[i] * 8 + j][mls[i]] = *(bf16*)&bj;
+      const int slot = wid + 4 * i;                    // m-slot (16B = 8 m)
+      u32* da = (u32*)&va[i];
+      u32* db = (u32*)&vb[i];
+      xpose8x8(da, lane);
+      xpose8x8(db, lane);
+      const int cha = u * 8 + mo;                      // ch row = lane
+      {
+        const int sw = slot ^ (cha & 7);
+        *(s16x8*)(&At[buf][cha][sw * 8]) = va[i];
+        *(s16x8*)(&Bt[buf][cha][sw * 8]) = vb[i];
       }
     }
   };
@@ -113,8 +161,12 @@ __global__ void wgrad_kernel(const bf16* __restrict__ dy, const bf16* __restrict
       bf16x8 afrag[2], bfrag[2];
 #pragma unroll
       for (int f = 0; f < 2; ++f) {
-        afrag[f] = *(const bf16x8*)(&At[buf][wr * 32 + f * 16 + l15][mc * 32 + l4 * 8]);
-        bfrag[f] = *(const bf16x8*)(&Bt[buf][wc * 32 + f * 16 + l15][mc * 32 + l4 * 8]);
+        const int ar = wr * 32 + f * 16 + l15;
+        const int as = (mc * 4 + l4) ^ (ar & 7);
+        afrag[f] = *(const bf16x8*)(&At[buf][ar][as * 8]);
+        const int br = wc * 32 + f * 16 + l15;
+        const int bs = (mc * 4 + l4) ^ (br & 7);
+        bfrag[f] = *(const bf16x8*)(&Bt[buf][br][bs * 8]);
       }
 #pragma unroll
       for (int mi = 0; mi < 2; ++mi)
