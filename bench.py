@@ -35,25 +35,39 @@ def get_args():
 
 def main():
     args = get_args()
-    assert torch.cuda.is_available(), "bench.py requires a GPU"
+    # test hooks: AL_BENCH_DEVICE=cpu + AL_BENCH_BACKEND=gloo let the full
+    # distributed bench flow (rendezvous, BucketedDDP, JSON contract) run in
+    # CI without GPUs; the real path is cuda + nccl(=RCCL).
+    dev_kind = os.environ.get("AL_BENCH_DEVICE", "cuda")
+    backend = os.environ.get("AL_BENCH_BACKEND", "nccl")
+    if dev_kind == "cuda":
+        assert torch.cuda.is_available(), "bench.py requires a GPU"
 
     rank = int(os.environ.get("RANK", 0))
     world = int(os.environ.get("WORLD_SIZE", 1))
     local_rank = int(os.environ.get("LOCAL_RANK", 0))
-    torch.cuda.set_device(local_rank)
-    device = torch.device("cuda", local_rank)
+    if dev_kind == "cuda":
+        torch.cuda.set_device(local_rank)
+        device = torch.device("cuda", local_rank)
+    else:
+        device = torch.device("cpu")
 
     dist = None
     if world > 1:
         import torch.distributed as dist_mod
         dist = dist_mod
-        dist.init_process_group("nccl")  # RCCL over xGMI
+        dist.init_process_group(backend)  # RCCL over xGMI on GPU
 
     from active_learning_amd.models.ssl_resnet import ResNetSimCLR
-    from active_learning_amd.ops.extension import require_extension
     from active_learning_amd.ops.loss import cross_entropy
     from active_learning_amd.ops.optim import FusedSGD
-    require_extension()
+    if dev_kind == "cuda":
+        from active_learning_amd.ops.extension import require_extension
+        require_extension()
+
+    def sync():
+        if dev_kind == "cuda":
+            torch.cuda.synchronize()
 
     torch.manual_seed(1234 + rank)
     base = "resnet50" if args.model == "resnet50" else "resnet18"
@@ -84,18 +98,18 @@ def main():
     net.train()
     for i in range(args.warmup):
         step(i)
-    torch.cuda.synchronize()
+    sync()
     if dist:
         dist.barrier()
-        torch.cuda.synchronize()
+        sync()
 
     t0 = time.perf_counter()
     for i in range(args.steps):
         step(i)
-    torch.cuda.synchronize()
+    sync()
     if dist:
         dist.barrier()
-        torch.cuda.synchronize()
+        sync()
     elapsed = time.perf_counter() - t0
 
     if dist:
