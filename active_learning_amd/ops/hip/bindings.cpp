@@ -22,6 +22,8 @@ void al_conv2d_wgrad(const void* dy, const void* x, float* dw, int N, int H, int
 void al_bn_stats(const void* x, float* sum, float* sumsq, long rows, int C,
                  void* stream);
 int al_bn_reduce_blocks(long rows, int C);
+void al_bn_part_reduce(const float* part_a, const float* part_b, float* out_a,
+                       float* out_b, int nb, int C, void* stream);
 void al_bn_norm_fwd(const void* x, void* y, const float* mean, const float* invstd,
                     const float* gamma, const float* beta, const void* res, int relu,
                     long rows, int C, void* stream);
@@ -138,7 +140,12 @@ std::vector<Tensor> bn_stats(const Tensor& x) {
   auto part_sumsq = torch::empty({nb, C}, opts);
   al_bn_stats(x.data_ptr(), part_sum.data_ptr<float>(), part_sumsq.data_ptr<float>(),
               rows, C, cur_stream());
-  return {part_sum.sum(0), part_sumsq.sum(0)};
+  auto sum = torch::empty({C}, opts);
+  auto sumsq = torch::empty({C}, opts);
+  al_bn_part_reduce(part_sum.data_ptr<float>(), part_sumsq.data_ptr<float>(),
+                    sum.data_ptr<float>(), sumsq.data_ptr<float>(), nb, C,
+                    cur_stream());
+  return {sum, sumsq};
 }
 
 Tensor bn_norm_fwd(const Tensor& x, const Tensor& mean, const Tensor& invstd,
@@ -175,7 +182,12 @@ std::vector<Tensor> bn_bwd_reduce(const Tensor& dy, const Tensor& x, const Tenso
                    invstd.contiguous().data_ptr<float>(), part_s.data_ptr<float>(),
                    part_sx.data_ptr<float>(), relu ? 1 : 0, rows, C,
                    cur_stream());
-  return {part_s.sum(0), part_sx.sum(0)};
+  auto sum_dy = torch::empty({C}, opts);
+  auto sum_dy_xhat = torch::empty({C}, opts);
+  al_bn_part_reduce(part_s.data_ptr<float>(), part_sx.data_ptr<float>(),
+                    sum_dy.data_ptr<float>(), sum_dy_xhat.data_ptr<float>(), nb, C,
+                    cur_stream());
+  return {sum_dy, sum_dy_xhat};
 }
 
 std::vector<Tensor> bn_bwd(const Tensor& dy, const Tensor& x, const Tensor& y,

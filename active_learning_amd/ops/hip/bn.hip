@@ -289,3 +289,30 @@ extern "C" void al_bn_bwd(const void* dy, const void* x, const void* y,
   }
 #undef CASE
 }
+
+// ---------------------------------------------------------------------------
+// stage-2: reduce the per-block partials of BOTH arrays in one launch
+// (replaces two torch .sum(0) calls per BN op — launch-bound at 53 BN layers)
+// ---------------------------------------------------------------------------
+
+__global__ void bn_part_reduce_kernel(const float* __restrict__ part_a,
+                                      const float* __restrict__ part_b,
+                                      float* __restrict__ out_a,
+                                      float* __restrict__ out_b, int nb, int C) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float sa = 0.f, sb = 0.f;
+  for (int b = 0; b < nb; ++b) {
+    sa += part_a[(long)b * C + c];
+    sb += part_b[(long)b * C + c];
+  }
+  out_a[c] = sa;
+  out_b[c] = sb;
+}
+
+extern "C" void al_bn_part_reduce(const float* part_a, const float* part_b,
+                                  float* out_a, float* out_b, int nb, int C,
+                                  hipStream_t stream) {
+  hipLaunchKernelGGL(bn_part_reduce_kernel, dim3((C + 255) / 256), dim3(256), 0,
+                     stream, part_a, part_b, out_a, out_b, nb, C);
+}

@@ -379,16 +379,40 @@ extern "C" void al_conv2d_mm(int mode, const void* A, const void* B, void* out,
 
 __global__ void im2col_pack_kernel(const bf16* __restrict__ x, bf16* __restrict__ out,
                                    ConvShape sh, int kdpad) {
-  const long total = sh.M * kdpad;
-  for (long i = grid_stride_begin(); i < total; i += grid_stride_step()) {
-    const long m = i / kdpad;
-    const int kd = (int)(i % kdpad);
-    float v = 0.f;
-    if (kd < sh.KD) {
-      const bf16* p = a_chunk_ptr<MODE_FWD>(x, nullptr, sh, m, kd);
-      if (p) v = bf2f(*p);
+  // thread handles 16 consecutive kd of one m row: one coordinate decode,
+  // carry-incremented across the segment, one s16x8-pair store
+  const long total16 = sh.M * (kdpad / 16);
+  for (long i = grid_stride_begin(); i < total16; i += grid_stride_step()) {
+    const int seg = (int)(i % (kdpad / 16));
+    const long m = i / (kdpad / 16);
+    int kd = seg * 16;
+    // decode m once
+    const int q = (int)(m % sh.Q);
+    long t = m / sh.Q;
+    const int p = (int)(t % sh.P);
+    const int n = (int)(t / sh.P);
+    const long pix_base = (long)n * sh.H * sh.W;
+    const int h0 = p * sh.stride - sh.pad;
+    const int w0 = q * sh.stride - sh.pad;
+    int c = kd % sh.C;
+    int rs = kd / sh.C;
+    int s = rs % sh.S;
+    int r = rs / sh.S;
+    s16x8 o[2] = {{0,0,0,0,0,0,0,0},{0,0,0,0,0,0,0,0}};
+    short* os = (short*)o;
+#pragma unroll 1
+    for (int j = 0; j < 16; ++j) {
+      if (kd + j < sh.KD) {
+        const int h = h0 + r, w = w0 + s;
+        if ((unsigned)h < (unsigned)sh.H && (unsigned)w < (unsigned)sh.W) {
+          const bf16 v = x[(pix_base + (long)h * sh.W + w) * sh.C + c];
+          os[j] = *(const short*)&v;
+        }
+        if (++c == sh.C) { c = 0; if (++s == sh.S) { s = 0; ++r; } }
+      }
     }
-    out[i] = f2bf(v);
+    ((s16x8*)out)[i * 2] = o[0];
+    ((s16x8*)out)[i * 2 + 1] = o[1];
   }
 }
 
