@@ -299,20 +299,38 @@ __global__ void bn_part_reduce_kernel(const float* __restrict__ part_a,
                                       const float* __restrict__ part_b,
                                       float* __restrict__ out_a,
                                       float* __restrict__ out_b, int nb, int C) {
-  const int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
+  // block = 8 channels x 32 block-lanes; each lane strides the nb partials,
+  // then an LDS tree folds the 32 lanes per channel.
+  const int cl = threadIdx.x & 7;            // channel within the block's 8
+  const int bl = threadIdx.x >> 3;           // 32 b-lanes
+  const int c = blockIdx.x * 8 + cl;
+  __shared__ float red[2][256];
   float sa = 0.f, sb = 0.f;
-  for (int b = 0; b < nb; ++b) {
-    sa += part_a[(long)b * C + c];
-    sb += part_b[(long)b * C + c];
+  if (c < C) {
+    for (int b = bl; b < nb; b += 32) {
+      sa += part_a[(long)b * C + c];
+      sb += part_b[(long)b * C + c];
+    }
   }
-  out_a[c] = sa;
-  out_b[c] = sb;
+  red[0][threadIdx.x] = sa;
+  red[1][threadIdx.x] = sb;
+  __syncthreads();
+  for (int step = 128; step >= 8; step >>= 1) {
+    if (threadIdx.x < step) {
+      red[0][threadIdx.x] += red[0][threadIdx.x + step];
+      red[1][threadIdx.x] += red[1][threadIdx.x + step];
+    }
+    __syncthreads();
+  }
+  if (threadIdx.x < 8 && c < C) {
+    out_a[c] = red[0][threadIdx.x];
+    out_b[c] = red[1][threadIdx.x];
+  }
 }
 
 extern "C" void al_bn_part_reduce(const float* part_a, const float* part_b,
                                   float* out_a, float* out_b, int nb, int C,
                                   hipStream_t stream) {
-  hipLaunchKernelGGL(bn_part_reduce_kernel, dim3((C + 255) / 256), dim3(256), 0,
+  hipLaunchKernelGGL(bn_part_reduce_kernel, dim3((C + 7) / 8), dim3(256), 0,
                      stream, part_a, part_b, out_a, out_b, nb, C);
 }
