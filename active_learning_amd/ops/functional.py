@@ -56,7 +56,7 @@ def wt_cached(w_c):
     """(K,R,S,C) -> (C,R,S,K) permutation for the bwd-data contraction,
     cached on the bf16 weight copy (which is itself version-cached)."""
     wt = getattr(w_c, "_al_wt", None)
-    if wt is None:
+    if wt is None:  # also reset to None by FusedSGD after in-place refresh
         wt = w_c.permute(3, 1, 2, 0).contiguous()
         try:
             w_c._al_wt = wt

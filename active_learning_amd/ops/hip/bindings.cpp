@@ -48,7 +48,7 @@ void al_im2col_pack(const void* x, void* out, int N, int H, int W, int C, int R,
                     int S, int P, int Q, int stride, int pad, int kdpad,
                     void* stream);
 void al_sgd_step(float* p, const float* g, float* buf, float lr, float momentum,
-                 float wd, long n, void* stream);
+                 float wd, long n, void* shadow, void* stream);
 void al_adam_step(float* p, const float* g, float* m, float* v, float lr, float b1,
                   float b2, float eps, float wd, float bc1, float bc2, long n,
                   void* stream);
@@ -277,11 +277,17 @@ Tensor ce_bwd(const Tensor& probs, const Tensor& targets, const Tensor& scale) {
 }
 
 void sgd_step(Tensor& p, const Tensor& g, Tensor& buf, double lr, double momentum,
-              double wd) {
+              double wd, const Tensor& shadow) {
   TORCH_CHECK(p.is_cuda() && p.scalar_type() == torch::kFloat32);
+  void* sh = nullptr;
+  if (shadow.numel()) {
+    TORCH_CHECK(shadow.numel() == p.numel() &&
+                shadow.scalar_type() == torch::kBFloat16);
+    sh = shadow.data_ptr();
+  }
   al_sgd_step(p.data_ptr<float>(), g.contiguous().data_ptr<float>(),
               buf.numel() ? buf.data_ptr<float>() : nullptr, (float)lr,
-              (float)momentum, (float)wd, p.numel(), cur_stream());
+              (float)momentum, (float)wd, p.numel(), sh, cur_stream());
 }
 
 void adam_step(Tensor& p, const Tensor& g, Tensor& m, Tensor& v, double lr, double b1,

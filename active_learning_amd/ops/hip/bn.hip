@@ -73,12 +73,21 @@ static inline int bn_cg_per_block(int C) {
   return cg;  // largest power-of-two divisor of C/8, capped at 32
 }
 
+static inline int bn_row_block_cap(int C) {
+  // bound the partial buffer (nb x C fp32) to ~512 KiB: wide layers need
+  // fewer row blocks, narrow layers keep the full 2048 for bandwidth
+  long cap = 131072 / C;
+  if (cap < 128) cap = 128;
+  if (cap > 2048) cap = 2048;
+  return (int)cap;
+}
+
 extern "C" void al_bn_stats(const void* x, float* sum, float* sumsq, long rows, int C,
                             hipStream_t stream) {
   dim3 block(256);
   const int cg = bn_cg_per_block(C);
   const int rpb = 256 / cg;
-  int row_blocks = (int)min((rows + rpb - 1) / rpb, (long)2048);
+  int row_blocks = (int)min((rows + rpb - 1) / rpb, (long)bn_row_block_cap(C));
   dim3 grid(row_blocks, (C / 8 + cg - 1) / cg);
   hipLaunchKernelGGL(bn_stats_kernel, grid, block, 0, stream, (const bf16*)x, sum,
                      sumsq, rows, C, cg);
@@ -87,7 +96,7 @@ extern "C" void al_bn_stats(const void* x, float* sum, float* sumsq, long rows, 
 extern "C" int al_bn_reduce_blocks(long rows, int C) {
   const int cg = bn_cg_per_block(C);
   const int rpb = 256 / cg;
-  return (int)min((rows + rpb - 1) / rpb, (long)2048);
+  return (int)min((rows + rpb - 1) / rpb, (long)bn_row_block_cap(C));
 }
 
 // ---------------------------------------------------------------------------
@@ -208,7 +217,7 @@ extern "C" void al_bn_bwd_reduce(const void* dy, const void* x, const void* y,
   dim3 block(256);
   const int cg = bn_cg_per_block(C);
   const int rpb = 256 / cg;
-  int row_blocks = (int)min((rows + rpb - 1) / rpb, (long)2048);
+  int row_blocks = (int)min((rows + rpb - 1) / rpb, (long)bn_row_block_cap(C));
   dim3 grid(row_blocks, (C / 8 + cg - 1) / cg);
   if (relu)
     hipLaunchKernelGGL((bn_bwd_reduce_kernel<true>), grid, block, 0, stream,

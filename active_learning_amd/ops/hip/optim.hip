@@ -5,9 +5,10 @@
 
 #include "al_common.h"
 
+template <bool SHADOW>
 __global__ void sgd_kernel(float* __restrict__ p, const float* __restrict__ g,
                            float* __restrict__ buf, float lr, float momentum,
-                           float wd, long n) {
+                           float wd, long n, bf16* __restrict__ shadow) {
   for (long i = grid_stride_begin(); i < n; i += grid_stride_step()) {
     float grad = g[i] + wd * p[i];
     if (momentum != 0.f) {
@@ -15,15 +16,22 @@ __global__ void sgd_kernel(float* __restrict__ p, const float* __restrict__ g,
       buf[i] = b;
       grad = b;
     }
-    p[i] -= lr * grad;
+    const float np = p[i] - lr * grad;
+    p[i] = np;
+    if (SHADOW) shadow[i] = f2bf(np);  // refresh the bf16 weight copy in-pass
   }
 }
 
 extern "C" void al_sgd_step(float* p, const float* g, float* buf, float lr,
-                            float momentum, float wd, long n, hipStream_t stream) {
+                            float momentum, float wd, long n, void* shadow,
+                            hipStream_t stream) {
   int blocks = (int)min((n + 255) / 256, (long)2048);
-  hipLaunchKernelGGL(sgd_kernel, dim3(blocks), dim3(256), 0, stream, p, g, buf, lr,
-                     momentum, wd, n);
+  if (shadow)
+    hipLaunchKernelGGL((sgd_kernel<true>), dim3(blocks), dim3(256), 0, stream, p, g,
+                       buf, lr, momentum, wd, n, (bf16*)shadow);
+  else
+    hipLaunchKernelGGL((sgd_kernel<false>), dim3(blocks), dim3(256), 0, stream, p, g,
+                       buf, lr, momentum, wd, n, nullptr);
 }
 
 __global__ void adam_kernel(float* __restrict__ p, const float* __restrict__ g,

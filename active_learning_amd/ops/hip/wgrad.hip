@@ -95,7 +95,7 @@ __launch_bounds__(256)
 __global__ void wgrad_kernel(const bf16* __restrict__ dy, const bf16* __restrict__ x,
                              float* __restrict__ dw, WgradShape sh, int grid_k,
                              long l_per_z) {
-  constexpr int BMK = 64, BNW = 128, BL = 64;
+  constexpr int BMK = 128, BNW = 128, BL = 64;
   const int bk = blockIdx.x % grid_k;
   const int bn = blockIdx.x / grid_k;
   const int k0 = bk * BMK;
@@ -116,70 +116,58 @@ __global__ void wgrad_kernel(const bf16* __restrict__ dy, const bf16* __restrict
   const int u = lane >> 3;       // channel chunk (8 ch)
   const int mo = lane & 7;       // m offset within the wave's 8-m slice
 
-  f32x4 acc[2][4];
+  f32x4 acc[4][4];
 #pragma unroll
-  for (int i = 0; i < 2; ++i)
+  for (int i = 0; i < 4; ++i)
 #pragma unroll
     for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  // A (dY): 2 sub-batches cover 64 m x 8 k-chunks; B (im2col x): 4 sub-
-  // batches cover 64 m x 16 rsc-chunks. Per (wave, i): one 8-m slice.
-  auto load_tile = [&](long m0, s16x8 va[2], s16x8 vb[4]) {
-#pragma unroll
-    for (int i = 0; i < 2; ++i) {
-      const long m = m0 + wid * 8 + 32 * i + mo;
-      va[i] = s16x8{0, 0, 0, 0, 0, 0, 0, 0};
-      if (m < lz1 && k0 + u * 8 < sh.K)
-        va[i] = *(const s16x8*)(dy + m * sh.K + k0 + u * 8);
-    }
+  // A (dY) and B (im2col x) each: 4 sub-batches of (8-m slice x 8-ch chunk)
+  // cover 64 m x 16 chunks (128 channels).
+  auto load_tile = [&](long m0, s16x8 va[4], s16x8 vb[4]) {
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
       const long m = m0 + wid * 8 + 32 * (i >> 1) + mo;
-      const int ub = u + 8 * (i & 1);
+      const int uc = u + 8 * (i & 1);
+      va[i] = s16x8{0, 0, 0, 0, 0, 0, 0, 0};
       vb[i] = s16x8{0, 0, 0, 0, 0, 0, 0, 0};
-      const bf16* src;
-      if (m < lz1 && x_chunk(x, sh, m, n0 + ub * 8, &src))
-        vb[i] = *(const s16x8*)src;
+      if (m < lz1) {
+        if (k0 + uc * 8 < sh.K)
+          va[i] = *(const s16x8*)(dy + m * sh.K + k0 + uc * 8);
+        const bf16* src;
+        if (x_chunk(x, sh, m, n0 + uc * 8, &src)) vb[i] = *(const s16x8*)src;
+      }
     }
   };
 
-  auto write_tile = [&](int buf, s16x8 va[2], s16x8 vb[4]) {
-#pragma unroll
-    for (int i = 0; i < 2; ++i) {
-      const int slot = wid + 4 * i;                    // m-slot (16B = 8 m)
-      xpose8x8((u32*)&va[i], lane);
-      const int cha = u * 8 + mo;                      // ch row = lane
-      const int sw = slot ^ (cha & 7);
-      *(s16x8*)(&At[buf][cha][sw * 8]) = va[i];
-    }
+  auto write_tile = [&](int buf, s16x8 va[4], s16x8 vb[4]) {
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
-      const int slot = wid + 4 * (i >> 1);
+      const int slot = wid + 4 * (i >> 1);             // m-slot (16B = 8 m)
+      const int ch = (u + 8 * (i & 1)) * 8 + mo;       // ch row
+      const int sw = slot ^ (ch & 7);
+      xpose8x8((u32*)&va[i], lane);
+      *(s16x8*)(&At[buf][ch][sw * 8]) = va[i];
       xpose8x8((u32*)&vb[i], lane);
-      const int chb = (u + 8 * (i & 1)) * 8 + mo;
-      const int sw = slot ^ (chb & 7);
-      *(s16x8*)(&Bt[buf][chb][sw * 8]) = vb[i];
+      *(s16x8*)(&Bt[buf][ch][sw * 8]) = vb[i];
     }
   };
 
   auto compute = [&](int buf) {
 #pragma unroll
     for (int mc = 0; mc < 2; ++mc) {
-      bf16x8 afrag[2], bfrag[4];
-#pragma unroll
-      for (int f = 0; f < 2; ++f) {
-        const int ar = wr * 32 + f * 16 + l15;
-        const int as = (mc * 4 + l4) ^ (ar & 7);
-        afrag[f] = *(const bf16x8*)(&At[buf][ar][as * 8]);
-      }
+      bf16x8 afrag[4], bfrag[4];
 #pragma unroll
       for (int f = 0; f < 4; ++f) {
+        const int ar = wr * 64 + f * 16 + l15;
+        const int as = (mc * 4 + l4) ^ (ar & 7);
+        afrag[f] = *(const bf16x8*)(&At[buf][ar][as * 8]);
         const int br = wc * 64 + f * 16 + l15;
         const int bs = (mc * 4 + l4) ^ (br & 7);
         bfrag[f] = *(const bf16x8*)(&Bt[buf][br][bs * 8]);
       }
 #pragma unroll
-      for (int mi = 0; mi < 2; ++mi)
+      for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
         for (int ni = 0; ni < 4; ++ni)
           acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
@@ -187,7 +175,7 @@ __global__ void wgrad_kernel(const bf16* __restrict__ dy, const bf16* __restrict
     }
   };
 
-  s16x8 va[2], vb[4];
+  s16x8 va[4], vb[4];
   load_tile(lz0, va, vb);
   write_tile(0, va, vb);
   int buf = 0;
@@ -202,14 +190,14 @@ __global__ void wgrad_kernel(const bf16* __restrict__ dy, const bf16* __restrict
 
   // accumulate into global dW (fp32): D row = k, col = rsc
 #pragma unroll
-  for (int mi = 0; mi < 2; ++mi) {
+  for (int mi = 0; mi < 4; ++mi) {
 #pragma unroll
     for (int ni = 0; ni < 4; ++ni) {
       const int col = n0 + wc * 64 + ni * 16 + l15;
       if (col >= sh.Nw) continue;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        const int row = k0 + wr * 32 + mi * 16 + l4 * 4 + r;
+        const int row = k0 + wr * 64 + mi * 16 + l4 * 4 + r;
         if (row < sh.K) atomicAdd(&dw[(long)row * sh.Nw + col], acc[mi][ni][r]);
       }
     }
@@ -249,7 +237,7 @@ extern "C" void al_conv2d_wgrad(const void* dy, const void* x, float* dw, int N,
   sh.L = (long)N * P * Q;
   sh.Nw = R * S * C;
   if (C % 8 == 0 && K % 8 == 0) {
-    const int grid_k = (K + 63) / 64;
+    const int grid_k = (K + 127) / 128;
     const int grid_n = (sh.Nw + 127) / 128;
     const int tiles = grid_k * grid_n;
     // split-K: aim for >= 512 blocks to fill 256 CUs

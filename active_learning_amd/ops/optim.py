@@ -51,9 +51,22 @@ class FusedSGD(torch.optim.Optimizer):
                     state["momentum_buffer"] = torch.zeros_like(p, dtype=torch.float32)
                 buf = state.get("momentum_buffer")
                 if p.is_cuda and ext is not None:
-                    ext.sgd_step(p, g.to(torch.float32), buf if buf is not None else p.new_empty(0),
-                                 lr, momentum, wd)
-                    _invalidate(p)
+                    cache = getattr(p, "_al_cast", None)
+                    shadow = (cache[1] if cache is not None
+                              and cache[1].numel() == p.numel()
+                              and cache[1].dtype == torch.bfloat16
+                              else p.new_empty(0, dtype=torch.bfloat16))
+                    ext.sgd_step(p, g.to(torch.float32),
+                                 buf if buf is not None else p.new_empty(0),
+                                 lr, momentum, wd, shadow)
+                    if shadow.numel():
+                        # the kernel refreshed the bf16 copy in-pass; derived
+                        # caches (bwd-data permute) must be recomputed
+                        if getattr(cache[1], "_al_wt", None) is not None:
+                            cache[1]._al_wt = None
+                        p._al_cast = (p._version, cache[1])
+                    else:
+                        _invalidate(p)
                 else:
                     gf = g.float()
                     if wd != 0:
