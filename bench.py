@@ -25,7 +25,7 @@ def get_args():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=30)
     p.add_argument("--warmup", type=int, default=10)
-    p.add_argument("--batch", type=int, default=128, help="per-GPU batch size")
+    p.add_argument("--batch", type=int, default=256, help="per-GPU batch size")
     p.add_argument("--model", default="resnet50", choices=["resnet50", "resnet18"])
     p.add_argument("--img", type=int, default=224)
     p.add_argument("--classes", type=int, default=1000)

@@ -1,0 +1,135 @@
+#!/usr/bin/env python3
+"""Query-round wall-clock benchmarks (BASELINE.json configs 2, 4, 5).
+
+Measures the MI355X-native query path with synthetic data and random-init
+weights, everything device-resident:
+  * config 2: ResNet-18/CIFAR entropy + margin scoring over a 50k pool
+  * config 4: SSLResNet50 linear-eval BADGE (factorized grad-embed Gram) +
+    k-means++ seeding, budget 10k over a 50k-labeled + 80k-unlabeled subset
+  * config 5: ResNet-50 coreset greedy k-center with the full-pool pairwise
+    distance matrix resident in HBM (N=130k -> 68 GB fp32)
+
+Pool inference uses on-device synthetic batches (no disk datasets exist in
+this environment); scoring/selection math is identical to the samplers'.
+"""
+
+import argparse
+import json
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import torch
+
+
+def sync():
+    torch.cuda.synchronize()
+
+
+def timed(label, fn):
+    sync()
+    t0 = time.perf_counter()
+    out = fn()
+    sync()
+    dt = time.perf_counter() - t0
+    print(f"  {label}: {dt:.3f} s", flush=True)
+    return out, dt
+
+
+@torch.no_grad()
+def pool_inference(net, n_images, batch, img, want_embedding=True):
+    """Forward the pool through the net, keeping logits/embeddings in HBM."""
+    logits_l, emb_l = [], []
+    done = 0
+    x = torch.randn(batch, 3, img, img, device="cuda")
+    while done < n_images:
+        b = min(batch, n_images - done)
+        xb = x[:b]
+        out, emb = net(xb, return_features="finalembed")
+        logits_l.append(out.float())
+        emb_l.append(emb.float())
+        done += b
+    return torch.cat(logits_l), torch.cat(emb_l)
+
+
+def config2_entropy_margin(steps_pool=50_000):
+    from active_learning_amd.models import get_networks
+    from active_learning_amd.ops.scoring import softmax_scores
+    net = get_networks("synthetic_cifar10", "SSLResNet18").cuda().eval()
+    res = {}
+    (logits, _), res["inference_s"] = timed(
+        "cifar pool inference (50k)", lambda: pool_inference(net, steps_pool, 1024, 32))
+    (_, res["scoring_s"]) = timed(
+        "fused entropy+margin scoring", lambda: softmax_scores(logits))
+    res["total_s"] = res["inference_s"] + res["scoring_s"]
+    return res
+
+
+def config4_badge(n_labeled=50_000, n_unlabeled=80_000, budget=10_000):
+    from active_learning_amd.models import get_networks
+    from active_learning_amd.ops.scoring import (badge_pairwise_sqdist, badge_vectors,
+                                                 kcenter_greedy)
+    net = get_networks("synthetic_imagenet", "SSLResNet50").cuda().eval()
+    n = n_labeled + n_unlabeled
+    res = {}
+    (out, res["inference_s"]) = timed(
+        f"imagenet pool inference ({n//1000}k)",
+        lambda: pool_inference(net, n, 256, 224))
+    logits, emb = out
+    (ae, res["gram_s"]) = timed(
+        "BADGE factorized Gram (130k x 130k)",
+        lambda: badge_pairwise_sqdist(*badge_vectors(logits, emb)))
+    labeled = torch.zeros(n, dtype=torch.bool, device="cuda")
+    labeled[:n_labeled] = True
+    (_, res["kcenter_s"]) = timed(
+        f"k-means++ seeding (b={budget})",
+        lambda: kcenter_greedy(ae, labeled, budget, randomize=True))
+    res["total_s"] = sum(v for k, v in res.items() if k.endswith("_s"))
+    return res
+
+
+def config5_coreset(n_labeled=50_000, n_unlabeled=80_000, budget=10_000):
+    from active_learning_amd.ops.scoring import kcenter_greedy, pairwise_sqdist
+    n = n_labeled + n_unlabeled
+    emb = torch.randn(n, 2048, device="cuda")
+    res = {}
+    (dist, res["pairwise_s"]) = timed(
+        f"pairwise sqdist ({n//1000}k x {n//1000}k = "
+        f"{n*n*4/1e9:.0f} GB resident)", lambda: pairwise_sqdist(emb))
+    labeled = torch.zeros(n, dtype=torch.bool, device="cuda")
+    labeled[:n_labeled] = True
+    (sel, res["kcenter_s"]) = timed(
+        f"greedy k-center (b={budget})",
+        lambda: kcenter_greedy(dist, labeled, budget, randomize=False))
+    assert len(sel) == budget
+    res["total_s"] = res["pairwise_s"] + res["kcenter_s"]
+    res["matrix_gb"] = round(n * n * 4 / 1e9, 1)
+    return res
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--configs", default="2,4,5")
+    ap.add_argument("--scale", type=float, default=1.0,
+                    help="pool-size scale factor for quick runs")
+    args = ap.parse_args()
+    s = args.scale
+    results = {}
+    if "2" in args.configs:
+        print("config 2: ResNet-18/CIFAR entropy+margin")
+        results["config2_entropy_margin"] = config2_entropy_margin(int(50_000 * s))
+    if "4" in args.configs:
+        print("config 4: BADGE grad-embed + k-means++")
+        results["config4_badge"] = config4_badge(int(50_000 * s), int(80_000 * s),
+                                                 int(10_000 * s))
+    if "5" in args.configs:
+        print("config 5: Coreset greedy k-center, HBM-resident N^2")
+        results["config5_coreset"] = config5_coreset(int(50_000 * s), int(80_000 * s),
+                                                     int(10_000 * s))
+    print(json.dumps(results))
+
+
+if __name__ == "__main__":
+    main()
