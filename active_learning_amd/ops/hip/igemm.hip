@@ -246,6 +246,9 @@ __global__ void igemm_kernel(const bf16* __restrict__ A, const bf16* __restrict_
   stage(0);
   int buf = 0;
   for (int kt = 0; kt < KT; ++kt) {
+    // own-queue drain of tile kt, then one barrier; the NEXT tile's glds
+    // (issued below) stay in flight across the whole compute phase — no
+    // bottom barrier, so hipcc never emits a mid-loop vmcnt(0) drain.
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __syncthreads();
     if (kt + 1 < KT) {
@@ -273,8 +276,7 @@ __global__ void igemm_kernel(const bf16* __restrict__ A, const bf16* __restrict_
           acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               afrag[mi], bfrag[ni], acc[mi][ni], 0, 0, 0);
     }
-    __syncthreads();
-    buf ^= 1;
+    buf ^= 1;  // next iteration's top barrier orders buffer reuse
   }
 
   // epilogue: C/D layout row = (l>>4)*4 + r, col = l&15

@@ -180,10 +180,10 @@ __global__ void wgrad_kernel(const bf16* __restrict__ dy, const bf16* __restrict
   write_tile(0, va, vb);
   int buf = 0;
   for (long m0 = lz0; m0 < lz1; m0 += BL) {
-    __syncthreads();
+    __syncthreads();  // write(buf) visible to all readers of this tile
     if (m0 + BL < lz1) load_tile(m0 + BL, va, vb);  // overlap with MFMAs below
     compute(buf);
-    __syncthreads();
+    // no barrier: write targets buf^1, last read two phases ago
     if (m0 + BL < lz1) write_tile(buf ^ 1, va, vb);
     buf ^= 1;
   }
