@@ -176,12 +176,15 @@ class BatchNormAct(Function):
                 xf = x.float()
                 s = xf.sum(dim=(0, 1, 2))
                 ss = (xf * xf).sum(dim=(0, 1, 2))
-            count = torch.tensor([float(n_local)], device=x.device)
             if pg is not None and dist.is_initialized() and dist.get_world_size(pg) > 1:
+                count = torch.tensor([float(n_local)], device=x.device)
                 packed = torch.cat([s, ss, count])
                 dist.all_reduce(packed, group=pg)
-                s, ss, count = packed[:len(s)], packed[len(s):2 * len(s)], packed[-1:]
-            n = float(count.item())
+                s, ss = packed[:len(s)], packed[len(s):2 * len(s)]
+                n = float(packed[-1].item())
+            else:
+                # no device round-trip (a .item() here would sync every BN)
+                n = float(n_local)
             mean = s / n
             var = (ss / n - mean * mean).clamp_min_(0)
             invstd = (var + eps).rsqrt()
