@@ -72,22 +72,36 @@ def _igemm_eligible(C, KD):
     return C % 8 == 0 and KD % 64 == 0
 
 
+def _wpad_cached(w_c, kdpad):
+    """Zero-padded (K,1,1,kdpad) copy of the stem weight, cached on the bf16
+    weight (rebuilt only when the bf16 copy is recreated)."""
+    wpad = getattr(w_c, "_al_wpad", None)
+    K = w_c.shape[0]
+    KD = w_c.numel() // K
+    if wpad is None:
+        wpad = torch.zeros(K, 1, 1, kdpad, dtype=w_c.dtype, device=w_c.device)
+        try:
+            w_c._al_wpad = wpad
+        except Exception:
+            pass
+    wpad.view(K, kdpad)[:, :KD] = w_c.reshape(K, KD)
+    return wpad
+
+
 def _gpu_conv_fwd_packed(ext, x, w_c, stride, padding):
-    """Stem path: A = im2col(x) zero-padded to kdpad cols, conv as 1x1 igemm."""
+    """Stem path: A = im2col(x) zero-padded to kdpad cols, conv as 1x1 igemm.
+    Returns (y, apack); apack is reused by the backward wgrad."""
     K, R, S, C = w_c.shape
     KD = R * S * C
     kdpad = ((KD + 63) // 64) * 64
     apack = ext.im2col_pack(x, R, S, stride, padding, kdpad)
-    wpad = torch.zeros(K, 1, 1, kdpad, dtype=w_c.dtype, device=w_c.device)
-    wpad.view(K, kdpad)[:, :KD] = w_c.reshape(K, KD)
-    return ext.conv2d_fwd(apack, wpad, 1, 0)
+    return ext.conv2d_fwd(apack, _wpad_cached(w_c, kdpad), 1, 0), apack
 
 
-def _gpu_conv_wgrad_packed(ext, dy, x, w_shape, stride, padding):
+def _gpu_conv_wgrad_packed(ext, dy, apack, w_shape):
     K, R, S, C = w_shape
     KD = R * S * C
-    kdpad = ((KD + 63) // 64) * 64
-    apack = ext.im2col_pack(x, R, S, stride, padding, kdpad)
+    kdpad = apack.shape[-1]
     dwpad = ext.conv2d_bwd_weight(dy, apack, 1, 1, 1, 0)  # (K,1,1,kdpad) fp32
     return dwpad.view(K, kdpad)[:, :KD].reshape(K, R, S, C).contiguous()
 
@@ -107,8 +121,11 @@ class Conv2dNHWC(Function):
             ext = require_extension()
             K, R, S, C = w_c.shape
             if _igemm_eligible(C, R * S * C):
+                ctx.apack = None
                 return ext.conv2d_fwd(x, w_c, stride, padding)
-            return _gpu_conv_fwd_packed(ext, x, w_c, stride, padding)
+            y, apack = _gpu_conv_fwd_packed(ext, x, w_c, stride, padding)
+            ctx.apack = apack
+            return y
         xc, wc = _cpu_conv_args(x, w_c)
         y = F.conv2d(xc.float(), wc.float(), stride=stride, padding=padding)
         return y.to(x.dtype).permute(0, 2, 3, 1).contiguous()
@@ -129,8 +146,7 @@ class Conv2dNHWC(Function):
                 if _igemm_eligible(C, R * S * C):
                     dw = ext.conv2d_bwd_weight(dy, x, R, S, ctx.stride, ctx.padding)
                 else:
-                    dw = _gpu_conv_wgrad_packed(ext, dy, x, w_c.shape, ctx.stride,
-                                                ctx.padding)
+                    dw = _gpu_conv_wgrad_packed(ext, dy, ctx.apack, w_c.shape)
         else:
             xc, wc = _cpu_conv_args(x, w_c)
             dyc = dy.permute(0, 3, 1, 2).float()
