@@ -44,12 +44,14 @@ class BasicBlock(nn.Module):
 class Bottleneck(nn.Module):
     expansion = 4
 
-    def __init__(self, in_planes, planes, stride=1, downsample=None):
+    def __init__(self, in_planes, planes, stride=1, downsample=None, v1=False):
         super().__init__()
-        # v1.5: stride on the 3x3 (torchvision default)
-        self.conv1 = Conv2dNHWC(in_planes, planes, 1, 1, 0)
+        # v1.5 (torchvision default): stride on the 3x3; v1 (resnet_hacks.py:36-46
+        # stride swap): stride on the first 1x1 instead
+        s1, s2 = (stride, 1) if v1 else (1, stride)
+        self.conv1 = Conv2dNHWC(in_planes, planes, 1, s1, 0)
         self.bn1 = BatchNormAct2d(planes, relu=True)
-        self.conv2 = Conv2dNHWC(planes, planes, 3, stride, 1)
+        self.conv2 = Conv2dNHWC(planes, planes, 3, s2, 1)
         self.bn2 = BatchNormAct2d(planes, relu=True)
         self.conv3 = Conv2dNHWC(planes, planes * self.expansion, 1, 1, 0)
         self.bn3 = BatchNormAct2d(planes * self.expansion, relu=True)
@@ -76,10 +78,12 @@ class Downsample(nn.Module):
 
 
 class ResNetEncoder(nn.Module):
-    def __init__(self, block, layers, cifar_stem=False, compute_dtype=torch.bfloat16):
+    def __init__(self, block, layers, cifar_stem=False, compute_dtype=torch.bfloat16,
+                 v1=False):
         super().__init__()
         self.compute_dtype = compute_dtype
         self.cifar_stem = cifar_stem
+        self.v1 = v1
         self.in_planes = 64
         if cifar_stem:
             self.conv1 = Conv2dNHWC(3, 64, 3, 1, 1)
@@ -96,7 +100,8 @@ class ResNetEncoder(nn.Module):
         downsample = None
         if stride != 1 or self.in_planes != planes * block.expansion:
             downsample = Downsample(self.in_planes, planes * block.expansion, stride)
-        blocks = [block(self.in_planes, planes, stride, downsample)]
+        kw = {"v1": self.v1} if block is Bottleneck else {}
+        blocks = [block(self.in_planes, planes, stride, downsample, **kw)]
         self.in_planes = planes * block.expansion
         for _ in range(1, num_blocks):
             blocks.append(block(self.in_planes, planes))
@@ -127,3 +132,21 @@ def resnet18_encoder(cifar_stem=False, **kw):
 
 def resnet50_encoder(cifar_stem=False, **kw):
     return ResNetEncoder(Bottleneck, [3, 4, 6, 3], cifar_stem=cifar_stem, **kw)
+
+
+class NormedLinear(nn.Module):
+    """Cosine-normalized linear head.
+
+    Parity with src/models/resnet_hacks.py:50-59 (defined there but unused
+    anywhere in the reference; provided for capability completeness).
+    """
+
+    def __init__(self, in_features, out_features):
+        super().__init__()
+        self.weight = nn.Parameter(torch.empty(in_features, out_features))
+        with torch.no_grad():
+            self.weight.uniform_(-1, 1).renorm_(2, 1, 1e-5).mul_(1e5)
+
+    def forward(self, x):
+        import torch.nn.functional as F
+        return F.normalize(x.float(), dim=1) @ F.normalize(self.weight, dim=0)
