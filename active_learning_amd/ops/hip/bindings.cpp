@@ -34,9 +34,9 @@ void al_bn_bwd(const void* dy, const void* x, const void* y, const float* mean,
                const float* invstd, const float* gamma, const float* sum_dy,
                const float* sum_dy_xhat, float n, int use_batch_stats, int relu,
                int has_res, void* dx, void* dres, long rows, int C, void* stream);
-void al_maxpool_fwd(const void* x, void* y, long* idx, int N, int H, int W, int C,
+void al_maxpool_fwd(const void* x, void* y, int* idx, int N, int H, int W, int C,
                     int P, int Q, int kernel, int stride, int pad, void* stream);
-void al_maxpool_bwd(const void* dy, const long* idx, void* dx, int N, int H, int W,
+void al_maxpool_bwd(const void* dy, const int* idx, void* dx, int N, int H, int W,
                     int C, int P, int Q, int kernel, int stride, int pad, void* stream);
 void al_global_avg_pool(const void* x, void* y, int N, int HW, int C, void* stream);
 void al_softmax_scores(const float* logits, float* out, int B, int C, void* stream);
@@ -219,8 +219,8 @@ std::vector<Tensor> maxpool2d_fwd(const Tensor& x, long kernel, long stride,
   const int N = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
   const int P = out_dim(H, kernel, stride, pad), Q = out_dim(W, kernel, stride, pad);
   auto y = torch::empty({N, P, Q, C}, x.options());
-  auto idx = torch::empty({N, P, Q, C}, x.options().dtype(torch::kInt64));
-  al_maxpool_fwd(x.data_ptr(), y.data_ptr(), idx.data_ptr<long>(), N, H, W, C, P, Q,
+  auto idx = torch::empty({N, P, Q, C}, x.options().dtype(torch::kInt32));
+  al_maxpool_fwd(x.data_ptr(), y.data_ptr(), idx.data_ptr<int>(), N, H, W, C, P, Q,
                  (int)kernel, (int)stride, (int)pad, cur_stream());
   return {y, idx};
 }
@@ -230,7 +230,7 @@ Tensor maxpool2d_bwd(const Tensor& dy, const Tensor& idx, long H, long W, long k
   check_bf16_contig(dy, "dy");
   const int N = dy.size(0), P = dy.size(1), Q = dy.size(2), C = dy.size(3);
   auto dx = torch::empty({N, (long)H, (long)W, C}, dy.options());
-  al_maxpool_bwd(dy.data_ptr(), idx.data_ptr<long>(), dx.data_ptr(), N, (int)H,
+  al_maxpool_bwd(dy.data_ptr(), idx.data_ptr<int>(), dx.data_ptr(), N, (int)H,
                  (int)W, C, P, Q, (int)kernel, (int)stride, (int)pad, cur_stream());
   return dx;
 }

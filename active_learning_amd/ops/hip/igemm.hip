@@ -101,12 +101,15 @@ AL_DEV const bf16* b_chunk_ptr(const bf16* __restrict__ b, const bf16* zero,
 // the tiled kernel
 // ---------------------------------------------------------------------------
 
-template <int MODE>
+// GWR x GWC: wave grid (4 waves). (2,2) -> 128x128 tile; (4,1) -> 256x64 for
+// narrow-Nout layers (K=64) where half a 128-wide tile would idle.
+template <int MODE, int GWR, int GWC>
 __launch_bounds__(256)
 __global__ void igemm_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
                              bf16* __restrict__ out, const bf16* __restrict__ zero,
                              ConvShape sh, int grid_m) {
-  constexpr int BM = 128, BN = 128, BK = 64;
+  constexpr int BM = GWR * 64, BN = GWC * 64, BK = 64;
+  constexpr int NA = 2 * GWR, NB = 2 * GWC;  // 16B chunks per thread/operand
   // XCD-aware block remap (T1): contiguous output tiles on one XCD share B
   // panels in its L2. Bijective variant for any grid size.
   const int nwg = gridDim.x;
@@ -128,7 +131,7 @@ __global__ void igemm_kernel(const bf16* __restrict__ A, const bf16* __restrict_
 
   const int tid = threadIdx.x;
   const int wid = tid >> 6, lane = tid & 63;
-  const int wr = wid >> 1, wc = wid & 1;     // 2x2 wave grid, 64x64 each
+  const int wr = wid / GWC, wc = wid % GWC;  // wave grid, 64x64 tiles each
 
   const int KT = sh.KD / BK;
 
@@ -139,16 +142,16 @@ __global__ void igemm_kernel(const bf16* __restrict__ A, const bf16* __restrict_
   // coordinate advances (+BK per tile) with cheap carry propagation — no
   // divisions inside the loop.
   // -------------------------------------------------------------------
-  int a_r[4], a_s[4], a_cf[4];          // filter pos + fast-channel coord
-  long a_pix[4];                        // (n, outer-spatial) base index
-  int a_p[4], a_q[4];                   // per-row spatial (fwd: p,q; bwd: h,w)
-  bool a_ok[4];
-  const bf16* b_ptr[4];
+  int a_r[NA], a_s[NA], a_cf[NA];       // filter pos + fast-channel coord
+  long a_pix[NA];                       // (n, outer-spatial) base index
+  int a_p[NA], a_q[NA];                 // per-row spatial (fwd: p,q; bwd: h,w)
+  bool a_ok[NA];
+  const bf16* b_ptr[NB];
   const int fastC = (MODE == MODE_FWD) ? sh.C : sh.K;
 
 #pragma unroll
-  for (int i = 0; i < 4; ++i) {
-    const int t = (wid * 4 + i) * 64 + lane;
+  for (int i = 0; i < NA; ++i) {
+    const int t = (wid * NA + i) * 64 + lane;
     const int row = t >> 3, u = t & 7;
     const int usw = u ^ (row & 7);
     const int kd0 = usw * 8;
@@ -181,9 +184,15 @@ __global__ void igemm_kernel(const bf16* __restrict__ A, const bf16* __restrict_
     } else {
       a_pix[i] = 0; a_p[i] = 0; a_q[i] = 0;
     }
+  }
+#pragma unroll
+  for (int i = 0; i < NB; ++i) {
+    const int t = (wid * NB + i) * 64 + lane;
+    const int row = t >> 3, u = t & 7;
+    const int usw = u ^ (row & 7);
     // B: row n0 + row fixed, contraction contiguous
     const int j = n0 + row;
-    b_ptr[i] = (j < sh.Nout) ? (B + (long)j * sh.KD + kd0) : nullptr;
+    b_ptr[i] = (j < sh.Nout) ? (B + (long)j * sh.KD + usw * 8) : nullptr;
   }
 
   auto a_src = [&](int i) -> const bf16* {
@@ -206,31 +215,33 @@ __global__ void igemm_kernel(const bf16* __restrict__ A, const bf16* __restrict_
 
   auto advance = [&]() {
 #pragma unroll
-    for (int i = 0; i < 4; ++i) {
+    for (int i = 0; i < NA; ++i) {
       a_cf[i] += BK;
       while (a_cf[i] >= fastC) {
         a_cf[i] -= fastC;
         if (++a_s[i] == sh.S) { a_s[i] = 0; ++a_r[i]; }
       }
-      if (b_ptr[i]) b_ptr[i] += BK;
     }
+#pragma unroll
+    for (int i = 0; i < NB; ++i)
+      if (b_ptr[i]) b_ptr[i] += BK;
   };
 
   auto stage = [&](int buf) {
     bf16* abase = As + buf * BM * BK;
     bf16* bbase = Bs + buf * BN * BK;
 #pragma unroll
-    for (int i = 0; i < 4; ++i) {
+    for (int i = 0; i < NA; ++i) {
       const bf16* src = a_src(i);
       __builtin_amdgcn_global_load_lds((const __attribute__((address_space(1))) void*)src,
-                                       (__attribute__((address_space(3))) void*)(abase + (wid * 4 + i) * 512),
+                                       (__attribute__((address_space(3))) void*)(abase + (wid * NA + i) * 512),
                                        16, 0, 0);
     }
 #pragma unroll
-    for (int i = 0; i < 4; ++i) {
+    for (int i = 0; i < NB; ++i) {
       const bf16* src = b_ptr[i] ? b_ptr[i] : zero;
       __builtin_amdgcn_global_load_lds((const __attribute__((address_space(1))) void*)src,
-                                       (__attribute__((address_space(3))) void*)(bbase + (wid * 4 + i) * 512),
+                                       (__attribute__((address_space(3))) void*)(bbase + (wid * NB + i) * 512),
                                        16, 0, 0);
     }
   };
@@ -347,18 +358,20 @@ extern "C" void al_conv2d_mm(int mode, const void* A, const void* B, void* out,
     sh.KD = R * S * K;
   }
   if (igemm_ok(mode, sh)) {
-    const int grid_m = (int)((sh.M + 127) / 128);
-    const int grid_n = (sh.Nout + 127) / 128;
-    const size_t lds = 2 * (128 * 64 + 128 * 64) * sizeof(bf16);  // 64 KiB
+    // narrow-Nout layers (K=64) use a 256x64 tile so no wave idles
+    const bool narrow = sh.Nout <= 64;
+    const int BM = narrow ? 256 : 128, BN = narrow ? 64 : 128;
+    const int grid_m = (int)((sh.M + BM - 1) / BM);
+    const int grid_n = (sh.Nout + BN - 1) / BN;
+    const size_t lds = 2 * (size_t)(BM + BN) * 64 * sizeof(bf16);
     dim3 grid(grid_m * grid_n), block(256);
-    if (mode == MODE_FWD)
-      hipLaunchKernelGGL((igemm_kernel<MODE_FWD>), grid, block, lds, stream,
-                         (const bf16*)A, (const bf16*)B, (bf16*)out,
-                         (const bf16*)zero_page, sh, grid_m);
-    else
-      hipLaunchKernelGGL((igemm_kernel<MODE_BWD_DATA>), grid, block, lds, stream,
-                         (const bf16*)A, (const bf16*)B, (bf16*)out,
-                         (const bf16*)zero_page, sh, grid_m);
+#define LAUNCH(MODE_, GWR_, GWC_)     hipLaunchKernelGGL((igemm_kernel<MODE_, GWR_, GWC_>), grid, block, lds, stream,                        (const bf16*)A, (const bf16*)B, (bf16*)out,                        (const bf16*)zero_page, sh, grid_m)
+    if (mode == MODE_FWD) {
+      if (narrow) LAUNCH(MODE_FWD, 4, 1); else LAUNCH(MODE_FWD, 2, 2);
+    } else {
+      if (narrow) LAUNCH(MODE_BWD_DATA, 4, 1); else LAUNCH(MODE_BWD_DATA, 2, 2);
+    }
+#undef LAUNCH
   } else {
     long total = sh.M * sh.Nout;
     int blocks = (int)min((total + 255) / 256, (long)8192);

@@ -9,7 +9,7 @@
 // ---------------------------------------------------------------------------
 
 __global__ void maxpool_fwd_kernel(const bf16* __restrict__ x, bf16* __restrict__ y,
-                                   long* __restrict__ idx, int N, int H, int W, int C,
+                                   int* __restrict__ idx, int N, int H, int W, int C,
                                    int P, int Q, int kernel, int stride, int pad) {
   const long total = (long)N * P * Q * C;
   for (long i = grid_stride_begin(); i < total; i += grid_stride_step()) {
@@ -37,7 +37,7 @@ __global__ void maxpool_fwd_kernel(const bf16* __restrict__ x, bf16* __restrict_
   }
 }
 
-extern "C" void al_maxpool_fwd(const void* x, void* y, long* idx, int N, int H, int W,
+extern "C" void al_maxpool_fwd(const void* x, void* y, int* idx, int N, int H, int W,
                                int C, int P, int Q, int kernel, int stride, int pad,
                                hipStream_t stream) {
   long total = (long)N * P * Q * C;
@@ -50,7 +50,7 @@ extern "C" void al_maxpool_fwd(const void* x, void* y, long* idx, int N, int H, 
 // backward: per INPUT pixel, sum dy over covering windows whose argmax == me
 // (deterministic, no atomics; each input is covered by <= ceil(k/s)^2 windows)
 __global__ void maxpool_bwd_kernel(const bf16* __restrict__ dy,
-                                   const long* __restrict__ idx,
+                                   const int* __restrict__ idx,
                                    bf16* __restrict__ dx, int N, int H, int W, int C,
                                    int P, int Q, int kernel, int stride, int pad) {
   const long total = (long)N * H * W * C;
@@ -76,7 +76,7 @@ __global__ void maxpool_bwd_kernel(const bf16* __restrict__ dy,
   }
 }
 
-extern "C" void al_maxpool_bwd(const void* dy, const long* idx, void* dx, int N, int H,
+extern "C" void al_maxpool_bwd(const void* dy, const int* idx, void* dx, int N, int H,
                                int W, int C, int P, int Q, int kernel, int stride,
                                int pad, hipStream_t stream) {
   long total = (long)N * H * W * C;

@@ -91,11 +91,15 @@ AL_DEV void xpose8x8(u32 d[4], int lane) {
   }
 }
 
+// GKR x GNC wave grid (4 waves, 64x64 tiles): (2,2) -> 128(k) x 128(rsc);
+// (1,4) -> 64 x 256 for K=64 layers.
+template <int GKR, int GNC>
 __launch_bounds__(256)
 __global__ void wgrad_kernel(const bf16* __restrict__ dy, const bf16* __restrict__ x,
                              float* __restrict__ dw, WgradShape sh, int grid_k,
                              long l_per_z) {
-  constexpr int BMK = 128, BNW = 128, BL = 64;
+  constexpr int BMK = GKR * 64, BNW = GNC * 64, BL = 64;
+  constexpr int NAB = 2 * GKR, NBB = 2 * GNC, HA = NAB / 2, HB = NBB / 2;
   const int bk = blockIdx.x % grid_k;
   const int bn = blockIdx.x / grid_k;
   const int k0 = bk * BMK;
@@ -111,7 +115,7 @@ __global__ void wgrad_kernel(const bf16* __restrict__ dy, const bf16* __restrict
 
   const int tid = threadIdx.x;
   const int wid = tid >> 6, lane = tid & 63;
-  const int wr = wid >> 1, wc = wid & 1;
+  const int wr = wid / GNC, wc = wid % GNC;
   const int l15 = lane & 15, l4 = lane >> 4;
   const int u = lane >> 3;       // channel chunk (8 ch)
   const int mo = lane & 7;       // m offset within the wave's 8-m slice
@@ -122,32 +126,42 @@ __global__ void wgrad_kernel(const bf16* __restrict__ dy, const bf16* __restrict
 #pragma unroll
     for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  // A (dY) and B (im2col x) each: 4 sub-batches of (8-m slice x 8-ch chunk)
-  // cover 64 m x 16 chunks (128 channels).
-  auto load_tile = [&](long m0, s16x8 va[4], s16x8 vb[4]) {
+  // sub-batches of (8-m slice x 8-ch chunk): A covers 64 m x BMK channels
+  // in NAB batches, B covers 64 m x BNW channels in NBB batches.
+  auto load_tile = [&](long m0, s16x8 va[NAB], s16x8 vb[NBB]) {
 #pragma unroll
-    for (int i = 0; i < 4; ++i) {
-      const long m = m0 + wid * 8 + 32 * (i >> 1) + mo;
-      const int uc = u + 8 * (i & 1);
+    for (int i = 0; i < NAB; ++i) {
+      const long m = m0 + wid * 8 + 32 * (i / HA) + mo;
+      const int uc = u + 8 * (i % HA);
       va[i] = s16x8{0, 0, 0, 0, 0, 0, 0, 0};
+      if (m < lz1 && k0 + uc * 8 < sh.K)
+        va[i] = *(const s16x8*)(dy + m * sh.K + k0 + uc * 8);
+    }
+#pragma unroll
+    for (int i = 0; i < NBB; ++i) {
+      const long m = m0 + wid * 8 + 32 * (i / HB) + mo;
+      const int uc = u + 8 * (i % HB);
       vb[i] = s16x8{0, 0, 0, 0, 0, 0, 0, 0};
-      if (m < lz1) {
-        if (k0 + uc * 8 < sh.K)
-          va[i] = *(const s16x8*)(dy + m * sh.K + k0 + uc * 8);
-        const bf16* src;
-        if (x_chunk(x, sh, m, n0 + uc * 8, &src)) vb[i] = *(const s16x8*)src;
-      }
+      const bf16* src;
+      if (m < lz1 && x_chunk(x, sh, m, n0 + uc * 8, &src))
+        vb[i] = *(const s16x8*)src;
     }
   };
 
-  auto write_tile = [&](int buf, s16x8 va[4], s16x8 vb[4]) {
+  auto write_tile = [&](int buf, s16x8 va[NAB], s16x8 vb[NBB]) {
 #pragma unroll
-    for (int i = 0; i < 4; ++i) {
-      const int slot = wid + 4 * (i >> 1);             // m-slot (16B = 8 m)
-      const int ch = (u + 8 * (i & 1)) * 8 + mo;       // ch row
+    for (int i = 0; i < NAB; ++i) {
+      const int slot = wid + 4 * (i / HA);             // m-slot (16B = 8 m)
+      const int ch = (u + 8 * (i % HA)) * 8 + mo;      // ch row
       const int sw = slot ^ (ch & 7);
       xpose8x8((u32*)&va[i], lane);
       *(s16x8*)(&At[buf][ch][sw * 8]) = va[i];
+    }
+#pragma unroll
+    for (int i = 0; i < NBB; ++i) {
+      const int slot = wid + 4 * (i / HB);
+      const int ch = (u + 8 * (i % HB)) * 8 + mo;
+      const int sw = slot ^ (ch & 7);
       xpose8x8((u32*)&vb[i], lane);
       *(s16x8*)(&Bt[buf][ch][sw * 8]) = vb[i];
     }
@@ -175,7 +189,7 @@ __global__ void wgrad_kernel(const bf16* __restrict__ dy, const bf16* __restrict
     }
   };
 
-  s16x8 va[4], vb[4];
+  s16x8 va[NAB], vb[NBB];
   load_tile(lz0, va, vb);
   write_tile(0, va, vb);
   int buf = 0;
@@ -237,8 +251,10 @@ extern "C" void al_conv2d_wgrad(const void* dy, const void* x, float* dw, int N,
   sh.L = (long)N * P * Q;
   sh.Nw = R * S * C;
   if (C % 8 == 0 && K % 8 == 0) {
-    const int grid_k = (K + 127) / 128;
-    const int grid_n = (sh.Nw + 127) / 128;
+    const bool narrow_k = K <= 64;
+    const int BMK = narrow_k ? 64 : 128, BNW = narrow_k ? 256 : 128;
+    const int grid_k = (K + BMK - 1) / BMK;
+    const int grid_n = (sh.Nw + BNW - 1) / BNW;
     const int tiles = grid_k * grid_n;
     // split-K: aim for >= 512 blocks to fill 256 CUs
     int z = (int)min((long)128, max((long)1, (512L + tiles - 1) / tiles));
@@ -247,8 +263,12 @@ extern "C" void al_conv2d_wgrad(const void* dy, const void* x, float* dw, int N,
     l_per_z = ((l_per_z + 63) / 64) * 64;
     z = (int)((sh.L + l_per_z - 1) / l_per_z);
     dim3 grid(tiles, z), block(256);
-    hipLaunchKernelGGL(wgrad_kernel, grid, block, 0, stream, (const bf16*)dy,
-                       (const bf16*)x, dw, sh, grid_k, l_per_z);
+    if (narrow_k)
+      hipLaunchKernelGGL((wgrad_kernel<1, 4>), grid, block, 0, stream,
+                         (const bf16*)dy, (const bf16*)x, dw, sh, grid_k, l_per_z);
+    else
+      hipLaunchKernelGGL((wgrad_kernel<2, 2>), grid, block, 0, stream,
+                         (const bf16*)dy, (const bf16*)x, dw, sh, grid_k, l_per_z);
   } else {
     long total = (long)K * sh.Nw;
     int bx = (int)min((total + 255) / 256, (long)1024);
