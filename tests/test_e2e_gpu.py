@@ -61,3 +61,33 @@ def test_scoring_pool_gpu():
     d_g = badge_pairwise_sqdist(a, e)
     d_c = badge_pairwise_sqdist(a.cpu(), e.cpu())
     assert torch.allclose(d_g.cpu(), d_c, rtol=1e-3, atol=1e-2)
+
+
+def test_vaal_round_gpu(tmp_path):
+    """VAAL on GPU: VAE (native conv + convT kernels) + discriminator + Adam
+    co-training and discriminator-scored query."""
+    from active_learning_amd.cli import get_args
+    from active_learning_amd.main_al import main
+    args = get_args([
+        "--dataset", "synthetic_cifar10", "--rounds", "2", "--round_budget", "6",
+        "--n_epoch", "1", "--early_stop_patience", "1", "--debug_mode",
+        "--ckpt_path", str(tmp_path / "ckpt"), "--log_dir", str(tmp_path / "logs"),
+        "--model", "SSLResNet18", "--strategy", "VAALSampler",
+        "--vae_latent_dim", "8"])
+    s = main(args)
+    assert s.idxs_lb.sum() == 11  # 5 init + 6 queried
+
+
+def test_freeze_feature_round_gpu(tmp_path):
+    """Linear-eval mode on GPU: frozen backbone (detached embedding), BN in
+    frozen-stats mode, only the head trains."""
+    from active_learning_amd.cli import get_args
+    from active_learning_amd.main_al import main
+    args = get_args([
+        "--dataset", "synthetic_cifar10", "--rounds", "1", "--round_budget", "6",
+        "--n_epoch", "2", "--early_stop_patience", "2", "--debug_mode",
+        "--ckpt_path", str(tmp_path / "ckpt"), "--log_dir", str(tmp_path / "logs"),
+        "--model", "SSLResNet18", "--strategy", "RandomSampler",
+        "--freeze_feature"])
+    s = main(args)
+    assert s.idxs_lb.sum() == 5
