@@ -140,8 +140,17 @@ class Conv2dNHWC(Function):
             ext = require_extension()
             K, R, S, C = w_c.shape
             if ctx.needs_input_grad[0]:
-                dx = ext.conv2d_bwd_data(dy, wt_cached(w_c), ctx.stride, ctx.padding,
-                                         x.shape[1], x.shape[2])
+                if R == 1 and S == 1 and ctx.stride > 1 and ctx.padding == 0:
+                    # 1x1 strided conv (ResNet downsample): dx is nonzero only
+                    # at stride-aligned pixels — dense GEMM on dY + strided
+                    # scatter instead of a 4x-redundant gather igemm
+                    tmp = ext.conv2d_fwd(dy, wt_cached(w_c), 1, 0)  # (N,P,Q,C)
+                    n_, h_, w_ = x.shape[0], x.shape[1], x.shape[2]
+                    dx = torch.zeros(n_, h_, w_, C, dtype=dy.dtype, device=dy.device)
+                    dx[:, ::ctx.stride, ::ctx.stride, :] = tmp
+                else:
+                    dx = ext.conv2d_bwd_data(dy, wt_cached(w_c), ctx.stride,
+                                             ctx.padding, x.shape[1], x.shape[2])
             if ctx.needs_input_grad[1]:
                 if _igemm_eligible(C, R * S * C):
                     dw = ext.conv2d_bwd_weight(dy, x, R, S, ctx.stride, ctx.padding)
