@@ -193,31 +193,42 @@ class BatchNormAct(Function):
     def forward(ctx, x, weight, bias, running_mean, running_var, use_batch_stats,
                 momentum, eps, relu, residual, pg):
         n_local = x.numel() // x.shape[-1]
+        sync = (pg is not None and dist.is_initialized()
+                and dist.get_world_size(pg) > 1)
         if use_batch_stats:
-            if x.is_cuda:
+            if x.is_cuda and not sync:
+                # fused path: partial sums -> mean/invstd + running update in
+                # two kernels, no small ATen ops, no host sync
                 ext = require_extension()
-                s, ss = ext.bn_stats(x)  # fp32 per-channel sum / sum of squares
-            else:
-                xf = x.float()
-                s = xf.sum(dim=(0, 1, 2))
-                ss = (xf * xf).sum(dim=(0, 1, 2))
-            if pg is not None and dist.is_initialized() and dist.get_world_size(pg) > 1:
-                count = torch.tensor([float(n_local)], device=x.device)
-                packed = torch.cat([s, ss, count])
-                dist.all_reduce(packed, group=pg)
-                s, ss = packed[:len(s)], packed[len(s):2 * len(s)]
-                n = float(packed[-1].item())
-            else:
-                # no device round-trip (a .item() here would sync every BN)
+                with torch.no_grad():
+                    mean, invstd = ext.bn_stats_finalize(
+                        x, running_mean, running_var, momentum, eps,
+                        running_mean is not None)
                 n = float(n_local)
-            mean = s / n
-            var = (ss / n - mean * mean).clamp_min_(0)
-            invstd = (var + eps).rsqrt()
-            with torch.no_grad():
-                if running_mean is not None:
-                    unbiased = var * (n / max(n - 1, 1))
-                    running_mean.mul_(1 - momentum).add_(mean, alpha=momentum)
-                    running_var.mul_(1 - momentum).add_(unbiased, alpha=momentum)
+            else:
+                if x.is_cuda:
+                    ext = require_extension()
+                    s, ss = ext.bn_stats(x)  # fp32 per-channel sums
+                else:
+                    xf = x.float()
+                    s = xf.sum(dim=(0, 1, 2))
+                    ss = (xf * xf).sum(dim=(0, 1, 2))
+                if sync:
+                    count = torch.tensor([float(n_local)], device=x.device)
+                    packed = torch.cat([s, ss, count])
+                    dist.all_reduce(packed, group=pg)
+                    s, ss = packed[:len(s)], packed[len(s):2 * len(s)]
+                    n = float(packed[-1].item())
+                else:
+                    n = float(n_local)
+                mean = s / n
+                var = (ss / n - mean * mean).clamp_min_(0)
+                invstd = (var + eps).rsqrt()
+                with torch.no_grad():
+                    if running_mean is not None:
+                        unbiased = var * (n / max(n - 1, 1))
+                        running_mean.mul_(1 - momentum).add_(mean, alpha=momentum)
+                        running_var.mul_(1 - momentum).add_(unbiased, alpha=momentum)
         else:
             mean = running_mean.to(torch.float32)
             invstd = (running_var.to(torch.float32) + eps).rsqrt()
@@ -259,14 +270,17 @@ class BatchNormAct(Function):
             sum_dy = dyf.sum(dim=(0, 1, 2))
             sum_dy_xhat = (dyf * xhat).sum(dim=(0, 1, 2))
 
-        dgamma = sum_dy_xhat.clone()
-        dbeta = sum_dy.clone()
-
         if ctx.use_batch_stats and ctx.pg is not None and dist.is_initialized() \
                 and dist.get_world_size(ctx.pg) > 1:
+            dgamma = sum_dy_xhat.clone()
+            dbeta = sum_dy.clone()
             packed = torch.cat([sum_dy, sum_dy_xhat])
             dist.all_reduce(packed, group=ctx.pg)
             sum_dy, sum_dy_xhat = packed[:len(sum_dy)], packed[len(sum_dy):]
+        else:
+            # no aliasing hazard without the all-reduce: the sums ARE the grads
+            dgamma = sum_dy_xhat
+            dbeta = sum_dy
         # ctx.n is already the GLOBAL element count when stats were synced
         # (the forward all-reduced the counts), else the local count.
         n_global = ctx.n

@@ -24,6 +24,10 @@ void al_bn_stats(const void* x, float* sum, float* sumsq, long rows, int C,
 int al_bn_reduce_blocks(long rows, int C);
 void al_bn_part_reduce(const float* part_a, const float* part_b, float* out_a,
                        float* out_b, int nb, int C, void* stream);
+void al_bn_finalize(const float* part_s, const float* part_ss, float* mean,
+                    float* invstd, float* running_mean, float* running_var, int nb,
+                    int C, float n, float momentum, float eps, int update_running,
+                    void* stream);
 void al_bn_norm_fwd(const void* x, void* y, const float* mean, const float* invstd,
                     const float* gamma, const float* beta, const void* res, int relu,
                     long rows, int C, void* stream);
@@ -128,6 +132,30 @@ Tensor im2col_pack(const Tensor& x, long R, long S, long stride, long pad,
   al_im2col_pack(x.data_ptr(), out.data_ptr(), N, H, W, C, (int)R, (int)S, P, Q,
                  (int)stride, (int)pad, (int)kdpad, cur_stream());
   return out;
+}
+
+std::vector<Tensor> bn_stats_finalize(const Tensor& x, Tensor& running_mean,
+                                      Tensor& running_var, double momentum,
+                                      double eps, bool update_running) {
+  // single-process fast path: partial sums -> mean/invstd (+ running update)
+  // without any host-side scalar math or small ATen kernels
+  check_bf16_contig(x, "x");
+  const int C = x.size(-1);
+  const long rows = x.numel() / C;
+  auto opts = x.options().dtype(torch::kFloat32);
+  const int nb = al_bn_reduce_blocks(rows, C);
+  auto part_sum = torch::empty({nb, C}, opts);
+  auto part_sumsq = torch::empty({nb, C}, opts);
+  al_bn_stats(x.data_ptr(), part_sum.data_ptr<float>(), part_sumsq.data_ptr<float>(),
+              rows, C, cur_stream());
+  auto mean = torch::empty({C}, opts);
+  auto invstd = torch::empty({C}, opts);
+  al_bn_finalize(part_sum.data_ptr<float>(), part_sumsq.data_ptr<float>(),
+                 mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                 running_mean.data_ptr<float>(), running_var.data_ptr<float>(), nb, C,
+                 (float)rows, (float)momentum, (float)eps, update_running ? 1 : 0,
+                 cur_stream());
+  return {mean, invstd};
 }
 
 std::vector<Tensor> bn_stats(const Tensor& x) {
@@ -306,6 +334,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv2d_bwd_data", &conv2d_bwd_data);
   m.def("conv2d_bwd_weight", &conv2d_bwd_weight);
   m.def("bn_stats", &bn_stats);
+  m.def("bn_stats_finalize", &bn_stats_finalize);
   m.def("bn_norm_fwd", &bn_norm_fwd);
   m.def("bn_bwd_reduce", &bn_bwd_reduce);
   m.def("bn_bwd", &bn_bwd);

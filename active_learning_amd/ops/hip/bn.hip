@@ -343,3 +343,60 @@ extern "C" void al_bn_part_reduce(const float* part_a, const float* part_b,
   hipLaunchKernelGGL(bn_part_reduce_kernel, dim3((C + 7) / 8), dim3(256), 0,
                      stream, part_a, part_b, out_a, out_b, nb, C);
 }
+
+// ---------------------------------------------------------------------------
+// finalize: partials -> mean/invstd (+ in-place running-stat update), fusing
+// what was ~6 tiny ATen ops per BN layer into the stage-2 reduce.
+// ---------------------------------------------------------------------------
+
+__global__ void bn_finalize_kernel(const float* __restrict__ part_s,
+                                   const float* __restrict__ part_ss,
+                                   float* __restrict__ mean,
+                                   float* __restrict__ invstd,
+                                   float* __restrict__ running_mean,
+                                   float* __restrict__ running_var, int nb, int C,
+                                   float inv_n, float unbias, float momentum,
+                                   float eps, int update_running) {
+  const int cl = threadIdx.x & 7;
+  const int bl = threadIdx.x >> 3;
+  const int c = blockIdx.x * 8 + cl;
+  __shared__ float red[2][256];
+  float sa = 0.f, sb = 0.f;
+  if (c < C) {
+    for (int b = bl; b < nb; b += 32) {
+      sa += part_s[(long)b * C + c];
+      sb += part_ss[(long)b * C + c];
+    }
+  }
+  red[0][threadIdx.x] = sa;
+  red[1][threadIdx.x] = sb;
+  __syncthreads();
+  for (int step = 128; step >= 8; step >>= 1) {
+    if (threadIdx.x < step) {
+      red[0][threadIdx.x] += red[0][threadIdx.x + step];
+      red[1][threadIdx.x] += red[1][threadIdx.x + step];
+    }
+    __syncthreads();
+  }
+  if (threadIdx.x < 8 && c < C) {
+    const float m = red[0][threadIdx.x] * inv_n;
+    float var = red[1][threadIdx.x] * inv_n - m * m;
+    var = var < 0.f ? 0.f : var;
+    mean[c] = m;
+    invstd[c] = rsqrtf(var + eps);
+    if (update_running) {
+      running_mean[c] = running_mean[c] * (1.f - momentum) + m * momentum;
+      running_var[c] = running_var[c] * (1.f - momentum) + var * unbias * momentum;
+    }
+  }
+}
+
+extern "C" void al_bn_finalize(const float* part_s, const float* part_ss, float* mean,
+                               float* invstd, float* running_mean, float* running_var,
+                               int nb, int C, float n, float momentum, float eps,
+                               int update_running, hipStream_t stream) {
+  const float unbias = n / (n - 1.f > 0.f ? n - 1.f : 1.f);
+  hipLaunchKernelGGL(bn_finalize_kernel, dim3((C + 7) / 8), dim3(256), 0, stream,
+                     part_s, part_ss, mean, invstd, running_mean, running_var, nb, C,
+                     1.f / n, unbias, momentum, eps, update_running);
+}
