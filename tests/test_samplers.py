@@ -158,3 +158,52 @@ def test_margin_clustering_consumes_clusters():
     assert cost == 6 and len(set(idxs)) == 6
     # assignment persists with consumed removed
     assert s.cluster_assignment is not None
+
+
+def test_coreset_cache_under_freeze_feature():
+    """freeze_feature + no subsetting caches the distance matrix across
+    rounds (coreset_sampler.py:112-121)."""
+    s = make_strategy(get_strategy("CoresetSampler"), freeze_feature=True)
+    s.update(np.arange(8), 8)
+    s.query(4)
+    assert s.saved_pairwise_l2_dist is not None
+    first = s.saved_pairwise_l2_dist
+    idxs, _ = s.query(4)
+    assert s.saved_pairwise_l2_dist is first  # reused, not recomputed
+
+
+def test_coreset_subset_caps():
+    s = make_strategy(get_strategy("CoresetSampler"), subset_labeled=5,
+                      subset_unlabeled=10)
+    s.update(np.arange(12), 12)
+    idxs_all, lab, unlab = s.get_idxs_for_coreset(return_sep_idxs=True)
+    assert len(lab) == 5
+    # reference semantics: unlabeled cap = subset_labeled + subset_unlabeled
+    # - len(labeled-after-cap) (coreset_sampler.py:28-34)
+    assert len(unlab) == 10
+    idxs, cost = s.query(6)
+    assert cost == 6 and all(not s.idxs_lb[i] for i in idxs)
+
+
+def test_partitioned_coreset_covers_partitions():
+    s = make_strategy(get_strategy("PartitionedCoresetSampler"), partitions=3)
+    s.update(np.arange(9), 9)
+    idxs, cost = s.query(9)
+    assert cost == 9 and len(set(idxs)) == 9
+
+
+def test_balancing_sampler_cache():
+    s = make_strategy(get_strategy("BalancingSampler"), freeze_feature=True)
+    s.update(np.arange(6), 6)
+    s.query(3)
+    assert s.saved_embeddings is not None
+
+
+def test_margin_clustering_assignment_persists():
+    s = make_strategy(get_strategy("MarginClusteringSampler"))
+    s.update(np.arange(5), 5)
+    n_before = len(s.available_query_idxs(shuffle=False))
+    idxs, _ = s.query(4)
+    s.update(idxs, 4)
+    # persisted assignment shrank by the consumed samples
+    assert len(s.cluster_assignment) == n_before - 4
