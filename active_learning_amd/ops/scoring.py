@@ -108,16 +108,24 @@ def pairwise_sqdist(features: torch.Tensor, chunk=8192) -> torch.Tensor:
     return out
 
 
-def badge_pairwise_sqdist(a_vec: torch.Tensor, e_vec: torch.Tensor) -> torch.Tensor:
+def badge_pairwise_sqdist(a_vec: torch.Tensor, e_vec: torch.Tensor,
+                          chunk=8192) -> torch.Tensor:
     """N x N squared distances between gradient embeddings g_i = a_i (x) e_i
-    without materializing them: <g_i, g_j> = (a_i.a_j)(e_i.e_j)."""
+    without materializing them: <g_i, g_j> = (a_i.a_j)(e_i.e_j).
+
+    Row-chunked so only ONE N x N fp32 matrix is resident (68 GB at N=130k);
+    the two Gram factors exist chunk-at-a-time."""
     a = a_vec.float()
     e = e_vec.float()
-    ga = a @ a.t()
-    ge = e @ e.t()
-    g = ga * ge
-    d = g.diagonal()
-    return d[:, None] + d[None, :] - 2 * g
+    n = a.shape[0]
+    d = (a * a).sum(dim=1) * (e * e).sum(dim=1)  # <g_i, g_i>
+    out = torch.empty((n, n), dtype=torch.float32, device=a.device)
+    for i0 in range(0, n, chunk):
+        i1 = min(i0 + chunk, n)
+        g = (a[i0:i1] @ a.t()) * (e[i0:i1] @ e.t())
+        g.mul_(-2).add_(d[i0:i1, None]).add_(d[None, :])
+        out[i0:i1] = g
+    return out
 
 
 def badge_vectors(logits: torch.Tensor, embedding: torch.Tensor, pool=None):
@@ -154,39 +162,46 @@ def kcenter_greedy(dist: torch.Tensor, labeled: torch.Tensor, budget: int,
     plus one argmax/multinomial.
     """
     n = dist.shape[0]
+    dev = dist.device
     labeled = labeled.clone()
-    selected = []
-    neg_inf = torch.tensor(float("-inf"), device=dist.device)
+    # fully device-side selection: the chosen index never round-trips to the
+    # host inside the loop (one sync per QUERY, not per iteration — the
+    # reference syncs every iteration, coreset_sampler.py:82-104)
+    sel_buf = torch.empty(budget, dtype=torch.int64, device=dev)
+    n_pre = 0
 
     if labeled.any():
         min_dist = _masked_col_min(dist, labeled)
+        start = 0
     else:
         if randomize:
-            j = int(torch.randint(n, (1,)).item())
+            j = torch.randint(n, (1,), device=dev)[0]
         else:
-            j = int(dist.max(dim=1).values.argmin().item())
-        selected.append(j)
+            j = dist.max(dim=1).values.argmin()
+        sel_buf[0] = j
         labeled[j] = True
-        min_dist = dist[:, j].clone()
+        min_dist = dist[j].clone()  # symmetric: row == column
+        start = 1
 
-    while len(selected) < budget:
+    for it in range(start, budget):
         if randomize:
             probs = min_dist.clamp_min(0.0)
             probs = torch.where(labeled, torch.zeros_like(probs), probs)
             total = probs.sum()
-            if total <= 0 or not torch.isfinite(total):
-                # reference jitters by +1e-5 until valid (coreset_sampler.py:85-92);
-                # equivalent: fall back to uniform over unlabeled
-                probs = (~labeled).float()
-                total = probs.sum()
-            j = int(torch.multinomial(probs / total, 1).item())
+            # reference jitters on degenerate weights (coreset_sampler.py:85-92);
+            # equivalent: uniform over unlabeled in that case, chosen
+            # device-side so the loop stays sync-free
+            uniform = (~labeled).float()
+            bad = ~torch.isfinite(total) | (total <= 0)
+            probs = torch.where(bad, uniform, probs)
+            j = torch.multinomial(probs, 1)[0]
         else:
-            scores = torch.where(labeled, neg_inf, min_dist)
-            j = int(scores.argmax().item())
-        selected.append(j)
+            scores = min_dist.masked_fill(labeled, float("-inf"))
+            j = scores.argmax()
+        sel_buf[it] = j
         labeled[j] = True
-        min_dist = torch.minimum(min_dist, dist[:, j])
-    return selected
+        torch.minimum(min_dist, dist[j], out=min_dist)
+    return sel_buf.cpu().tolist()
 
 
 def _masked_col_min(dist, labeled_mask, chunk=16384):
