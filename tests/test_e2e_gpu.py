@@ -60,7 +60,10 @@ def test_scoring_pool_gpu():
     a, e = badge_vectors(logits, emb)
     d_g = badge_pairwise_sqdist(a, e)
     d_c = badge_pairwise_sqdist(a.cpu(), e.cpu())
-    assert torch.allclose(d_g.cpu(), d_c, rtol=1e-3, atol=1e-2)
+    # fp32 Gram distances cancel catastrophically relative to the ~2k-scale
+    # Gram entries; compare at matrix-norm precision (GPU vs CPU GEMM order)
+    err = (d_g.cpu() - d_c).norm() / d_c.norm().clamp_min(1e-6)
+    assert err < 1e-4, f"badge gram relerr {err}"
 
 
 def test_vaal_round_gpu(tmp_path):
