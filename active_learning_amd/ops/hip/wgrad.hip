@@ -17,6 +17,7 @@
 // ops/functional.py, so the direct fallback below is a safety net only.
 
 #include "al_common.h"
+#include <stdlib.h>
 
 typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
 
@@ -256,8 +257,18 @@ extern "C" void al_conv2d_wgrad(const void* dy, const void* x, float* dw, int N,
     const int grid_k = (K + BMK - 1) / BMK;
     const int grid_n = (sh.Nw + BNW - 1) / BNW;
     const int tiles = grid_k * grid_n;
-    // split-K: aim for >= 512 blocks to fill 256 CUs
-    int z = (int)min((long)128, max((long)1, (512L + tiles - 1) / tiles));
+    // split-K: aim for >= 512 blocks to fill 256 CUs (tunable via env)
+    static int zcap = -1;
+    static long ztarget = -1;
+    if (zcap < 0) {
+      const char* e = getenv("AL_WGRAD_ZCAP");
+      zcap = e ? atoi(e) : 128;
+      const char* t = getenv("AL_WGRAD_ZTARGET");
+      ztarget = t ? atol(t) : 512;
+    }
+    long zt = ztarget, zc = zcap;
+    if (tiles <= 4) { zt = 2 * ztarget; zc = 4 * zcap; }  // tiny-grid layers
+    int z = (int)min(zc, max((long)1, (zt + tiles - 1) / tiles));
     z = (int)min((long)z, max((long)1, sh.L / 64));
     long l_per_z = (sh.L + z - 1) / z;
     l_per_z = ((l_per_z + 63) / 64) * 64;
