@@ -34,6 +34,9 @@ typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
 
 #define MODE_FWD 0
 #define MODE_BWD_DATA 1
+#define MODE_BWD_S2 2    // stride-2 bwd-data, one (h%2, w%2) parity class per
+                         // launch: contraction runs only the VALID (r,s) of
+                         // the class, eliminating the 4x zero-chunk redundancy
 
 struct ConvShape {
   int N, H, W, C;       // input (or dx) spatial shape
@@ -44,6 +47,14 @@ struct ConvShape {
   long M;               // GEMM rows: N*P*Q (fwd) or N*H*W (bwd)
   int Nout;             // GEMM cols: K (fwd) or C (bwd)
   int KD;               // contraction: R*S*C (fwd) or R*S*K (bwd)
+  // MODE_BWD_S2 only (H/W hold the class spatial dims Hc/Wc; R/S hold the
+  // class filter counts Rc/Sc):
+  int Horig, Worig;     // full dx spatial dims
+  int hcl, wcl;         // parity class
+  int poff, qoff;       // p = (hh + poff) - ri, q = (ww + qoff) - si
+  long b_rowstride;     // Rorig*Sorig*K (Wt row pitch)
+  long b_rstride;       // 2*Sorig*K    (ri step within a Wt row)
+  long b_coff;          // (rpar*Sorig + spar)*K (class base inside a row)
 };
 
 // ---------------------------------------------------------------------------
@@ -147,6 +158,8 @@ __global__ void igemm_kernel(const bf16* __restrict__ A, const bf16* __restrict_
   int a_p[NA], a_q[NA];                 // per-row spatial (fwd: p,q; bwd: h,w)
   bool a_ok[NA];
   const bf16* b_ptr[NB];
+  int b_r[NB], b_s[NB], b_k[NB];        // MODE_BWD_S2: per-slot Wt coords
+  long b_base[NB];
   const int fastC = (MODE == MODE_FWD) ? sh.C : sh.K;
 
 #pragma unroll
@@ -172,7 +185,7 @@ __global__ void igemm_kernel(const bf16* __restrict__ A, const bf16* __restrict_
         a_pix[i] = (long)n * sh.H * sh.W;
         a_p[i] = p * sh.stride - sh.pad;
         a_q[i] = q * sh.stride - sh.pad;
-      } else {
+      } else if (MODE == MODE_BWD_DATA) {
         const int w = (int)(m % sh.W);
         long tt = m / sh.W;
         const int h = (int)(tt % sh.H);
@@ -180,6 +193,14 @@ __global__ void igemm_kernel(const bf16* __restrict__ A, const bf16* __restrict_
         a_pix[i] = (long)n * sh.P * sh.Q;
         a_p[i] = h + sh.pad;
         a_q[i] = w + sh.pad;
+      } else {  // MODE_BWD_S2: m -> (n, hh, ww) within the parity class
+        const int ww = (int)(m % sh.W);
+        long tt = m / sh.W;
+        const int hh = (int)(tt % sh.H);
+        const int n = (int)(tt / sh.H);
+        a_pix[i] = (long)n * sh.P * sh.Q;
+        a_p[i] = hh + sh.poff;
+        a_q[i] = ww + sh.qoff;
       }
     } else {
       a_pix[i] = 0; a_p[i] = 0; a_q[i] = 0;
@@ -190,9 +211,16 @@ __global__ void igemm_kernel(const bf16* __restrict__ A, const bf16* __restrict_
     const int t = (wid * NB + i) * 64 + lane;
     const int row = t >> 3, u = t & 7;
     const int usw = u ^ (row & 7);
-    // B: row n0 + row fixed, contraction contiguous
+    // B: row n0 + row fixed; contraction contiguous except MODE_BWD_S2,
+    // where only the class's (r,s) of Wt participate (strided, stateful)
     const int j = n0 + row;
-    b_ptr[i] = (j < sh.Nout) ? (B + (long)j * sh.KD + usw * 8) : nullptr;
+    if (MODE == MODE_BWD_S2) {
+      b_base[i] = (j < sh.Nout) ? (long)j * sh.b_rowstride : -1;
+      b_r[i] = 0; b_s[i] = 0; b_k[i] = usw * 8;  // kd0 < 64 <= K
+      b_ptr[i] = nullptr;
+    } else {
+      b_ptr[i] = (j < sh.Nout) ? (B + (long)j * sh.KD + usw * 8) : nullptr;
+    }
   }
 
   auto a_src = [&](int i) -> const bf16* {
@@ -202,7 +230,7 @@ __global__ void igemm_kernel(const bf16* __restrict__ A, const bf16* __restrict_
       const int w = a_q[i] + a_s[i];
       if ((unsigned)h >= (unsigned)sh.H || (unsigned)w >= (unsigned)sh.W) return zero;
       return A + (a_pix[i] + (long)h * sh.W + w) * sh.C + a_cf[i];
-    } else {
+    } else if (MODE == MODE_BWD_DATA) {
       const int hp = a_p[i] - a_r[i];
       const int wp = a_q[i] - a_s[i];
       if (hp < 0 || wp < 0) return zero;
@@ -210,7 +238,19 @@ __global__ void igemm_kernel(const bf16* __restrict__ A, const bf16* __restrict_
       const int p = hp / sh.stride, q = wp / sh.stride;
       if (p >= sh.P || q >= sh.Q) return zero;
       return A + (a_pix[i] + (long)p * sh.Q + q) * sh.K + a_cf[i];
+    } else {  // MODE_BWD_S2: every (ri, si) of the class divides exactly
+      const int p = a_p[i] - a_r[i];
+      const int q = a_q[i] - a_s[i];
+      if ((unsigned)p >= (unsigned)sh.P || (unsigned)q >= (unsigned)sh.Q) return zero;
+      return A + (a_pix[i] + (long)p * sh.Q + q) * sh.K + a_cf[i];
     }
+  };
+
+  auto b_src = [&](int i) -> const bf16* {
+    if (MODE != MODE_BWD_S2) return b_ptr[i] ? b_ptr[i] : zero;
+    if (b_base[i] < 0) return zero;
+    return B + b_base[i] + sh.b_coff + (long)b_r[i] * sh.b_rstride
+           + (long)b_s[i] * 2 * sh.K + b_k[i];
   };
 
   auto advance = [&]() {
@@ -223,8 +263,17 @@ __global__ void igemm_kernel(const bf16* __restrict__ A, const bf16* __restrict_
       }
     }
 #pragma unroll
-    for (int i = 0; i < NB; ++i)
-      if (b_ptr[i]) b_ptr[i] += BK;
+    for (int i = 0; i < NB; ++i) {
+      if (MODE == MODE_BWD_S2) {
+        b_k[i] += BK;
+        while (b_k[i] >= sh.K) {
+          b_k[i] -= sh.K;
+          if (++b_s[i] == sh.S) { b_s[i] = 0; ++b_r[i]; }
+        }
+      } else if (b_ptr[i]) {
+        b_ptr[i] += BK;
+      }
+    }
   };
 
   auto stage = [&](int buf) {
@@ -239,7 +288,7 @@ __global__ void igemm_kernel(const bf16* __restrict__ A, const bf16* __restrict_
     }
 #pragma unroll
     for (int i = 0; i < NB; ++i) {
-      const bf16* src = b_ptr[i] ? b_ptr[i] : zero;
+      const bf16* src = b_src(i);
       __builtin_amdgcn_global_load_lds((const __attribute__((address_space(1))) void*)src,
                                        (__attribute__((address_space(3))) void*)(bbase + (wid * NB + i) * 512),
                                        16, 0, 0);
@@ -290,7 +339,8 @@ __global__ void igemm_kernel(const bf16* __restrict__ A, const bf16* __restrict_
     buf ^= 1;  // next iteration's top barrier orders buffer reuse
   }
 
-  // epilogue: C/D layout row = (l>>4)*4 + r, col = l&15
+  // epilogue: C/D layout row = (l>>4)*4 + r, col = l&15. For MODE_BWD_S2 the
+  // GEMM row is a class-local pixel; scatter to the strided dx location.
 #pragma unroll
   for (int mi = 0; mi < 4; ++mi) {
 #pragma unroll
@@ -300,8 +350,20 @@ __global__ void igemm_kernel(const bf16* __restrict__ A, const bf16* __restrict_
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const long row = m0 + wr * 64 + mi * 16 + l4 * 4 + r;
-        if (row < sh.M)
-          out[row * sh.Nout + col] = f2bf(acc[mi][ni][r]);
+        if (row < sh.M) {
+          long off;
+          if (MODE == MODE_BWD_S2) {
+            const int ww = (int)(row % sh.W);
+            long tt = row / sh.W;
+            const int hh = (int)(tt % sh.H);
+            const long n = tt / sh.H;
+            off = ((n * sh.Horig + 2 * hh + sh.hcl) * sh.Worig + 2 * ww + sh.wcl)
+                      * sh.Nout + col;
+          } else {
+            off = row * sh.Nout + col;
+          }
+          out[off] = f2bf(acc[mi][ni][r]);
+        }
       }
     }
   }
@@ -357,6 +419,53 @@ extern "C" void al_conv2d_mm(int mode, const void* A, const void* B, void* out,
     sh.Nout = C;
     sh.KD = R * S * K;
   }
+
+  // stride-2 bwd-data: run the 4 parity classes as dense gather GEMMs
+  // (each contracts only its valid (r,s): no zero-chunk redundancy)
+  if (mode == MODE_BWD_DATA && stride == 2 && (K % 64) == 0 && R * S > 1) {
+    bool classes_ok = true;
+    for (int hc = 0; hc < 2 && classes_ok; ++hc)
+      for (int wc = 0; wc < 2 && classes_ok; ++wc) {
+        const int rpar = (hc + pad) & 1, spar = (wc + pad) & 1;
+        if ((R - rpar + 1) / 2 <= 0 || (S - spar + 1) / 2 <= 0) classes_ok = false;
+      }
+    if (classes_ok) {
+      for (int hc = 0; hc < 2; ++hc)
+        for (int wc = 0; wc < 2; ++wc) {
+          ConvShape c = sh;
+          const int rpar = (hc + pad) & 1, spar = (wc + pad) & 1;
+          const int Rc = (R - rpar + 1) / 2, Sc = (S - spar + 1) / 2;
+          const int Hc = (H - hc + 1) / 2, Wc = (W - wc + 1) / 2;
+          if (Hc <= 0 || Wc <= 0) continue;
+          c.H = Hc; c.W = Wc; c.R = Rc; c.S = Sc;
+          c.M = (long)N * Hc * Wc;
+          c.KD = Rc * Sc * K;
+          c.Horig = H; c.Worig = W;
+          c.hcl = hc; c.wcl = wc;
+          c.poff = (hc + pad - rpar) / 2;
+          c.qoff = (wc + pad - spar) / 2;
+          c.b_rowstride = (long)R * S * K;
+          c.b_rstride = 2L * S * K;
+          c.b_coff = ((long)rpar * S + spar) * K;
+          const bool narrow = c.Nout <= 64;
+          const int BM = narrow ? 256 : 128, BN = narrow ? 64 : 128;
+          const int gm = (int)((c.M + BM - 1) / BM);
+          const int gn = (c.Nout + BN - 1) / BN;
+          const size_t lds = 2 * (size_t)(BM + BN) * 64 * sizeof(bf16);
+          dim3 grid(gm * gn), block(256);
+          if (narrow)
+            hipLaunchKernelGGL((igemm_kernel<MODE_BWD_S2, 4, 1>), grid, block, lds,
+                               stream, (const bf16*)A, (const bf16*)B, (bf16*)out,
+                               (const bf16*)zero_page, c, gm);
+          else
+            hipLaunchKernelGGL((igemm_kernel<MODE_BWD_S2, 2, 2>), grid, block, lds,
+                               stream, (const bf16*)A, (const bf16*)B, (bf16*)out,
+                               (const bf16*)zero_page, c, gm);
+        }
+      return;
+    }
+  }
+
   if (igemm_ok(mode, sh)) {
     // narrow-Nout layers (K=64) use a 256x64 tile so no wave idles
     const bool narrow = sh.Nout <= 64;
