@@ -92,19 +92,43 @@ extern "C" void al_maxpool_bwd(const void* dy, const int* idx, void* dx, int N, 
 // ---------------------------------------------------------------------------
 
 __global__ void gap_kernel(const bf16* __restrict__ x, bf16* __restrict__ y, int HW,
-                           int C) {
+                           int C, int cg_per_block) {
+  // thread -> (c8 chunk, row lane): 16B vector loads, coalesced across lanes
   const int n = blockIdx.x;
+  const int C8 = C / 8;
+  const int cg = blockIdx.y * cg_per_block + (threadIdx.x % cg_per_block);
+  const int row_lane = threadIdx.x / cg_per_block;
+  const int rows_per_block = 256 / cg_per_block;
+  if (cg >= C8) return;
   const float inv = 1.0f / HW;
-  for (int c = threadIdx.x; c < C; c += blockDim.x) {
-    float acc = 0.f;
-    const bf16* base = x + (long)n * HW * C + c;
-    for (int r = 0; r < HW; ++r) acc += bf2f(base[(long)r * C]);
-    y[(long)n * C + c] = f2bf(acc * inv);
+  float acc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  const s16x8* base = (const s16x8*)x + (long)n * HW * C8 + cg;
+  for (int r = row_lane; r < HW; r += rows_per_block) {
+    s16x8 v = base[(long)r * C8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) acc[j] += bits2f(v[j]);
+  }
+  __shared__ float red[256][8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) red[threadIdx.x][j] = acc[j];
+  __syncthreads();
+  if (row_lane == 0) {
+    for (int rl = 1; rl < rows_per_block; ++rl)
+#pragma unroll
+      for (int j = 0; j < 8; ++j) acc[j] += red[rl * cg_per_block + (threadIdx.x % cg_per_block)][j];
+    s16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) o[j] = f2bits(acc[j] * inv);
+    ((s16x8*)y)[(long)n * C8 + cg] = o;
   }
 }
 
 extern "C" void al_global_avg_pool(const void* x, void* y, int N, int HW, int C,
                                    hipStream_t stream) {
-  hipLaunchKernelGGL(gap_kernel, dim3(N), dim3(256), 0, stream, (const bf16*)x,
-                     (bf16*)y, HW, C);
+  int c8 = C / 8;
+  int cg = 1;
+  while (cg < 32 && cg * 2 <= c8 && (c8 % (cg * 2)) == 0) cg *= 2;
+  dim3 grid(N, (c8 + cg - 1) / cg);
+  hipLaunchKernelGGL(gap_kernel, grid, dim3(256), 0, stream, (const bf16*)x,
+                     (bf16*)y, HW, C, cg);
 }
