@@ -20,6 +20,7 @@ import torch
 import torch.nn as nn
 
 from ..ops import functional as AF
+from ..ops.fused import conv_bn_act
 from .layers import BatchNormAct2d, Conv2dNHWC, nchw_to_nhwc
 
 
@@ -36,8 +37,8 @@ class BasicBlock(nn.Module):
 
     def forward(self, x):
         identity = x if self.downsample is None else self.downsample(x)
-        out = self.bn1(self.conv1(x))
-        out = self.bn2(self.conv2(out), residual=identity)
+        out = conv_bn_act(self.conv1, self.bn1, x)
+        out = conv_bn_act(self.conv2, self.bn2, out, residual=identity)
         return out
 
 
@@ -59,9 +60,9 @@ class Bottleneck(nn.Module):
 
     def forward(self, x):
         identity = x if self.downsample is None else self.downsample(x)
-        out = self.bn1(self.conv1(x))
-        out = self.bn2(self.conv2(out))
-        out = self.bn3(self.conv3(out), residual=identity)
+        out = conv_bn_act(self.conv1, self.bn1, x)
+        out = conv_bn_act(self.conv2, self.bn2, out)
+        out = conv_bn_act(self.conv3, self.bn3, out, residual=identity)
         return out
 
 
@@ -74,7 +75,7 @@ class Downsample(nn.Module):
         self.bn = BatchNormAct2d(out_planes, relu=False)
 
     def forward(self, x):
-        return self.bn(self.conv(x))
+        return conv_bn_act(self.conv, self.bn, x)
 
 
 class ResNetEncoder(nn.Module):
@@ -115,7 +116,7 @@ class ResNetEncoder(nn.Module):
         x = nchw_to_nhwc(x)
         if x.is_cuda and self.compute_dtype is not None:
             x = x.to(self.compute_dtype)
-        x = self.bn1(self.conv1(x))
+        x = conv_bn_act(self.conv1, self.bn1, x)
         if not self.cifar_stem:
             x = AF.max_pool2d(x, 3, 2, 1)
         x = self.layer1(x)

@@ -1,0 +1,49 @@
+"""Fused inference ops: conv + frozen-stats BN (+ residual + ReLU) in one
+kernel pass.
+
+Used on the query/eval paths (always under torch.no_grad with eval-mode BN):
+folds scale = gamma/sqrt(var+eps), shift = beta - mean*scale into the conv
+epilogue, removing one full read+write of every conv output (53 BN passes per
+ResNet-50 forward). Training-mode or grad-enabled calls fall back to the
+separate autograd ops.
+"""
+
+import torch
+
+from .extension import extension_available, load_extension
+from .functional import _igemm_eligible, _wpad_cached, cast_cached
+
+
+def _bn_fold_cached(bn):
+    key = (bn.weight._version, bn.bias._version, bn.running_mean._version,
+           bn.running_var._version)
+    cache = getattr(bn, "_al_fold", None)
+    if cache is not None and cache[0] == key:
+        return cache[1], cache[2]
+    invstd = (bn.running_var.float() + bn.eps).rsqrt()
+    scale = (bn.weight.float() * invstd).contiguous()
+    shift = (bn.bias.float() - bn.running_mean.float() * scale).contiguous()
+    try:
+        bn._al_fold = (key, scale, shift)
+    except Exception:
+        pass
+    return scale, shift
+
+
+def conv_bn_act(conv, bn, x, residual=None):
+    """bn(conv(x), residual) with the fused single-kernel path when eligible."""
+    if (x.is_cuda and not bn.training and not torch.is_grad_enabled()
+            and extension_available()):
+        ext = load_extension()
+        w_c = cast_cached(conv.weight, x.dtype)
+        K, R, S, C = w_c.shape
+        scale, shift = _bn_fold_cached(bn)
+        res = residual if residual is not None else x.new_empty(0)
+        if _igemm_eligible(C, R * S * C):
+            return ext.conv2d_fwd_fused(x, w_c, conv.stride, conv.padding,
+                                        scale, shift, bn.relu, res)
+        kdpad = ((R * S * C + 63) // 64) * 64
+        apack = ext.im2col_pack(x, R, S, conv.stride, conv.padding, kdpad)
+        return ext.conv2d_fwd_fused(apack, _wpad_cached(w_c, kdpad), 1, 0,
+                                    scale, shift, bn.relu, res)
+    return bn(conv(x), residual=residual)

@@ -15,7 +15,9 @@ using torch::Tensor;
 extern "C" {
 void al_conv2d_mm(int mode, const void* A, const void* B, void* out,
                   const void* zero_page, int N, int H, int W, int C, int K, int R,
-                  int S, int P, int Q, int stride, int pad, void* stream);
+                  int S, int P, int Q, int stride, int pad, const float* epi_scale,
+                  const float* epi_shift, const void* epi_res, int epi_relu,
+                  void* stream);
 void al_conv2d_wgrad(const void* dy, const void* x, float* dw, int N, int H, int W,
                      int C, int K, int R, int S, int P, int Q, int stride, int pad,
                      void* stream);
@@ -90,7 +92,30 @@ Tensor conv2d_fwd(const Tensor& x, const Tensor& w, long stride, long pad) {
   auto y = torch::empty({N, P, Q, K}, x.options());
   al_conv2d_mm(0, x.data_ptr(), w.data_ptr(), y.data_ptr(),
                zero_page(x).data_ptr(), N, H, W, C, K, R, S, P, Q, (int)stride,
-               (int)pad, cur_stream());
+               (int)pad, nullptr, nullptr, nullptr, 0, cur_stream());
+  return y;
+}
+
+Tensor conv2d_fwd_fused(const Tensor& x, const Tensor& w, long stride, long pad,
+                        const Tensor& scale, const Tensor& shift, bool relu,
+                        const Tensor& residual) {
+  // inference path: y = [relu](conv(x,w) * scale + shift [+ residual]) —
+  // frozen-stats BN folded into the conv epilogue (query/eval passes)
+  check_bf16_contig(x, "x");
+  check_bf16_contig(w, "w");
+  const int N = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
+  const int K = w.size(0), R = w.size(1), S = w.size(2);
+  TORCH_CHECK(w.size(3) == C, "conv2d_fwd_fused: channel mismatch");
+  TORCH_CHECK(scale.numel() == K && shift.numel() == K);
+  const int P = out_dim(H, R, stride, pad), Q = out_dim(W, S, stride, pad);
+  auto y = torch::empty({N, P, Q, K}, x.options());
+  const bool has_res = residual.numel() > 0;
+  if (has_res) check_bf16_contig(residual, "residual");
+  al_conv2d_mm(0, x.data_ptr(), w.data_ptr(), y.data_ptr(),
+               zero_page(x).data_ptr(), N, H, W, C, K, R, S, P, Q, (int)stride,
+               (int)pad, scale.contiguous().data_ptr<float>(),
+               shift.contiguous().data_ptr<float>(),
+               has_res ? residual.data_ptr() : nullptr, relu ? 1 : 0, cur_stream());
   return y;
 }
 
@@ -106,7 +131,7 @@ Tensor conv2d_bwd_data(const Tensor& dy, const Tensor& wt, long stride, long pad
   auto dx = torch::empty({N, (long)H, (long)W, C}, dy.options());
   al_conv2d_mm(1, dy.data_ptr(), wt.data_ptr(), dx.data_ptr(),
                zero_page(dy).data_ptr(), N, (int)H, (int)W, C, K, R, S, P, Q,
-               (int)stride, (int)pad, cur_stream());
+               (int)stride, (int)pad, nullptr, nullptr, nullptr, 0, cur_stream());
   return dx;
 }
 
@@ -331,6 +356,7 @@ void adam_step(Tensor& p, const Tensor& g, Tensor& m, Tensor& v, double lr, doub
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv2d_fwd", &conv2d_fwd);
+  m.def("conv2d_fwd_fused", &conv2d_fwd_fused);
   m.def("conv2d_bwd_data", &conv2d_bwd_data);
   m.def("conv2d_bwd_weight", &conv2d_bwd_weight);
   m.def("bn_stats", &bn_stats);

@@ -114,11 +114,17 @@ AL_DEV const bf16* b_chunk_ptr(const bf16* __restrict__ b, const bf16* zero,
 
 // GWR x GWC: wave grid (4 waves). (2,2) -> 128x128 tile; (4,1) -> 256x64 for
 // narrow-Nout layers (K=64) where half a 128-wide tile would idle.
+// epi_scale/epi_shift: optional fused inference epilogue
+//   y = [relu]( acc * scale[col] + shift[col] [+ residual] )
+// (frozen-stats BN folded into the conv — query/eval path)
 template <int MODE, int GWR, int GWC>
 __launch_bounds__(256)
 __global__ void igemm_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
                              bf16* __restrict__ out, const bf16* __restrict__ zero,
-                             ConvShape sh, int grid_m) {
+                             ConvShape sh, int grid_m,
+                             const float* __restrict__ epi_scale,
+                             const float* __restrict__ epi_shift,
+                             const bf16* __restrict__ epi_res, int epi_relu) {
   constexpr int BM = GWR * 64, BN = GWC * 64, BK = 64;
   constexpr int NA = 2 * GWR, NB = 2 * GWC;  // 16B chunks per thread/operand
   // XCD-aware block remap (T1): contiguous output tiles on one XCD share B
@@ -362,7 +368,13 @@ __global__ void igemm_kernel(const bf16* __restrict__ A, const bf16* __restrict_
           } else {
             off = row * sh.Nout + col;
           }
-          out[off] = f2bf(acc[mi][ni][r]);
+          float v = acc[mi][ni][r];
+          if (epi_scale) {
+            v = v * epi_scale[col] + epi_shift[col];
+            if (epi_res) v += bf2f(epi_res[off]);
+            if (epi_relu) v = fmaxf(v, 0.f);
+          }
+          out[off] = f2bf(v);
         }
       }
     }
@@ -377,7 +389,10 @@ __global__ void igemm_kernel(const bf16* __restrict__ A, const bf16* __restrict_
 template <int MODE>
 __global__ void conv_direct_kernel(const bf16* __restrict__ A,
                                    const bf16* __restrict__ B,
-                                   bf16* __restrict__ out, ConvShape sh) {
+                                   bf16* __restrict__ out, ConvShape sh,
+                                   const float* __restrict__ epi_scale,
+                                   const float* __restrict__ epi_shift,
+                                   const bf16* __restrict__ epi_res, int epi_relu) {
   const long total = sh.M * sh.Nout;
   for (long i = grid_stride_begin(); i < total; i += grid_stride_step()) {
     const long m = i / sh.Nout;
@@ -387,6 +402,11 @@ __global__ void conv_direct_kernel(const bf16* __restrict__ A,
       const bf16* pa = a_chunk_ptr<MODE>(A, nullptr, sh, m, kd);
       if (pa == nullptr) continue;
       acc += bf2f(*pa) * bf2f(B[(long)j * sh.KD + kd]);
+    }
+    if (epi_scale) {
+      acc = acc * epi_scale[j] + epi_shift[j];
+      if (epi_res) acc += bf2f(epi_res[i]);
+      if (epi_relu) acc = fmaxf(acc, 0.f);
     }
     out[i] = f2bf(acc);
   }
@@ -406,7 +426,8 @@ static inline bool igemm_ok(int mode, const ConvShape& sh) {
 extern "C" void al_conv2d_mm(int mode, const void* A, const void* B, void* out,
                              const void* zero_page, int N, int H, int W, int C,
                              int K, int R, int S, int P, int Q, int stride, int pad,
-                             hipStream_t stream) {
+                             const float* epi_scale, const float* epi_shift,
+                             const void* epi_res, int epi_relu, hipStream_t stream) {
   ConvShape sh;
   sh.N = N; sh.H = H; sh.W = W; sh.C = C; sh.K = K; sh.R = R; sh.S = S;
   sh.P = P; sh.Q = Q; sh.stride = stride; sh.pad = pad;
@@ -456,11 +477,13 @@ extern "C" void al_conv2d_mm(int mode, const void* A, const void* B, void* out,
           if (narrow)
             hipLaunchKernelGGL((igemm_kernel<MODE_BWD_S2, 4, 1>), grid, block, lds,
                                stream, (const bf16*)A, (const bf16*)B, (bf16*)out,
-                               (const bf16*)zero_page, c, gm);
+                               (const bf16*)zero_page, c, gm, epi_scale, epi_shift,
+                               (const bf16*)epi_res, epi_relu);
           else
             hipLaunchKernelGGL((igemm_kernel<MODE_BWD_S2, 2, 2>), grid, block, lds,
                                stream, (const bf16*)A, (const bf16*)B, (bf16*)out,
-                               (const bf16*)zero_page, c, gm);
+                               (const bf16*)zero_page, c, gm, epi_scale, epi_shift,
+                               (const bf16*)epi_res, epi_relu);
         }
       return;
     }
@@ -474,7 +497,7 @@ extern "C" void al_conv2d_mm(int mode, const void* A, const void* B, void* out,
     const int grid_n = (sh.Nout + BN - 1) / BN;
     const size_t lds = 2 * (size_t)(BM + BN) * 64 * sizeof(bf16);
     dim3 grid(grid_m * grid_n), block(256);
-#define LAUNCH(MODE_, GWR_, GWC_)     hipLaunchKernelGGL((igemm_kernel<MODE_, GWR_, GWC_>), grid, block, lds, stream,                        (const bf16*)A, (const bf16*)B, (bf16*)out,                        (const bf16*)zero_page, sh, grid_m)
+#define LAUNCH(MODE_, GWR_, GWC_) hipLaunchKernelGGL((igemm_kernel<MODE_, GWR_, GWC_>), grid, block, lds, stream, (const bf16*)A, (const bf16*)B, (bf16*)out, (const bf16*)zero_page, sh, grid_m, epi_scale, epi_shift, (const bf16*)epi_res, epi_relu)
     if (mode == MODE_FWD) {
       if (narrow) LAUNCH(MODE_FWD, 4, 1); else LAUNCH(MODE_FWD, 2, 2);
     } else {
@@ -486,10 +509,12 @@ extern "C" void al_conv2d_mm(int mode, const void* A, const void* B, void* out,
     int blocks = (int)min((total + 255) / 256, (long)8192);
     if (mode == MODE_FWD)
       hipLaunchKernelGGL((conv_direct_kernel<MODE_FWD>), dim3(blocks), dim3(256), 0,
-                         stream, (const bf16*)A, (const bf16*)B, (bf16*)out, sh);
+                         stream, (const bf16*)A, (const bf16*)B, (bf16*)out, sh,
+                         epi_scale, epi_shift, (const bf16*)epi_res, epi_relu);
     else
       hipLaunchKernelGGL((conv_direct_kernel<MODE_BWD_DATA>), dim3(blocks), dim3(256),
-                         0, stream, (const bf16*)A, (const bf16*)B, (bf16*)out, sh);
+                         0, stream, (const bf16*)A, (const bf16*)B, (bf16*)out, sh,
+                         epi_scale, epi_shift, (const bf16*)epi_res, epi_relu);
   }
 }
 

@@ -257,3 +257,28 @@ def test_resnet50_smoke_gpu():
     for p in net.parameters():
         if p.grad is not None:
             assert torch.isfinite(p.grad).all()
+
+
+def test_fused_inference_path_matches_eval():
+    """Eval-mode no-grad forward uses the fused conv+BN epilogue; must match
+    the CPU eval forward."""
+    from active_learning_amd.models import get_networks
+    torch.manual_seed(3)
+    net_c = get_networks("synthetic_cifar10", "SSLResNet18")
+    net_g = get_networks("synthetic_cifar10", "SSLResNet18")
+    net_g.load_state_dict(net_c.state_dict())
+    net_g = net_g.cuda()
+    # give BN non-trivial running stats
+    x_warm = torch.randn(16, 3, 32, 32)
+    net_c.train()
+    net_g.train()
+    with torch.no_grad():
+        net_c(x_warm)
+        net_g(x_warm.cuda())
+    net_c.eval()
+    net_g.eval()
+    x = torch.randn(8, 3, 32, 32)
+    with torch.no_grad():
+        out_c = net_c(x)
+        out_g = net_g(x.cuda())
+    assert relerr(out_g, out_c) < 0.1, f"fused eval divergence {relerr(out_g, out_c)}"
