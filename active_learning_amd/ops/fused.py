@@ -8,9 +8,13 @@ ResNet-50 forward). Training-mode or grad-enabled calls fall back to the
 separate autograd ops.
 """
 
+import os
+
 import torch
 
 from .extension import extension_available, load_extension
+
+_DISABLED = os.environ.get("AL_AMD_DISABLE_FUSED_EVAL") == "1"
 from .functional import _igemm_eligible, _wpad_cached, cast_cached
 
 
@@ -32,8 +36,8 @@ def _bn_fold_cached(bn):
 
 def conv_bn_act(conv, bn, x, residual=None):
     """bn(conv(x), residual) with the fused single-kernel path when eligible."""
-    if (x.is_cuda and not bn.training and not torch.is_grad_enabled()
-            and extension_available()):
+    if (not _DISABLED and x.is_cuda and not bn.training
+            and not torch.is_grad_enabled() and extension_available()):
         ext = load_extension()
         w_c = cast_cached(conv.weight, x.dtype)
         K, R, S, C = w_c.shape
