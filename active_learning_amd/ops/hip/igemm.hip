@@ -345,37 +345,70 @@ __global__ void igemm_kernel(const bf16* __restrict__ A, const bf16* __restrict_
     buf ^= 1;  // next iteration's top barrier orders buffer reuse
   }
 
-  // epilogue: C/D layout row = (l>>4)*4 + r, col = l&15. For MODE_BWD_S2 the
-  // GEMM row is a class-local pixel; scatter to the strided dx location.
+  // ------------------------------------------------------------------
+  // epilogue: the MFMA C/D fragment layout (row = (l>>4)*4 + r, col = l&15)
+  // would store 16 scattered 2-byte elements per lane. Instead stage the
+  // accumulator tile through LDS (reusing the A staging buffers) and emit
+  // row-contiguous 16-byte stores — also making the fused-BN epilogue's
+  // residual reads and scale/shift loads coalesced.
+  // ------------------------------------------------------------------
+  __syncthreads();  // all waves done reading As/Bs
+  bf16* etile = As;  // [BM][BN] bf16 == 2*BM*BK elements (BN <= 2*BK)
 #pragma unroll
-  for (int mi = 0; mi < 4; ++mi) {
+  for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
-    for (int ni = 0; ni < 4; ++ni) {
-      const int col = n0 + wc * 64 + ni * 16 + l15;
-      if (col >= sh.Nout) continue;
+    for (int ni = 0; ni < 4; ++ni)
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const long row = m0 + wr * 64 + mi * 16 + l4 * 4 + r;
-        if (row < sh.M) {
-          long off;
-          if (MODE == MODE_BWD_S2) {
-            const int ww = (int)(row % sh.W);
-            long tt = row / sh.W;
-            const int hh = (int)(tt % sh.H);
-            const long n = tt / sh.H;
-            off = ((n * sh.Horig + 2 * hh + sh.hcl) * sh.Worig + 2 * ww + sh.wcl)
-                      * sh.Nout + col;
-          } else {
-            off = row * sh.Nout + col;
-          }
-          float v = acc[mi][ni][r];
-          if (epi_scale) {
-            v = v * epi_scale[col] + epi_shift[col];
-            if (epi_res) v += bf2f(epi_res[off]);
-            if (epi_relu) v = fmaxf(v, 0.f);
-          }
-          out[off] = f2bf(v);
+      for (int r = 0; r < 4; ++r)
+        etile[(wr * 64 + mi * 16 + l4 * 4 + r) * BN + wc * 64 + ni * 16 + l15] =
+            f2bf(acc[mi][ni][r]);
+  __syncthreads();
+
+  constexpr int CPR = BN / 8;              // 16B chunks per row
+  constexpr int NCH = BM * CPR / 256;      // chunks per thread
+#pragma unroll
+  for (int i = 0; i < NCH; ++i) {
+    const int chunk = tid + 256 * i;
+    const int row = chunk / CPR, cc = chunk % CPR;
+    const long m = m0 + row;
+    if (m >= sh.M) continue;
+    const int col0 = n0 + cc * 8;
+    if (col0 >= sh.Nout) continue;
+    long off;
+    if (MODE == MODE_BWD_S2) {
+      const int ww = (int)(m % sh.W);
+      long tt = m / sh.W;
+      const int hh = (int)(tt % sh.H);
+      const long n = tt / sh.H;
+      off = ((n * sh.Horig + 2 * hh + sh.hcl) * sh.Worig + 2 * ww + sh.wcl)
+                * sh.Nout + col0;
+    } else {
+      off = m * sh.Nout + col0;
+    }
+    s16x8 v = *(const s16x8*)(etile + row * BN + cc * 8);
+    if (col0 + 8 <= sh.Nout) {
+      if (epi_scale) {
+        s16x8 rv;
+        if (epi_res) rv = *(const s16x8*)(epi_res + off);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float f = bits2f(v[j]) * epi_scale[col0 + j] + epi_shift[col0 + j];
+          if (epi_res) f += bits2f(rv[j]);
+          if (epi_relu) f = fmaxf(f, 0.f);
+          v[j] = f2bits(f);
         }
+      }
+      *(s16x8*)(out + off) = v;
+    } else {  // column tail: elementwise, all epi reads guarded
+      for (int j = 0; col0 + j < sh.Nout; ++j) {
+        float f = bits2f(v[j]);
+        if (epi_scale) {
+          f = f * epi_scale[col0 + j] + epi_shift[col0 + j];
+          if (epi_res) f += bf2f(epi_res[off + j]);
+          if (epi_relu) f = fmaxf(f, 0.f);
+        }
+        short sj = f2bits(f);
+        out[off + j] = *(bf16*)&sj;
       }
     }
   }
