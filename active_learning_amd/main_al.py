@@ -131,7 +131,10 @@ def main(args):
 
         t0 = time()
         strategy.test()
-        print(f"Rd {rd} test_time is {time() - t0}")
+        test_time = time() - t0
+        print(f"Rd {rd} test_time is {test_time}")
+        strategy.comet_experiment.log_metrics(
+            {f"rd_{rd}_test_time_s": test_time}, step=rd)
 
         save_experiment(strategy, args, logger)
         args.resume_training = True
