@@ -104,17 +104,21 @@ def main():
     if use_graph:
         # warm up eagerly, then capture one hipGraph per data buffer; replay
         # covers fwd + CE + bwd + fused SGD with zero per-kernel launch gaps
-        for i in range(max(3, args.warmup)):
-            step(i)
-        sync()
-        for i in range(n_buf):
-            g = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(g):
+        try:
+            for i in range(max(3, args.warmup)):
                 step(i)
-            graphs.append(g)
+            sync()
+            for i in range(n_buf):
+                g = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g):
+                    step(i)
+                graphs.append(g)
 
-        def step(i):  # noqa: F811 — replay path
-            graphs[i % n_buf].replay()
+            def step(i):  # noqa: F811 — replay path
+                graphs[i % n_buf].replay()
+        except Exception as e:  # never lose the bench to a capture failure
+            print(f"# hipGraph capture unavailable ({e!r}); running eager", flush=True)
+            graphs = []
     for i in range(args.warmup):
         step(i)
     sync()
