@@ -176,6 +176,49 @@ def conv2d(x, weight, stride=1, padding=0):
     return Conv2dNHWC.apply(x, weight, stride, padding)
 
 
+class Conv2dNHWCStats(Function):
+    """Conv that also returns per-column (sum, sumsq) of its output, computed
+    in the conv epilogue — feeds the following BatchNorm's batch statistics
+    without a separate full read of the activation (training path)."""
+
+    @staticmethod
+    def forward(ctx, x, weight, stride, padding):
+        w_c = cast_cached(weight, x.dtype)
+        ctx.save_for_backward(x)
+        ctx.w_c = w_c
+        ctx.stride, ctx.padding = stride, padding
+        ctx.weight_dtype = weight.dtype
+        if x.is_cuda:
+            ext = require_extension()
+            K, R, S, C = w_c.shape
+            if _igemm_eligible(C, R * S * C):
+                ctx.apack = None
+                y, s, ss = ext.conv2d_fwd_stats(x, w_c, stride, padding)
+            else:
+                K, R, S, C = w_c.shape
+                kdpad = ((R * S * C + 63) // 64) * 64
+                apack = ext.im2col_pack(x, R, S, stride, padding, kdpad)
+                ctx.apack = apack
+                y, s, ss = ext.conv2d_fwd_stats(apack, _wpad_cached(w_c, kdpad), 1, 0)
+        else:
+            xc, wc = _cpu_conv_args(x, w_c)
+            yt = F.conv2d(xc.float(), wc.float(), stride=stride, padding=padding)
+            y = yt.to(x.dtype).permute(0, 2, 3, 1).contiguous()
+            yf = y.float()
+            s = yf.sum(dim=(0, 1, 2))
+            ss = (yf * yf).sum(dim=(0, 1, 2))
+        ctx.mark_non_differentiable(s, ss)
+        return y, s, ss
+
+    @staticmethod
+    def backward(ctx, dy, _ds, _dss):
+        return Conv2dNHWC.backward(ctx, dy)
+
+
+def conv2d_with_stats(x, weight, stride=1, padding=0):
+    return Conv2dNHWCStats.apply(x, weight, stride, padding)
+
+
 class BatchNormAct(Function):
     """Fused BatchNorm (+ optional residual add + optional ReLU) over NHWC.
 
@@ -191,12 +234,48 @@ class BatchNormAct(Function):
 
     @staticmethod
     def forward(ctx, x, weight, bias, running_mean, running_var, use_batch_stats,
-                momentum, eps, relu, residual, pg):
+                momentum, eps, relu, residual, pg, pre_sums=None):
         n_local = x.numel() // x.shape[-1]
         sync = (pg is not None and dist.is_initialized()
                 and dist.get_world_size(pg) > 1)
         if use_batch_stats:
-            if x.is_cuda and not sync:
+            if pre_sums is not None:
+                # sums produced in the preceding conv's epilogue
+                s, ss = pre_sums
+                if sync:
+                    count = torch.tensor([float(n_local)], device=x.device)
+                    packed = torch.cat([s, ss, count])
+                    dist.all_reduce(packed, group=pg)
+                    s, ss = packed[:len(s)], packed[len(s):2 * len(s)]
+                    n = float(packed[-1].item())
+                    mean = s / n
+                    var = (ss / n - mean * mean).clamp_min_(0)
+                    invstd = (var + eps).rsqrt()
+                    with torch.no_grad():
+                        if running_mean is not None:
+                            unbiased = var * (n / max(n - 1, 1))
+                            running_mean.mul_(1 - momentum).add_(mean, alpha=momentum)
+                            running_var.mul_(1 - momentum).add_(unbiased,
+                                                                alpha=momentum)
+                elif x.is_cuda:
+                    from .extension import require_extension as _re
+                    with torch.no_grad():
+                        mean, invstd = _re().bn_finalize(
+                            s, ss, running_mean, running_var, float(n_local),
+                            momentum, eps, running_mean is not None)
+                    n = float(n_local)
+                else:
+                    n = float(n_local)
+                    mean = s / n
+                    var = (ss / n - mean * mean).clamp_min_(0)
+                    invstd = (var + eps).rsqrt()
+                    with torch.no_grad():
+                        if running_mean is not None:
+                            unbiased = var * (n / max(n - 1, 1))
+                            running_mean.mul_(1 - momentum).add_(mean, alpha=momentum)
+                            running_var.mul_(1 - momentum).add_(unbiased,
+                                                                alpha=momentum)
+            elif x.is_cuda and not sync:
                 # fused path: partial sums -> mean/invstd + running update in
                 # two kernels, no small ATen ops, no host sync
                 ext = require_extension()
@@ -306,13 +385,15 @@ class BatchNormAct(Function):
             else:
                 dx = g * dyf
             dx = dx.to(x.dtype)
-        return dx, dgamma, dbeta, None, None, None, None, None, None, dres, None
+        return dx, dgamma, dbeta, None, None, None, None, None, None, dres, None, None
 
 
 def batch_norm_act(x, weight, bias, running_mean, running_var, use_batch_stats,
-                   momentum=0.1, eps=1e-5, relu=True, residual=None, pg=None):
+                   momentum=0.1, eps=1e-5, relu=True, residual=None, pg=None,
+                   pre_sums=None):
     return BatchNormAct.apply(x, weight, bias, running_mean, running_var,
-                              use_batch_stats, momentum, eps, relu, residual, pg)
+                              use_batch_stats, momentum, eps, relu, residual, pg,
+                              pre_sums)
 
 
 class MaxPool2dNHWC(Function):

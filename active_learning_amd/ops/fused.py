@@ -15,6 +15,7 @@ import torch
 from .extension import extension_available, load_extension
 
 _DISABLED = os.environ.get("AL_AMD_DISABLE_FUSED_EVAL") == "1"
+_STATS_DISABLED = os.environ.get("AL_AMD_DISABLE_CONV_STATS") == "1"
 from .functional import _igemm_eligible, _wpad_cached, cast_cached
 
 
@@ -36,6 +37,14 @@ def _bn_fold_cached(bn):
 
 def conv_bn_act(conv, bn, x, residual=None):
     """bn(conv(x), residual) with the fused single-kernel path when eligible."""
+    if (not _STATS_DISABLED and x.is_cuda and bn.training
+            and extension_available()):
+        # training: conv computes the BN batch statistics in its epilogue
+        from .functional import batch_norm_act, conv2d_with_stats
+        y, s, ss = conv2d_with_stats(x, conv.weight, conv.stride, conv.padding)
+        return batch_norm_act(y, bn.weight, bn.bias, bn.running_mean,
+                              bn.running_var, True, bn.momentum, bn.eps, bn.relu,
+                              residual, bn._pg(), pre_sums=(s, ss))
     if (not _DISABLED and x.is_cuda and not bn.training
             and not torch.is_grad_enabled() and extension_available()):
         ext = load_extension()

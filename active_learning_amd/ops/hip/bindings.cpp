@@ -17,7 +17,7 @@ void al_conv2d_mm(int mode, const void* A, const void* B, void* out,
                   const void* zero_page, int N, int H, int W, int C, int K, int R,
                   int S, int P, int Q, int stride, int pad, const float* epi_scale,
                   const float* epi_shift, const void* epi_res, int epi_relu,
-                  void* stream);
+                  float* stat_sum, float* stat_sumsq, void* stream);
 void al_conv2d_wgrad(const void* dy, const void* x, float* dw, int N, int H, int W,
                      int C, int K, int R, int S, int P, int Q, int stride, int pad,
                      void* stream);
@@ -92,8 +92,30 @@ Tensor conv2d_fwd(const Tensor& x, const Tensor& w, long stride, long pad) {
   auto y = torch::empty({N, P, Q, K}, x.options());
   al_conv2d_mm(0, x.data_ptr(), w.data_ptr(), y.data_ptr(),
                zero_page(x).data_ptr(), N, H, W, C, K, R, S, P, Q, (int)stride,
-               (int)pad, nullptr, nullptr, nullptr, 0, cur_stream());
+               (int)pad, nullptr, nullptr, nullptr, 0, nullptr, nullptr,
+               cur_stream());
   return y;
+}
+
+std::vector<Tensor> conv2d_fwd_stats(const Tensor& x, const Tensor& w, long stride,
+                                     long pad) {
+  // training-path fusion: conv + per-column sum/sumsq of the output in the
+  // same kernel (feeds the following BatchNorm's batch statistics)
+  check_bf16_contig(x, "x");
+  check_bf16_contig(w, "w");
+  const int N = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
+  const int K = w.size(0), R = w.size(1), S = w.size(2);
+  TORCH_CHECK(w.size(3) == C, "conv2d_fwd_stats: channel mismatch");
+  const int P = out_dim(H, R, stride, pad), Q = out_dim(W, S, stride, pad);
+  auto y = torch::empty({N, P, Q, K}, x.options());
+  auto opts = x.options().dtype(torch::kFloat32);
+  auto sum = torch::zeros({K}, opts);
+  auto sumsq = torch::zeros({K}, opts);
+  al_conv2d_mm(0, x.data_ptr(), w.data_ptr(), y.data_ptr(),
+               zero_page(x).data_ptr(), N, H, W, C, K, R, S, P, Q, (int)stride,
+               (int)pad, nullptr, nullptr, nullptr, 0, sum.data_ptr<float>(),
+               sumsq.data_ptr<float>(), cur_stream());
+  return {y, sum, sumsq};
 }
 
 Tensor conv2d_fwd_fused(const Tensor& x, const Tensor& w, long stride, long pad,
@@ -115,7 +137,8 @@ Tensor conv2d_fwd_fused(const Tensor& x, const Tensor& w, long stride, long pad,
                zero_page(x).data_ptr(), N, H, W, C, K, R, S, P, Q, (int)stride,
                (int)pad, scale.contiguous().data_ptr<float>(),
                shift.contiguous().data_ptr<float>(),
-               has_res ? residual.data_ptr() : nullptr, relu ? 1 : 0, cur_stream());
+               has_res ? residual.data_ptr() : nullptr, relu ? 1 : 0, nullptr,
+               nullptr, cur_stream());
   return y;
 }
 
@@ -131,7 +154,8 @@ Tensor conv2d_bwd_data(const Tensor& dy, const Tensor& wt, long stride, long pad
   auto dx = torch::empty({N, (long)H, (long)W, C}, dy.options());
   al_conv2d_mm(1, dy.data_ptr(), wt.data_ptr(), dx.data_ptr(),
                zero_page(dy).data_ptr(), N, (int)H, (int)W, C, K, R, S, P, Q,
-               (int)stride, (int)pad, nullptr, nullptr, nullptr, 0, cur_stream());
+               (int)stride, (int)pad, nullptr, nullptr, nullptr, 0, nullptr, nullptr,
+               cur_stream());
   return dx;
 }
 
@@ -180,6 +204,20 @@ std::vector<Tensor> bn_stats_finalize(const Tensor& x, Tensor& running_mean,
                  running_mean.data_ptr<float>(), running_var.data_ptr<float>(), nb, C,
                  (float)rows, (float)momentum, (float)eps, update_running ? 1 : 0,
                  cur_stream());
+  return {mean, invstd};
+}
+
+std::vector<Tensor> bn_finalize(const Tensor& s, const Tensor& ss, Tensor& rm,
+                                Tensor& rv, double n, double momentum, double eps,
+                                bool update_running) {
+  const int C = s.numel();
+  auto opts = s.options();
+  auto mean = torch::empty({C}, opts);
+  auto invstd = torch::empty({C}, opts);
+  al_bn_finalize(s.contiguous().data_ptr<float>(), ss.contiguous().data_ptr<float>(),
+                 mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                 rm.data_ptr<float>(), rv.data_ptr<float>(), 1, C, (float)n,
+                 (float)momentum, (float)eps, update_running ? 1 : 0, cur_stream());
   return {mean, invstd};
 }
 
@@ -357,10 +395,12 @@ void adam_step(Tensor& p, const Tensor& g, Tensor& m, Tensor& v, double lr, doub
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv2d_fwd", &conv2d_fwd);
   m.def("conv2d_fwd_fused", &conv2d_fwd_fused);
+  m.def("conv2d_fwd_stats", &conv2d_fwd_stats);
   m.def("conv2d_bwd_data", &conv2d_bwd_data);
   m.def("conv2d_bwd_weight", &conv2d_bwd_weight);
   m.def("bn_stats", &bn_stats);
   m.def("bn_stats_finalize", &bn_stats_finalize);
+  m.def("bn_finalize", &bn_finalize);
   m.def("bn_norm_fwd", &bn_norm_fwd);
   m.def("bn_bwd_reduce", &bn_bwd_reduce);
   m.def("bn_bwd", &bn_bwd);

@@ -124,7 +124,9 @@ __global__ void igemm_kernel(const bf16* __restrict__ A, const bf16* __restrict_
                              ConvShape sh, int grid_m,
                              const float* __restrict__ epi_scale,
                              const float* __restrict__ epi_shift,
-                             const bf16* __restrict__ epi_res, int epi_relu) {
+                             const bf16* __restrict__ epi_res, int epi_relu,
+                             float* __restrict__ stat_sum,
+                             float* __restrict__ stat_sumsq) {
   constexpr int BM = GWR * 64, BN = GWC * 64, BK = 64;
   constexpr int NA = 2 * GWR, NB = 2 * GWC;  // 16B chunks per thread/operand
   // XCD-aware block remap (T1): contiguous output tiles on one XCD share B
@@ -354,6 +356,15 @@ __global__ void igemm_kernel(const bf16* __restrict__ A, const bf16* __restrict_
   // ------------------------------------------------------------------
   __syncthreads();  // all waves done reading As/Bs
   bf16* etile = As;  // [BM][BN] bf16 == 2*BM*BK elements (BN <= 2*BK)
+  // optional per-column statistics for the following BatchNorm (training):
+  // block-local sums in the free B-staging region, one global atomic per
+  // column per block at the end (SURVEY.md §2.4 BN-stats fusion)
+  float* ssum = (float*)Bs;          // [BN]
+  float* ssq = ssum + BN;            // [BN]
+  if (stat_sum && tid < BN) {
+    ssum[tid] = 0.f;
+    ssq[tid] = 0.f;
+  }
 #pragma unroll
   for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
@@ -386,6 +397,16 @@ __global__ void igemm_kernel(const bf16* __restrict__ A, const bf16* __restrict_
       off = m * sh.Nout + col0;
     }
     s16x8 v = *(const s16x8*)(etile + row * BN + cc * 8);
+    if (stat_sum) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        if (col0 + j < sh.Nout) {
+          const float f = bits2f(v[j]);
+          atomicAdd(&ssum[cc * 8 + j], f);
+          atomicAdd(&ssq[cc * 8 + j], f * f);
+        }
+      }
+    }
     if (col0 + 8 <= sh.Nout) {
       if (epi_scale) {
         s16x8 rv;
@@ -412,6 +433,13 @@ __global__ void igemm_kernel(const bf16* __restrict__ A, const bf16* __restrict_
       }
     }
   }
+  if (stat_sum) {
+    __syncthreads();
+    if (tid < BN && n0 + tid < sh.Nout) {
+      atomicAdd(&stat_sum[n0 + tid], ssum[tid]);
+      atomicAdd(&stat_sumsq[n0 + tid], ssq[tid]);
+    }
+  }
 }
 
 // ---------------------------------------------------------------------------
@@ -425,7 +453,9 @@ __global__ void conv_direct_kernel(const bf16* __restrict__ A,
                                    bf16* __restrict__ out, ConvShape sh,
                                    const float* __restrict__ epi_scale,
                                    const float* __restrict__ epi_shift,
-                                   const bf16* __restrict__ epi_res, int epi_relu) {
+                                   const bf16* __restrict__ epi_res, int epi_relu,
+                                   float* __restrict__ stat_sum,
+                                   float* __restrict__ stat_sumsq) {
   const long total = sh.M * sh.Nout;
   for (long i = grid_stride_begin(); i < total; i += grid_stride_step()) {
     const long m = i / sh.Nout;
@@ -440,6 +470,10 @@ __global__ void conv_direct_kernel(const bf16* __restrict__ A,
       acc = acc * epi_scale[j] + epi_shift[j];
       if (epi_res) acc += bf2f(epi_res[i]);
       if (epi_relu) acc = fmaxf(acc, 0.f);
+    }
+    if (stat_sum) {
+      atomicAdd(&stat_sum[j], acc);
+      atomicAdd(&stat_sumsq[j], acc * acc);
     }
     out[i] = f2bf(acc);
   }
@@ -460,7 +494,8 @@ extern "C" void al_conv2d_mm(int mode, const void* A, const void* B, void* out,
                              const void* zero_page, int N, int H, int W, int C,
                              int K, int R, int S, int P, int Q, int stride, int pad,
                              const float* epi_scale, const float* epi_shift,
-                             const void* epi_res, int epi_relu, hipStream_t stream) {
+                             const void* epi_res, int epi_relu, float* stat_sum,
+                             float* stat_sumsq, hipStream_t stream) {
   ConvShape sh;
   sh.N = N; sh.H = H; sh.W = W; sh.C = C; sh.K = K; sh.R = R; sh.S = S;
   sh.P = P; sh.Q = Q; sh.stride = stride; sh.pad = pad;
@@ -511,12 +546,12 @@ extern "C" void al_conv2d_mm(int mode, const void* A, const void* B, void* out,
             hipLaunchKernelGGL((igemm_kernel<MODE_BWD_S2, 4, 1>), grid, block, lds,
                                stream, (const bf16*)A, (const bf16*)B, (bf16*)out,
                                (const bf16*)zero_page, c, gm, epi_scale, epi_shift,
-                               (const bf16*)epi_res, epi_relu);
+                               (const bf16*)epi_res, epi_relu, stat_sum, stat_sumsq);
           else
             hipLaunchKernelGGL((igemm_kernel<MODE_BWD_S2, 2, 2>), grid, block, lds,
                                stream, (const bf16*)A, (const bf16*)B, (bf16*)out,
                                (const bf16*)zero_page, c, gm, epi_scale, epi_shift,
-                               (const bf16*)epi_res, epi_relu);
+                               (const bf16*)epi_res, epi_relu, stat_sum, stat_sumsq);
         }
       return;
     }
@@ -530,7 +565,7 @@ extern "C" void al_conv2d_mm(int mode, const void* A, const void* B, void* out,
     const int grid_n = (sh.Nout + BN - 1) / BN;
     const size_t lds = 2 * (size_t)(BM + BN) * 64 * sizeof(bf16);
     dim3 grid(grid_m * grid_n), block(256);
-#define LAUNCH(MODE_, GWR_, GWC_) hipLaunchKernelGGL((igemm_kernel<MODE_, GWR_, GWC_>), grid, block, lds, stream, (const bf16*)A, (const bf16*)B, (bf16*)out, (const bf16*)zero_page, sh, grid_m, epi_scale, epi_shift, (const bf16*)epi_res, epi_relu)
+#define LAUNCH(MODE_, GWR_, GWC_) hipLaunchKernelGGL((igemm_kernel<MODE_, GWR_, GWC_>), grid, block, lds, stream, (const bf16*)A, (const bf16*)B, (bf16*)out, (const bf16*)zero_page, sh, grid_m, epi_scale, epi_shift, (const bf16*)epi_res, epi_relu, stat_sum, stat_sumsq)
     if (mode == MODE_FWD) {
       if (narrow) LAUNCH(MODE_FWD, 4, 1); else LAUNCH(MODE_FWD, 2, 2);
     } else {
@@ -543,11 +578,11 @@ extern "C" void al_conv2d_mm(int mode, const void* A, const void* B, void* out,
     if (mode == MODE_FWD)
       hipLaunchKernelGGL((conv_direct_kernel<MODE_FWD>), dim3(blocks), dim3(256), 0,
                          stream, (const bf16*)A, (const bf16*)B, (bf16*)out, sh,
-                         epi_scale, epi_shift, (const bf16*)epi_res, epi_relu);
+                         epi_scale, epi_shift, (const bf16*)epi_res, epi_relu, stat_sum, stat_sumsq);
     else
       hipLaunchKernelGGL((conv_direct_kernel<MODE_BWD_DATA>), dim3(blocks), dim3(256),
                          0, stream, (const bf16*)A, (const bf16*)B, (bf16*)out, sh,
-                         epi_scale, epi_shift, (const bf16*)epi_res, epi_relu);
+                         epi_scale, epi_shift, (const bf16*)epi_res, epi_relu, stat_sum, stat_sumsq);
   }
 }
 
