@@ -359,12 +359,8 @@ __global__ void igemm_kernel(const bf16* __restrict__ A, const bf16* __restrict_
   // optional per-column statistics for the following BatchNorm (training):
   // block-local sums in the free B-staging region, one global atomic per
   // column per block at the end (SURVEY.md §2.4 BN-stats fusion)
-  float* ssum = (float*)Bs;          // [BN]
-  float* ssq = ssum + BN;            // [BN]
-  if (stat_sum && tid < BN) {
-    ssum[tid] = 0.f;
-    ssq[tid] = 0.f;
-  }
+  // layout: [4 waves][CPR groups][8 cols] fp32, in the free B region
+  float* sred = (float*)Bs;
 #pragma unroll
   for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
@@ -377,6 +373,8 @@ __global__ void igemm_kernel(const bf16* __restrict__ A, const bf16* __restrict_
 
   constexpr int CPR = BN / 8;              // 16B chunks per row
   constexpr int NCH = BM * CPR / 256;      // chunks per thread
+  static_assert(256 % CPR == 0, "chunk->column-group invariant");
+  float st_s[8] = {0, 0, 0, 0, 0, 0, 0, 0}, st_q[8] = {0, 0, 0, 0, 0, 0, 0, 0};
 #pragma unroll
   for (int i = 0; i < NCH; ++i) {
     const int chunk = tid + 256 * i;
@@ -398,13 +396,13 @@ __global__ void igemm_kernel(const bf16* __restrict__ A, const bf16* __restrict_
     }
     s16x8 v = *(const s16x8*)(etile + row * BN + cc * 8);
     if (stat_sum) {
+      // every chunk of this thread shares the same column group
+      // (cc = tid % CPR since 256 % CPR == 0): accumulate in registers
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        if (col0 + j < sh.Nout) {
-          const float f = bits2f(v[j]);
-          atomicAdd(&ssum[cc * 8 + j], f);
-          atomicAdd(&ssq[cc * 8 + j], f * f);
-        }
+        const float f = (col0 + j < sh.Nout) ? bits2f(v[j]) : 0.f;
+        st_s[j] += f;
+        st_q[j] += f * f;
       }
     }
     if (col0 + 8 <= sh.Nout) {
@@ -434,10 +432,34 @@ __global__ void igemm_kernel(const bf16* __restrict__ A, const bf16* __restrict_
     }
   }
   if (stat_sum) {
+    // lanes at stride CPR within a wave share a column group: shuffle-fold
+    // them, then one LDS slot per (wave, group) and one global atomic per
+    // column per block
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      for (int off = CPR; off < 64; off <<= 1) {
+        st_s[j] += __shfl_xor(st_s[j], off, 64);
+        st_q[j] += __shfl_xor(st_q[j], off, 64);
+      }
+    const int cc0 = tid % CPR;
+    if (lane < CPR) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        sred[(wid * CPR + cc0) * 16 + j] = st_s[j];
+        sred[(wid * CPR + cc0) * 16 + 8 + j] = st_q[j];
+      }
+    }
     __syncthreads();
     if (tid < BN && n0 + tid < sh.Nout) {
-      atomicAdd(&stat_sum[n0 + tid], ssum[tid]);
-      atomicAdd(&stat_sumsq[n0 + tid], ssq[tid]);
+      const int cc = tid / 8, j = tid % 8;
+      float ts = 0.f, tq = 0.f;
+#pragma unroll
+      for (int w = 0; w < 4; ++w) {
+        ts += sred[(w * CPR + cc) * 16 + j];
+        tq += sred[(w * CPR + cc) * 16 + 8 + j];
+      }
+      atomicAdd(&stat_sum[n0 + tid], ts);
+      atomicAdd(&stat_sumsq[n0 + tid], tq);
     }
   }
 }
