@@ -142,6 +142,22 @@ def _syncbn_check(rank, world):
     ref_dx = ref_in.grad.permute(0, 2, 3, 1)[rank * 4:(rank + 1) * 4]
     assert torch.allclose(shard.grad, ref_dx, atol=1e-5), "SyncBN backward mismatch"
 
+    # pre-computed-sums path (conv-epilogue stats fusion, ops/fused.py):
+    # identical result when local (sum, sumsq) are passed in
+    from active_learning_amd.ops.functional import batch_norm_act
+    bn2 = BatchNormAct2d(3, relu=False)
+    convert_sync_batchnorm(bn2)
+    bn2.train()
+    xf = shard.detach()
+    s = xf.float().sum(dim=(0, 1, 2))
+    ss = (xf.float() ** 2).sum(dim=(0, 1, 2))
+    y2 = batch_norm_act(xf, bn2.weight, bn2.bias, bn2.running_mean,
+                        bn2.running_var, True, bn2.momentum, bn2.eps, False,
+                        None, bn2._pg(), pre_sums=(s, ss))
+    assert torch.allclose(y2, ref_y_nhwc[rank * 4:(rank + 1) * 4].detach(),
+                          atol=1e-5), "SyncBN pre_sums mismatch"
+    assert torch.allclose(bn2.running_mean, ref_bn.running_mean, atol=1e-6)
+
 
 def test_syncbn_matches_fullbatch():
     _run_dist(_syncbn_check)
