@@ -168,7 +168,6 @@ def kcenter_greedy(dist: torch.Tensor, labeled: torch.Tensor, budget: int,
     # host inside the loop (one sync per QUERY, not per iteration — the
     # reference syncs every iteration, coreset_sampler.py:82-104)
     sel_buf = torch.empty(budget, dtype=torch.int64, device=dev)
-    n_pre = 0
 
     if labeled.any():
         min_dist = _masked_col_min(dist, labeled)

@@ -178,22 +178,9 @@ def test_gather_parallel_eval():
     _run_dist(_eval_gather_check)
 
 
-def _strategy_train_check(rank, world):
-    """Full Strategy.parallel_train_fn over gloo with world=2."""
-    from active_learning_amd.strategies import RandomSampler
-    import helpers
-    s = helpers.make_strategy(RandomSampler)
-    s.world_size = world
-    s.backend = "gloo"
-    s.update(np.arange(20), 20)
-    # emulate mp.spawn path: pg already initialized by _entry
-    s.parallel_train_fn(rank)
-    assert os.path.exists(s.generate_weight_paths()["best_ckpt"])
-
-
 def test_strategy_parallel_train_gloo():
-    # parallel_train_fn calls _init_distributed which would re-init; run the
-    # whole thing through Strategy.train-like spawn instead
+    """Full Strategy.parallel_train_fn over gloo with world=2 (the process
+    group is provided by the harness, mirroring the mp.spawn environment)."""
     _run_dist(_strategy_train_spawned)
 
 
