@@ -257,12 +257,12 @@ class BatchNormAct(Function):
                             running_mean.mul_(1 - momentum).add_(mean, alpha=momentum)
                             running_var.mul_(1 - momentum).add_(unbiased,
                                                                 alpha=momentum)
-                elif x.is_cuda:
+                elif x.is_cuda and running_mean is not None:
                     from .extension import require_extension as _re
                     with torch.no_grad():
                         mean, invstd = _re().bn_finalize(
                             s, ss, running_mean, running_var, float(n_local),
-                            momentum, eps, running_mean is not None)
+                            momentum, eps, True)
                     n = float(n_local)
                 else:
                     n = float(n_local)
