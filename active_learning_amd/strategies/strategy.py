@@ -364,7 +364,13 @@ class Strategy:
     # -- validation / early stop (strategy.py:383-442) --------------------- #
 
     def validation_and_early_stopping(self, rank, epoch, weight_paths):
-        if not self.es_params["use_es"]:
+        if not self.es_params["use_es"] or len(self.eval_idxs) == 0:
+            # no validation data (early stop off, or the balanced eval split
+            # collapsed to zero on a tiny pool): keep checkpoints current
+            if rank == 0:
+                torch.save(state_dict_with_marker(self.net), weight_paths["best_ckpt"])
+                torch.save(state_dict_with_marker(self.net),
+                           weight_paths["current_ckpt"])
             return False
         validation_data = Subset(self.al_set, indices=self.eval_idxs)
         if self.world_size > 1:

@@ -28,6 +28,19 @@ def accuracy(dataloader, net, top_k=(1, 5), **kwargs):
             ds = ds.dataset
         num_classes = ds.num_classes
 
+    n_dataset = len(dataloader.dataset)
+    if n_dataset == 0:
+        # empty eval split (tiny pools with many classes reduce the balanced
+        # split to zero, generate_initial_pool.py:20-24 semantics): report
+        # zeros instead of dividing by zero
+        zero = torch.tensor(0.0)
+        out = {f"top_{k}_correct_count": 0.0 for k in top_k}
+        out.update({f"top_{k}_accuracy": zero for k in top_k})
+        out.update(accuracy=zero, accuracy_byclass=torch.zeros(num_classes),
+                   count_byclass=torch.zeros(num_classes),
+                   corrects_byclass=torch.zeros(num_classes), count=0)
+        return out
+
     max_k = min(max(top_k), num_classes)
     corrects = {k: torch.zeros((), dtype=torch.float64) for k in top_k}
     corrects_byclass = torch.zeros(num_classes)
@@ -52,7 +65,6 @@ def accuracy(dataloader, net, top_k=(1, 5), **kwargs):
             if batch_idx % 25 == 0:
                 logger.info(f"\tEval Batch {batch_idx + 1}/{len(dataloader)}")
 
-    n_dataset = len(dataloader.dataset)
     out = {}
     for k in top_k:
         out[f"top_{k}_correct_count"] = corrects[k].item()

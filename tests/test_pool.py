@@ -83,3 +83,42 @@ def test_available_boolean_matches_list():
     mask = s.available_query_idxs(boolean=True)
     lst = s.available_query_idxs(boolean=False, shuffle=False)
     assert set(np.where(mask)[0].tolist()) == set(lst.tolist())
+
+
+def test_empty_eval_split_is_survivable():
+    """A tiny pool over many classes truncates the balanced eval split to zero
+    (reference generate_initial_pool.py:20-24 does size -= size % num_classes);
+    accuracy() and the early-stop path must tolerate the resulting empty eval
+    set instead of dividing by zero."""
+    import torch
+    from torch.utils.data import DataLoader, Subset
+    from active_learning_amd.utils.pool_init import generate_idxs
+    from active_learning_amd.utils.evaluation import accuracy
+    import helpers
+
+    from active_learning_amd.data.synthetic import get_data_synthetic
+    ds, _, _ = get_data_synthetic(10, 50, 10, (3, 8, 8), seed=0)
+    idxs = generate_idxs(ds, 7, "random_balance", random_seed=0)
+    assert len(idxs) == 0  # 7 % 10 truncates to zero — documented semantics
+
+    loader = DataLoader(Subset(ds, []), batch_size=4)
+    net = torch.nn.Sequential(torch.nn.Flatten(), torch.nn.LazyLinear(10))
+    out = accuracy(loader, net, num_classes=10)
+    assert out["count"] == 0
+    assert out["accuracy"].item() == 0.0
+    assert out["top_5_accuracy"].item() == 0.0
+
+
+def test_strategy_trains_with_empty_eval(tmp_path):
+    """End-to-end single-process round with an empty eval split: validation is
+    skipped, best checkpoint still written (strategy.py guard)."""
+    import numpy as np
+    from active_learning_amd.strategies import RandomSampler
+    import helpers
+
+    s = helpers.make_strategy(RandomSampler, ckpt_path=str(tmp_path))
+    s.eval_idxs = np.array([], dtype=np.int64)
+    s.update(np.arange(16), 16)
+    s.train()
+    import os
+    assert os.path.exists(s.generate_weight_paths()["best_ckpt"])
