@@ -80,7 +80,7 @@ def generate_idxs(train_set, size, generation_type: str, avoid_idxs=None, random
                 result.append(idx)
                 remaining[y] -= 1
                 size -= 1
-        return np.asarray(result)
+        return np.asarray(result, dtype=np.int64)  # empty list must stay int-indexable
 
     raise ValueError(f"Init pool type {generation_type!r} not implemented")
 

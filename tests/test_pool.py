@@ -100,6 +100,10 @@ def test_empty_eval_split_is_survivable():
     ds, _, _ = get_data_synthetic(10, 50, 10, (3, 8, 8), seed=0)
     idxs = generate_idxs(ds, 7, "random_balance", random_seed=0)
     assert len(idxs) == 0  # 7 % 10 truncates to zero — documented semantics
+    assert idxs.dtype == np.int64  # must stay usable as an index array
+    mask = np.ones(50, dtype=bool)
+    mask[idxs] = False  # the available_query_idxs pattern must not raise
+    assert mask.all()
 
     loader = DataLoader(Subset(ds, []), batch_size=4)
     net = torch.nn.Sequential(torch.nn.Flatten(), torch.nn.LazyLinear(10))
