@@ -207,3 +207,47 @@ def test_margin_clustering_assignment_persists():
     s.update(idxs, 4)
     # persisted assignment shrank by the consumed samples
     assert len(s.cluster_assignment) == n_before - 4
+
+
+def test_sharded_forward_pool_matches_single(monkeypatch, tmp_path):
+    """Multi-process sharded query (one worker per device; CPU here) must
+    return elementwise-identical logits/embeddings/labels in pool order."""
+    import numpy as np
+    import torch
+    from active_learning_amd.strategies import MarginSampler
+    from active_learning_amd.strategies.common import forward_pool, _should_shard
+    from helpers import make_strategy
+
+    s = make_strategy(MarginSampler, ckpt_path=str(tmp_path))
+    idxs = np.arange(30)
+
+    monkeypatch.setenv("AL_SHARD_QUERY", "0")
+    ref_logits, ref_emb, ref_y = forward_pool(s, idxs, want_embedding=True)
+
+    monkeypatch.setenv("AL_SHARD_QUERY", "1")
+    monkeypatch.setenv("AL_SHARD_QUERY_MIN", "8")
+    s.world_size = 2
+    assert _should_shard(s, idxs)
+    logits, emb, y = forward_pool(s, idxs, want_embedding=True)
+
+    assert torch.equal(y, ref_y)
+    assert torch.allclose(logits, ref_logits, atol=1e-6)
+    assert torch.allclose(emb, ref_emb, atol=1e-6)
+
+
+def test_shard_gating(monkeypatch, tmp_path):
+    import numpy as np
+    from active_learning_amd.strategies import MarginSampler
+    from active_learning_amd.strategies.common import _should_shard
+    from helpers import make_strategy
+
+    s = make_strategy(MarginSampler, ckpt_path=str(tmp_path))
+    idxs = np.arange(100)
+    monkeypatch.setenv("AL_SHARD_QUERY_MIN", "50")
+    s.world_size = 1
+    assert not _should_shard(s, idxs)        # single-device training -> no shard
+    s.world_size = 2
+    assert _should_shard(s, idxs)
+    assert not _should_shard(s, idxs[:10])   # too small to amortize the spawn
+    monkeypatch.setenv("AL_SHARD_QUERY", "0")
+    assert not _should_shard(s, idxs)        # kill switch
