@@ -465,6 +465,240 @@ __global__ void igemm_kernel(const bf16* __restrict__ A, const bf16* __restrict_
 }
 
 // ---------------------------------------------------------------------------
+// 256x256 deep-pipelined NT GEMM for 1x1 stride-1 convolutions (and their
+// bwd-data): D[M, Nout] = A[M, KD] @ B[Nout, KD]^T. A 1x1 s1 p0 conv needs no
+// im2col gather — A is the raw NHWC activation (fwd: x, KD=C) or gradient
+// (bwd: dY, KD=K), B the (Nout, KD) weight rows already used by the igemm —
+// so the guide's 8-phase 256^2 schedule applies directly: 8 waves (512
+// threads), BK=64, 128 KiB LDS as a ring of 4 half-tiles (128 rows) per
+// operand, one half-tile refilled per phase into the slot freed by the
+// previous phase, vmcnt(4) once per K-tile, no vmcnt(0) drain in the loop.
+// Phase = one C-quadrant (qm,qn) x K=64: 12 ds_read_b128 + 2 glds + 16 MFMA.
+// ---------------------------------------------------------------------------
+
+__launch_bounds__(512, 1)
+__global__ void gemm256_nt_kernel(const bf16* __restrict__ A,
+                                  const bf16* __restrict__ B,
+                                  bf16* __restrict__ out,
+                                  const bf16* __restrict__ zero,
+                                  long M, int Nout, int KD, int grid_m,
+                                  const float* __restrict__ epi_scale,
+                                  const float* __restrict__ epi_shift,
+                                  const bf16* __restrict__ epi_res, int epi_relu,
+                                  float* __restrict__ stat_sum,
+                                  float* __restrict__ stat_sumsq) {
+  constexpr int BM = 256, BN = 256, BK = 64;
+  constexpr int HALF = 128 * BK;        // bf16 elements per half-tile (16 KiB)
+  // XCD-aware bijective block remap (T1)
+  const int nwg = gridDim.x;
+  int bid = blockIdx.x;
+  {
+    const int nx = 8;
+    const int q = nwg / nx, r = nwg % nx;
+    const int xcd = bid % nx, pos = bid / nx;
+    if (pos < (xcd < r ? q + 1 : q))
+      bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + pos;
+  }
+  const int bm = bid % grid_m, bn = bid / grid_m;
+  const long m0 = (long)bm * BM;
+  const int n0 = bn * BN;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  // slot layout: A half-tiles [buf][half] then B half-tiles [buf][half]
+  bf16* As = (bf16*)smem;               // 4 x HALF
+  bf16* Bs = As + 4 * HALF;             // 4 x HALF
+
+  const int tid = threadIdx.x;
+  const int wid = tid >> 6, lane = tid & 63;
+  const int wrq = wid >> 2, wcq = wid & 3;  // wave grid inside a 128x128 quadrant
+  const int l15 = lane & 15, l4 = lane >> 4;
+  const int KT = KD / BK;
+
+  // ---- half-tile staging: chunk dc = load*512 + wid*64 + lane lands at LDS
+  // byte dc*16 of the slot; the bank-conflict swizzle (slot ^= row&7, as in
+  // igemm_kernel) is applied by remapping which SOURCE chunk each lane loads.
+  auto stage_half = [&](const bf16* base, long ld, long row0, long rows_max,
+                        int ktile, bf16* slot) {
+#pragma unroll
+    for (int li = 0; li < 2; ++li) {
+      const int dc = li * 512 + wid * 64 + lane;
+      const int drow = dc >> 3, du = dc & 7;
+      const int su = du ^ (drow & 7);
+      const long row = row0 + drow;
+      const bf16* src = (ktile < KT && row < rows_max)
+                            ? base + row * ld + (long)ktile * BK + su * 8
+                            : zero;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)src,
+          (__attribute__((address_space(3))) void*)(slot + li * 512 * 8 + wid * 512),
+          16, 0, 0);
+    }
+  };
+  // refill schedule within a K-tile group g (consuming buf = g&1): the q-order
+  // (0,0),(0,1),(1,0),(1,1) frees one slot per phase; each phase refills the
+  // slot last read in the previous phase:
+  //   ph0: A[buf^1] h1 <- ktile g+1   ph1: B[buf^1] h1 <- g+1
+  //   ph2: A[buf]   h0 <- ktile g+2   ph3: B[buf]   h0 <- g+2
+  auto refill = [&](int g, int ph) {
+    const int buf = g & 1;
+    switch (ph) {
+      case 0: stage_half(A, KD, m0 + 128, M, g + 1, As + ((buf ^ 1) * 2 + 1) * HALF); break;
+      case 1: stage_half(B, KD, n0 + 128, Nout, g + 1, Bs + ((buf ^ 1) * 2 + 1) * HALF); break;
+      case 2: stage_half(A, KD, m0, M, g + 2, As + (buf * 2) * HALF); break;
+      case 3: stage_half(B, KD, n0, Nout, g + 2, Bs + (buf * 2) * HALF); break;
+    }
+  };
+
+  f32x4 acc[2][2][4][2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j)
+#pragma unroll
+      for (int g = 0; g < 4; ++g)
+#pragma unroll
+        for (int n = 0; n < 2; ++n) acc[i][j][g][n] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  // prologue: K-tile 0 fully + K-tile 1's h0 halves (6 half-tiles); K-tile
+  // 1's h1 halves arrive via group 0's ph0/ph1 refills. Order matters for
+  // the vmcnt(4) accounting below.
+  stage_half(A, KD, m0, M, 0, As);
+  stage_half(B, KD, n0, Nout, 0, Bs);
+  stage_half(A, KD, m0 + 128, M, 0, As + HALF);
+  stage_half(B, KD, n0 + 128, Nout, 0, Bs + HALF);
+  stage_half(A, KD, m0, M, 1, As + 2 * HALF);
+  stage_half(B, KD, n0, Nout, 1, Bs + 2 * HALF);
+
+  for (int g = 0; g < KT; ++g) {
+    const int buf = g & 1;
+    // once per K-tile: the 2 half-tiles issued since this tile's last half
+    // (4 loads/wave) may stay in flight — never drain to vmcnt(0) (T3+T4)
+    asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+#pragma unroll
+    for (int ph = 0; ph < 4; ++ph) {
+      const int qm = ph >> 1, qn = ph & 1;
+      const bf16* ah = As + (buf * 2 + qm) * HALF;
+      const bf16* bh = Bs + (buf * 2 + qn) * HALF;
+      bf16x8 af[4][2], bf[2][2];
+#pragma unroll
+      for (int fg = 0; fg < 4; ++fg) {
+        const int rowh = wrq * 64 + fg * 16 + l15;
+#pragma unroll
+        for (int kc = 0; kc < 2; ++kc)
+          af[fg][kc] = *(const bf16x8*)(ah + rowh * BK + (((kc * 4 + l4) ^ (rowh & 7)) * 8));
+      }
+#pragma unroll
+      for (int ng = 0; ng < 2; ++ng) {
+        const int colh = wcq * 32 + ng * 16 + l15;
+#pragma unroll
+        for (int kc = 0; kc < 2; ++kc)
+          bf[ng][kc] = *(const bf16x8*)(bh + colh * BK + (((kc * 4 + l4) ^ (colh & 7)) * 8));
+      }
+      refill(g, ph);
+      __builtin_amdgcn_s_barrier();
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int kc = 0; kc < 2; ++kc)
+#pragma unroll
+        for (int fg = 0; fg < 4; ++fg)
+#pragma unroll
+          for (int ng = 0; ng < 2; ++ng)
+            acc[qm][qn][fg][ng] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                af[fg][kc], bf[ng][kc], acc[qm][qn][fg][ng], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+      __builtin_amdgcn_s_barrier();
+    }
+  }
+
+  // ---- epilogue: LDS round trip for contiguous 16B stores (same scheme as
+  // igemm_kernel; etile = [256][256] bf16 = the whole 128 KiB LDS)
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // tail zero-page glds
+  __syncthreads();
+  bf16* etile = As;
+#pragma unroll
+  for (int qm = 0; qm < 2; ++qm)
+#pragma unroll
+    for (int qn = 0; qn < 2; ++qn)
+#pragma unroll
+      for (int fg = 0; fg < 4; ++fg)
+#pragma unroll
+        for (int ng = 0; ng < 2; ++ng)
+#pragma unroll
+          for (int r = 0; r < 4; ++r)
+            etile[(qm * 128 + wrq * 64 + fg * 16 + l4 * 4 + r) * BN +
+                  qn * 128 + wcq * 32 + ng * 16 + l15] = f2bf(acc[qm][qn][fg][ng][r]);
+  __syncthreads();
+
+  constexpr int CPR = BN / 8;          // 32 16B chunks per row
+  constexpr int NCH = BM * CPR / 512;  // 16 chunks per thread
+  static_assert(512 % CPR == 0, "chunk->column-group invariant");
+  float st_s[8] = {0, 0, 0, 0, 0, 0, 0, 0}, st_q[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+#pragma unroll
+  for (int i = 0; i < NCH; ++i) {
+    const int chunk = tid + 512 * i;
+    const int row = chunk / CPR, cc = chunk % CPR;
+    const long m = m0 + row;
+    if (m >= M) continue;
+    const int col0 = n0 + cc * 8;     // Nout % 256 == 0: no column tail
+    const long off = m * Nout + col0;
+    s16x8 v = *(const s16x8*)(etile + row * BN + cc * 8);
+    if (stat_sum) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const float f = bits2f(v[j]);
+        st_s[j] += f;
+        st_q[j] += f * f;
+      }
+    }
+    if (epi_scale) {
+      s16x8 rv;
+      if (epi_res) rv = *(const s16x8*)(epi_res + off);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float f = bits2f(v[j]) * epi_scale[col0 + j] + epi_shift[col0 + j];
+        if (epi_res) f += bits2f(rv[j]);
+        if (epi_relu) f = fmaxf(f, 0.f);
+        v[j] = f2bits(f);
+      }
+    }
+    *(s16x8*)(out + off) = v;
+  }
+  if (stat_sum) {
+    // lanes at stride CPR=32 share a column group: one xor fold, then the
+    // per-wave partials go through LDS (etile reads are done) and one global
+    // atomic per column per block
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      st_s[j] += __shfl_xor(st_s[j], 32, 64);
+      st_q[j] += __shfl_xor(st_q[j], 32, 64);
+    }
+    __syncthreads();
+    float* sred = (float*)As;         // [8 waves][32 groups][16]
+    const int cc0 = tid % CPR;
+    if (lane < CPR) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        sred[(wid * CPR + cc0) * 16 + j] = st_s[j];
+        sred[(wid * CPR + cc0) * 16 + 8 + j] = st_q[j];
+      }
+    }
+    __syncthreads();
+    if (tid < BN) {
+      const int cc = tid / 8, j = tid % 8;
+      float ts = 0.f, tq = 0.f;
+#pragma unroll
+      for (int w = 0; w < 8; ++w) {
+        ts += sred[(w * CPR + cc) * 16 + j];
+        tq += sred[(w * CPR + cc) * 16 + 8 + j];
+      }
+      atomicAdd(&stat_sum[n0 + tid], ts);
+      atomicAdd(&stat_sumsq[n0 + tid], tq);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // direct fallback (any shape; used for the C=3 stems): one output element per
 // thread, fp32 accumulate. Slow path by design — stems are ~3% of FLOPs.
 // ---------------------------------------------------------------------------
@@ -577,6 +811,32 @@ extern "C" void al_conv2d_mm(int mode, const void* A, const void* B, void* out,
         }
       return;
     }
+  }
+
+  // 1x1 stride-1 convs are plain NT GEMMs: route big ones to the 8-phase
+  // 256^2 kernel (fwd and bwd-data share it — B is (Nout, KD) in both)
+  static int gemm256_on = -1;
+  if (gemm256_on < 0) {
+    const char* e = getenv("AL_DISABLE_GEMM256");
+    gemm256_on = (e && e[0] == '1') ? 0 : 1;
+  }
+  if (gemm256_on && (mode == MODE_FWD || mode == MODE_BWD_DATA) && R == 1 &&
+      S == 1 && stride == 1 && pad == 0 && sh.Nout % 256 == 0 &&
+      sh.KD % 64 == 0 && sh.M >= 128) {
+    static bool attr_set = false;
+    if (!attr_set) {
+      (void)hipFuncSetAttribute((const void*)gemm256_nt_kernel,
+                          hipFuncAttributeMaxDynamicSharedMemorySize, 131072);
+      attr_set = true;
+    }
+    const int grid_m = (int)((sh.M + 255) / 256);
+    const int grid_n = sh.Nout / 256;
+    hipLaunchKernelGGL(gemm256_nt_kernel, dim3(grid_m * grid_n), dim3(512),
+                       131072, stream, (const bf16*)A, (const bf16*)B,
+                       (bf16*)out, (const bf16*)zero_page, sh.M, sh.Nout, sh.KD,
+                       grid_m, epi_scale, epi_shift, (const bf16*)epi_res,
+                       epi_relu, stat_sum, stat_sumsq);
+    return;
   }
 
   if (igemm_ok(mode, sh)) {

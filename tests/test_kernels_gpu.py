@@ -39,6 +39,11 @@ CONV_CASES = [
     # direct fallback (stem-like)
     (2, 16, 16, 3, 64, 3, 1, 1),
     (1, 32, 32, 3, 64, 7, 2, 3),
+    # 1x1 s1 shapes routed to the 8-phase 256^2 NT GEMM (gemm256_nt_kernel):
+    (2, 16, 16, 64, 256, 1, 1, 0),    # KT=1 edge (KD=64), M=512
+    (1, 16, 16, 128, 256, 1, 1, 0),   # KT=2 edge, M=256 exact
+    (2, 14, 14, 256, 512, 1, 1, 0),   # M=392 tail, KT=4
+    (2, 16, 16, 512, 64, 1, 1, 0),    # bwd-data routes (Nout=C=512), fwd narrow
 ]
 
 
@@ -54,7 +59,7 @@ def test_conv_fwd(case):
     assert relerr(y, ref) < 0.02, f"conv fwd {case}: relerr {relerr(y, ref)}"
 
 
-@pytest.mark.parametrize("case", CONV_CASES[:5])
+@pytest.mark.parametrize("case", CONV_CASES[:5] + CONV_CASES[7:])
 def test_conv_bwd_data(case):
     n, h, w, c, k, r, stride, pad = case
     p = (h + 2 * pad - r) // stride + 1
@@ -284,16 +289,35 @@ def test_fused_inference_path_matches_eval():
     assert relerr(out_g, out_c) < 0.1, f"fused eval divergence {relerr(out_g, out_c)}"
 
 
-def test_conv_stats_fusion_matches():
+@pytest.mark.parametrize("shape", [
+    (4, 14, 14, 128, 128, 3, 1, 1),
+    (4, 14, 14, 256, 512, 1, 1, 0),   # gemm256 path with stats epilogue
+])
+def test_conv_stats_fusion_matches(shape):
     """conv2d_fwd_stats sums must equal torch sums of the conv output."""
-    x = torch.randn(4, 14, 14, 128, device="cuda").to(torch.bfloat16)
-    w = (torch.randn(128, 3, 3, 128, device="cuda") * 0.05).to(torch.bfloat16)
-    y, s, ss = EXT.conv2d_fwd_stats(x, w, 1, 1)
-    y2 = EXT.conv2d_fwd(x, w, 1, 1)
+    n, h, w_, c, k, r, stride, pad = shape
+    x = torch.randn(n, h, w_, c, device="cuda").to(torch.bfloat16)
+    w = (torch.randn(k, r, r, c, device="cuda") * 0.05).to(torch.bfloat16)
+    y, s, ss = EXT.conv2d_fwd_stats(x, w, stride, pad)
+    y2 = EXT.conv2d_fwd(x, w, stride, pad)
     assert torch.equal(y, y2)
     yf = y.float()
     assert relerr(s, yf.sum(dim=(0, 1, 2))) < 1e-3
     assert relerr(ss, (yf * yf).sum(dim=(0, 1, 2))) < 1e-3
+
+
+def test_gemm256_fused_epilogue():
+    """1x1 route with the folded-BN inference epilogue (scale/shift/res/relu)."""
+    n, h, w_, c, k = 2, 14, 14, 256, 256
+    x = torch.randn(n, h, w_, c, device="cuda").to(torch.bfloat16)
+    w = (torch.randn(k, 1, 1, c, device="cuda") * 0.05).to(torch.bfloat16)
+    scale = torch.rand(k, device="cuda") + 0.5
+    shift = torch.randn(k, device="cuda") * 0.1
+    res = torch.randn(n, h, w_, k, device="cuda").to(torch.bfloat16)
+    y = EXT.conv2d_fwd_fused(x, w, 1, 0, scale, shift, res, True)
+    base = EXT.conv2d_fwd(x, w, 1, 0).float()
+    ref = (base * scale + shift + res.float()).clamp_min(0)
+    assert relerr(y, ref) < 0.02
 
 
 def test_train_forward_with_stats_fusion_matches_cpu():

@@ -1,0 +1,60 @@
+#!/usr/bin/env python3
+"""A/B the 8-phase 256^2 NT GEMM (1x1 s1 conv route) against the 128^2 igemm.
+Run twice on a GPU box (the route switch is read once per process):
+  python tools/ab_gemm256.py            # gemm256 enabled
+  AL_DISABLE_GEMM256=1 python tools/ab_gemm256.py
+"""
+
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import torch
+
+from active_learning_amd.ops.extension import require_extension
+
+
+def timeit(fn, iters=30, warmup=8):
+    for _ in range(warmup):
+        fn()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        fn()
+    torch.cuda.synchronize()
+    return (time.perf_counter() - t0) / iters
+
+
+# R50 @ B=256 1x1 stride-1 shapes: (N, HW, C_in, K_out)
+SHAPES = [
+    ("l2.conv3", 256, 28, 128, 512),
+    ("l3.conv1", 256, 14, 1024, 256),
+    ("l3.conv3", 256, 14, 256, 1024),
+    ("l4.conv3", 256, 7, 512, 2048),
+]
+
+
+def main():
+    ext = require_extension()
+    tag = "OFF(128^2)" if os.environ.get("AL_DISABLE_GEMM256") == "1" else "ON(256^2)"
+    print(f"== gemm256 {tag} ==")
+    tot_f = tot_b = 0.0
+    for name, n, hw, c, k in SHAPES:
+        x = torch.randn(n, hw, hw, c, device="cuda").to(torch.bfloat16)
+        w = (torch.randn(k, 1, 1, c, device="cuda") * 0.05).to(torch.bfloat16)
+        flops = 2.0 * n * hw * hw * c * k
+        tf = timeit(lambda: ext.conv2d_fwd(x, w, 1, 0))
+        dy = torch.randn(n, hw, hw, k, device="cuda").to(torch.bfloat16)
+        wt = w.permute(3, 1, 2, 0).contiguous()
+        tb = timeit(lambda: ext.conv2d_bwd_data(dy, wt, 1, 0, hw, hw))
+        tot_f += tf
+        tot_b += tb
+        print(f"{name:10s} fwd {tf*1e3:7.3f} ms {flops/tf/1e12:7.1f} TF   "
+              f"bwd {tb*1e3:7.3f} ms {flops/tb/1e12:7.1f} TF")
+    print(f"TOTAL fwd {tot_f*1e3:.3f} ms  bwd {tot_b*1e3:.3f} ms")
+
+
+if __name__ == "__main__":
+    main()
