@@ -465,28 +465,36 @@ __global__ void igemm_kernel(const bf16* __restrict__ A, const bf16* __restrict_
 }
 
 // ---------------------------------------------------------------------------
-// 256x256 deep-pipelined NT GEMM for 1x1 stride-1 convolutions (and their
-// bwd-data): D[M, Nout] = A[M, KD] @ B[Nout, KD]^T. A 1x1 s1 p0 conv needs no
-// im2col gather — A is the raw NHWC activation (fwd: x, KD=C) or gradient
-// (bwd: dY, KD=K), B the (Nout, KD) weight rows already used by the igemm —
-// so the guide's 8-phase 256^2 schedule applies directly: 8 waves (512
-// threads), BK=64, 128 KiB LDS as a ring of 4 half-tiles (128 rows) per
-// operand, one half-tile refilled per phase into the slot freed by the
-// previous phase, vmcnt(4) once per K-tile, no vmcnt(0) drain in the loop.
-// Phase = one C-quadrant (qm,qn) x K=64: 12 ds_read_b128 + 2 glds + 16 MFMA.
+// 256x256 deep-pipelined conv GEMM: D[M, Nout] = A[M, KD] @ B[Nout, KD]^T.
+// The guide's 8-phase 256^2 schedule: 8 waves (512 threads), BK=64, 128 KiB
+// LDS as a ring of 4 half-tiles (128 rows) per operand, one half-tile
+// refilled per phase into the slot freed by the previous phase, vmcnt(4)
+// once per K-tile, no vmcnt(0) drain in the loop. Phase = one C-quadrant
+// (qm,qn) x K=64: 12 ds_read_b128 + 2 glds + 16 MFMA.
+//
+// MODE_PURE: A is a plain (M, KD) row-major matrix — 1x1 s1 p0 convs (fwd:
+// x with KD=C; bwd-data: dY with KD=K) need no im2col gather at all.
+// MODE_FWD / MODE_BWD_DATA: A rows are im2col/scatter gathers exactly as in
+// igemm_kernel, with the per-slot carry state (fixed GEMM row per slot, only
+// the contraction coordinate advances — no divisions in the K-loop).
 // ---------------------------------------------------------------------------
 
+#define MODE_PURE 3
+
+template <int MODE>
 __launch_bounds__(512, 1)
 __global__ void gemm256_nt_kernel(const bf16* __restrict__ A,
                                   const bf16* __restrict__ B,
                                   bf16* __restrict__ out,
                                   const bf16* __restrict__ zero,
-                                  long M, int Nout, int KD, int grid_m,
+                                  ConvShape sh, int grid_m,
                                   const float* __restrict__ epi_scale,
                                   const float* __restrict__ epi_shift,
                                   const bf16* __restrict__ epi_res, int epi_relu,
                                   float* __restrict__ stat_sum,
                                   float* __restrict__ stat_sumsq) {
+  const long M = sh.M;
+  const int Nout = sh.Nout, KD = sh.KD;
   constexpr int BM = 256, BN = 256, BK = 64;
   constexpr int HALF = 128 * BK;        // bf16 elements per half-tile (16 KiB)
   // XCD-aware bijective block remap (T1)
@@ -514,19 +522,115 @@ __global__ void gemm256_nt_kernel(const bf16* __restrict__ A,
   const int l15 = lane & 15, l4 = lane >> 4;
   const int KT = KD / BK;
 
+  // ---- per-slot A gather state (gather modes): each thread stages 2 chunks
+  // per A half-tile refill; the GEMM row m of slot (h, li) is FIXED, so its
+  // (n, spatial) decomposition is computed once and only the contraction
+  // coordinate (r, s, cf) advances by +BK per refill — same carry scheme as
+  // igemm_kernel, four independent slot states.
+  const int fastC = (MODE == MODE_FWD) ? sh.C : sh.K;
+  int ga_r[4], ga_s[4], ga_cf[4];
+  long ga_pix[4];
+  int ga_p[4], ga_q[4];
+  bool ga_ok[4];
+  if (MODE != MODE_PURE) {
+#pragma unroll
+    for (int sl = 0; sl < 4; ++sl) {
+      const int h = sl >> 1, li = sl & 1;
+      const int dc = li * 512 + wid * 64 + lane;
+      const int drow = dc >> 3, du = dc & 7;
+      const int su = du ^ (drow & 7);
+      const int kd0 = su * 8;           // kd0 < 64 <= fastC
+      ga_cf[sl] = kd0 % fastC;
+      const int rs = kd0 / fastC;
+      ga_s[sl] = rs % sh.S;
+      ga_r[sl] = rs / sh.S;
+      const long m = m0 + h * 128 + drow;
+      ga_ok[sl] = m < M;
+      if (ga_ok[sl]) {
+        if (MODE == MODE_FWD) {
+          const int q = (int)(m % sh.Q);
+          long tt = m / sh.Q;
+          const int p = (int)(tt % sh.P);
+          const int n = (int)(tt / sh.P);
+          ga_pix[sl] = (long)n * sh.H * sh.W;
+          ga_p[sl] = p * sh.stride - sh.pad;
+          ga_q[sl] = q * sh.stride - sh.pad;
+        } else {
+          const int w = (int)(m % sh.W);
+          long tt = m / sh.W;
+          const int h_ = (int)(tt % sh.H);
+          const int n = (int)(tt / sh.H);
+          ga_pix[sl] = (long)n * sh.P * sh.Q;
+          ga_p[sl] = h_ + sh.pad;
+          ga_q[sl] = w + sh.pad;
+        }
+      } else {
+        ga_pix[sl] = 0; ga_p[sl] = 0; ga_q[sl] = 0;
+      }
+    }
+  }
+  auto ga_src = [&](int sl) -> const bf16* {
+    if (!ga_ok[sl]) return zero;
+    if (MODE == MODE_FWD) {
+      const int h = ga_p[sl] + ga_r[sl];
+      const int w = ga_q[sl] + ga_s[sl];
+      if ((unsigned)h >= (unsigned)sh.H || (unsigned)w >= (unsigned)sh.W) return zero;
+      return A + (ga_pix[sl] + (long)h * sh.W + w) * sh.C + ga_cf[sl];
+    } else {
+      const int hp = ga_p[sl] - ga_r[sl];
+      const int wp = ga_q[sl] - ga_s[sl];
+      if (hp < 0 || wp < 0) return zero;
+      if (hp % sh.stride || wp % sh.stride) return zero;
+      const int p = hp / sh.stride, q = wp / sh.stride;
+      if (p >= sh.P || q >= sh.Q) return zero;
+      return A + (ga_pix[sl] + (long)p * sh.Q + q) * sh.K + ga_cf[sl];
+    }
+  };
+  auto ga_advance = [&](int h) {      // +BK for the two slots of half h
+#pragma unroll
+    for (int li = 0; li < 2; ++li) {
+      const int sl = h * 2 + li;
+      ga_cf[sl] += BK;
+      while (ga_cf[sl] >= fastC) {
+        ga_cf[sl] -= fastC;
+        if (++ga_s[sl] == sh.S) { ga_s[sl] = 0; ++ga_r[sl]; }
+      }
+    }
+  };
+
   // ---- half-tile staging: chunk dc = load*512 + wid*64 + lane lands at LDS
   // byte dc*16 of the slot; the bank-conflict swizzle (slot ^= row&7, as in
   // igemm_kernel) is applied by remapping which SOURCE chunk each lane loads.
-  auto stage_half = [&](const bf16* base, long ld, long row0, long rows_max,
-                        int ktile, bf16* slot) {
+  // A-halves of gather modes read through the slot state (advance first!).
+  auto stage_a = [&](int h, int ktile, bf16* slot) {
+#pragma unroll
+    for (int li = 0; li < 2; ++li) {
+      const bf16* src;
+      if (MODE == MODE_PURE) {
+        const int dc = li * 512 + wid * 64 + lane;
+        const int drow = dc >> 3, du = dc & 7;
+        const int su = du ^ (drow & 7);
+        const long row = m0 + h * 128 + drow;
+        src = (ktile < KT && row < M) ? A + row * KD + (long)ktile * BK + su * 8
+                                      : zero;
+      } else {
+        src = (ktile < KT) ? ga_src(h * 2 + li) : zero;
+      }
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)src,
+          (__attribute__((address_space(3))) void*)(slot + li * 512 * 8 + wid * 512),
+          16, 0, 0);
+    }
+  };
+  auto stage_b = [&](int h, int ktile, bf16* slot) {
 #pragma unroll
     for (int li = 0; li < 2; ++li) {
       const int dc = li * 512 + wid * 64 + lane;
       const int drow = dc >> 3, du = dc & 7;
       const int su = du ^ (drow & 7);
-      const long row = row0 + drow;
-      const bf16* src = (ktile < KT && row < rows_max)
-                            ? base + row * ld + (long)ktile * BK + su * 8
+      const long row = n0 + h * 128 + drow;
+      const bf16* src = (ktile < KT && row < Nout)
+                            ? B + row * KD + (long)ktile * BK + su * 8
                             : zero;
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) void*)src,
@@ -542,10 +646,16 @@ __global__ void gemm256_nt_kernel(const bf16* __restrict__ A,
   auto refill = [&](int g, int ph) {
     const int buf = g & 1;
     switch (ph) {
-      case 0: stage_half(A, KD, m0 + 128, M, g + 1, As + ((buf ^ 1) * 2 + 1) * HALF); break;
-      case 1: stage_half(B, KD, n0 + 128, Nout, g + 1, Bs + ((buf ^ 1) * 2 + 1) * HALF); break;
-      case 2: stage_half(A, KD, m0, M, g + 2, As + (buf * 2) * HALF); break;
-      case 3: stage_half(B, KD, n0, Nout, g + 2, Bs + (buf * 2) * HALF); break;
+      case 0:
+        if (MODE != MODE_PURE) ga_advance(1);
+        stage_a(1, g + 1, As + ((buf ^ 1) * 2 + 1) * HALF);
+        break;
+      case 1: stage_b(1, g + 1, Bs + ((buf ^ 1) * 2 + 1) * HALF); break;
+      case 2:
+        if (MODE != MODE_PURE) ga_advance(0);
+        stage_a(0, g + 2, As + (buf * 2) * HALF);
+        break;
+      case 3: stage_b(0, g + 2, Bs + (buf * 2) * HALF); break;
     }
   };
 
@@ -561,13 +671,15 @@ __global__ void gemm256_nt_kernel(const bf16* __restrict__ A,
 
   // prologue: K-tile 0 fully + K-tile 1's h0 halves (6 half-tiles); K-tile
   // 1's h1 halves arrive via group 0's ph0/ph1 refills. Order matters for
-  // the vmcnt(4) accounting below.
-  stage_half(A, KD, m0, M, 0, As);
-  stage_half(B, KD, n0, Nout, 0, Bs);
-  stage_half(A, KD, m0 + 128, M, 0, As + HALF);
-  stage_half(B, KD, n0 + 128, Nout, 0, Bs + HALF);
-  stage_half(A, KD, m0, M, 1, As + 2 * HALF);
-  stage_half(B, KD, n0, Nout, 1, Bs + 2 * HALF);
+  // the vmcnt(4) accounting below. Gather-state after prologue: h0 @ kt1,
+  // h1 @ kt0 — each subsequent refill advances its half by one K-tile.
+  stage_a(0, 0, As);
+  stage_b(0, 0, Bs);
+  stage_a(1, 0, As + HALF);
+  stage_b(1, 0, Bs + HALF);
+  if (MODE != MODE_PURE) ga_advance(0);
+  stage_a(0, 1, As + 2 * HALF);
+  stage_b(0, 1, Bs + 2 * HALF);
 
   for (int g = 0; g < KT; ++g) {
     const int buf = g & 1;
@@ -813,30 +925,48 @@ extern "C" void al_conv2d_mm(int mode, const void* A, const void* B, void* out,
     }
   }
 
-  // 1x1 stride-1 convs are plain NT GEMMs: route big ones to the 8-phase
-  // 256^2 kernel (fwd and bwd-data share it — B is (Nout, KD) in both)
+  // 8-phase 256^2 route. 1x1 s1 p0 convs are plain NT GEMMs (MODE_PURE);
+  // larger-filter s1 convs use the gathered variant when the grid still
+  // fills the 256 CUs (1 block/CU at 128 KiB LDS: require >= min_grid
+  // workgroups or the deep pipeline loses to the 128^2 tile's occupancy).
   static int gemm256_on = -1;
+  static int min_grid = 192;
   if (gemm256_on < 0) {
     const char* e = getenv("AL_DISABLE_GEMM256");
     gemm256_on = (e && e[0] == '1') ? 0 : 1;
+    const char* g = getenv("AL_GEMM256_MIN_GRID");
+    if (g) min_grid = atoi(g);
   }
-  if (gemm256_on && (mode == MODE_FWD || mode == MODE_BWD_DATA) && R == 1 &&
-      S == 1 && stride == 1 && pad == 0 && sh.Nout % 256 == 0 &&
-      sh.KD % 64 == 0 && sh.M >= 128) {
-    static bool attr_set = false;
-    if (!attr_set) {
-      (void)hipFuncSetAttribute((const void*)gemm256_nt_kernel,
-                          hipFuncAttributeMaxDynamicSharedMemorySize, 131072);
-      attr_set = true;
-    }
+  if (gemm256_on && (mode == MODE_FWD || mode == MODE_BWD_DATA) &&
+      stride == 1 && sh.Nout % 256 == 0 && sh.KD % 64 == 0 && sh.M >= 128) {
+    const bool pure = (R == 1 && S == 1 && pad == 0);
+    const int fast = (mode == MODE_FWD) ? sh.C : sh.K;
     const int grid_m = (int)((sh.M + 255) / 256);
     const int grid_n = sh.Nout / 256;
-    hipLaunchKernelGGL(gemm256_nt_kernel, dim3(grid_m * grid_n), dim3(512),
-                       131072, stream, (const bf16*)A, (const bf16*)B,
-                       (bf16*)out, (const bf16*)zero_page, sh.M, sh.Nout, sh.KD,
-                       grid_m, epi_scale, epi_shift, (const bf16*)epi_res,
-                       epi_relu, stat_sum, stat_sumsq);
-    return;
+    const bool gather_ok = !pure && fast % 8 == 0 &&
+                           grid_m * grid_n >= min_grid;
+    if (pure || gather_ok) {
+      static bool attr_set = false;
+      if (!attr_set) {
+        (void)hipFuncSetAttribute((const void*)gemm256_nt_kernel<MODE_PURE>,
+                                  hipFuncAttributeMaxDynamicSharedMemorySize, 131072);
+        (void)hipFuncSetAttribute((const void*)gemm256_nt_kernel<MODE_FWD>,
+                                  hipFuncAttributeMaxDynamicSharedMemorySize, 131072);
+        (void)hipFuncSetAttribute((const void*)gemm256_nt_kernel<MODE_BWD_DATA>,
+                                  hipFuncAttributeMaxDynamicSharedMemorySize, 131072);
+        attr_set = true;
+      }
+      dim3 grid(grid_m * grid_n), block(512);
+#define LAUNCH256(MODE_) hipLaunchKernelGGL((gemm256_nt_kernel<MODE_>), grid, \
+      block, 131072, stream, (const bf16*)A, (const bf16*)B, (bf16*)out, \
+      (const bf16*)zero_page, sh, grid_m, epi_scale, epi_shift, \
+      (const bf16*)epi_res, epi_relu, stat_sum, stat_sumsq)
+      if (pure) LAUNCH256(MODE_PURE);
+      else if (mode == MODE_FWD) LAUNCH256(MODE_FWD);
+      else LAUNCH256(MODE_BWD_DATA);
+#undef LAUNCH256
+      return;
+    }
   }
 
   if (igemm_ok(mode, sh)) {

@@ -44,6 +44,9 @@ CONV_CASES = [
     (1, 16, 16, 128, 256, 1, 1, 0),   # KT=2 edge, M=256 exact
     (2, 14, 14, 256, 512, 1, 1, 0),   # M=392 tail, KT=4
     (2, 16, 16, 512, 64, 1, 1, 0),    # bwd-data routes (Nout=C=512), fwd narrow
+    # gathered 256^2 route (grid >= 192 workgroups):
+    (8, 56, 56, 64, 512, 3, 1, 1),    # fwd gather (grid 98x2)
+    (16, 56, 56, 256, 64, 3, 1, 1),   # bwd-data gather (Nout=256, grid 196x1)
 ]
 
 
@@ -314,7 +317,7 @@ def test_gemm256_fused_epilogue():
     scale = torch.rand(k, device="cuda") + 0.5
     shift = torch.randn(k, device="cuda") * 0.1
     res = torch.randn(n, h, w_, k, device="cuda").to(torch.bfloat16)
-    y = EXT.conv2d_fwd_fused(x, w, 1, 0, scale, shift, res, True)
+    y = EXT.conv2d_fwd_fused(x, w, 1, 0, scale, shift, True, res)
     base = EXT.conv2d_fwd(x, w, 1, 0).float()
     ref = (base * scale + shift + res.float()).clamp_min(0)
     assert relerr(y, ref) < 0.02

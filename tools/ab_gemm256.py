@@ -27,12 +27,14 @@ def timeit(fn, iters=30, warmup=8):
     return (time.perf_counter() - t0) / iters
 
 
-# R50 @ B=256 1x1 stride-1 shapes: (N, HW, C_in, K_out)
+# R50 @ B=256 stride-1 shapes: (name, N, HW, C_in, K_out, R)
 SHAPES = [
-    ("l2.conv3", 256, 28, 128, 512),
-    ("l3.conv1", 256, 14, 1024, 256),
-    ("l3.conv3", 256, 14, 256, 1024),
-    ("l4.conv3", 256, 7, 512, 2048),
+    ("l2.conv3", 256, 28, 128, 512, 1),
+    ("l3.conv1", 256, 14, 1024, 256, 1),
+    ("l3.conv3", 256, 14, 256, 1024, 1),
+    ("l4.conv3", 256, 7, 512, 2048, 1),
+    ("l3.conv2", 256, 14, 256, 256, 3),   # gathered 256^2 (grid 196)
+    ("l4.conv2", 256, 7, 512, 512, 3),    # grid 98 < 192: stays 128^2
 ]
 
 
@@ -41,14 +43,15 @@ def main():
     tag = "OFF(128^2)" if os.environ.get("AL_DISABLE_GEMM256") == "1" else "ON(256^2)"
     print(f"== gemm256 {tag} ==")
     tot_f = tot_b = 0.0
-    for name, n, hw, c, k in SHAPES:
+    for name, n, hw, c, k, r in SHAPES:
+        pad = r // 2
         x = torch.randn(n, hw, hw, c, device="cuda").to(torch.bfloat16)
-        w = (torch.randn(k, 1, 1, c, device="cuda") * 0.05).to(torch.bfloat16)
-        flops = 2.0 * n * hw * hw * c * k
-        tf = timeit(lambda: ext.conv2d_fwd(x, w, 1, 0))
+        w = (torch.randn(k, r, r, c, device="cuda") * 0.05).to(torch.bfloat16)
+        flops = 2.0 * n * hw * hw * c * k * r * r
+        tf = timeit(lambda: ext.conv2d_fwd(x, w, 1, pad))
         dy = torch.randn(n, hw, hw, k, device="cuda").to(torch.bfloat16)
         wt = w.permute(3, 1, 2, 0).contiguous()
-        tb = timeit(lambda: ext.conv2d_bwd_data(dy, wt, 1, 0, hw, hw))
+        tb = timeit(lambda: ext.conv2d_bwd_data(dy, wt, 1, pad, hw, hw))
         tot_f += tf
         tot_b += tb
         print(f"{name:10s} fwd {tf*1e3:7.3f} ms {flops/tf/1e12:7.1f} TF   "
