@@ -18,7 +18,8 @@ void al_conv2d_mm(int mode, const void* A, const void* B, void* out,
                   int S, int P, int Q, int stride, int pad, const float* epi_scale,
                   const float* epi_shift, const void* epi_res, int epi_relu,
                   float* stat_sum, float* stat_sumsq, void* stream);
-void al_conv2d_wgrad(const void* dy, const void* x, float* dw, int N, int H, int W,
+void al_conv2d_wgrad(const void* dy, const void* x, float* dw,
+                     const void* zero_page, int N, int H, int W,
                      int C, int K, int R, int S, int P, int Q, int stride, int pad,
                      void* stream);
 void al_bn_stats(const void* x, float* sum, float* sumsq, long rows, int C,
@@ -167,7 +168,8 @@ Tensor conv2d_bwd_weight(const Tensor& dy, const Tensor& x, long R, long S,
   const int P = dy.size(1), Q = dy.size(2), K = dy.size(3);
   auto dw = torch::zeros({(long)K, R, S, (long)C},
                          x.options().dtype(torch::kFloat32));
-  al_conv2d_wgrad(dy.data_ptr(), x.data_ptr(), dw.data_ptr<float>(), N, H, W, C, K,
+  al_conv2d_wgrad(dy.data_ptr(), x.data_ptr(), dw.data_ptr<float>(),
+                  zero_page(x).data_ptr(), N, H, W, C, K,
                   (int)R, (int)S, P, Q, (int)stride, (int)pad, cur_stream());
   return dw;
 }

@@ -219,6 +219,199 @@ __global__ void wgrad_kernel(const bf16* __restrict__ dy, const bf16* __restrict
   }
 }
 
+// ---------------------------------------------------------------------------
+// wgrad v2: row-major staging + hardware transpose reads.
+//
+// v1 stages [ch][m] tiles via an in-register 8x8 bf16 transpose (3 shfl
+// butterflies + v_perm per chunk) so fragments can ds_read_b128. gfx950's
+// ds_read_b64_tr_b16 makes that transpose free: stage both operands ROW-major
+// ([l][ch] — every 16B global gather is contiguous, issued as
+// global_load_lds with no VGPR round trip or ds_write pass) and read MFMA
+// fragments with the transpose load. Probe-verified semantics
+// (tools/tr_probe.hip): within a [4][16]-u16 128B-aligned block, lane l
+// receives rows 4*(l>>4)+0..3 of column l&15 — exactly the A/B fragment
+// gather (lane's own 8B-aligned address selects block and column).
+//
+// LDS per operand per buffer: [width/16 groups][64 l][16 ch] bf16 (row-major
+// within each 16-channel group — the conflict-free tr_b16 subtiling).
+// dy chunks advance along L with a pointer bump; x chunks keep per-slot
+// (n,p,q) pixel state advanced by precomputed (dn,dp,dq) carries.
+// ---------------------------------------------------------------------------
+
+template <int GKR, int GNC>
+__launch_bounds__(256)
+__global__ void wgrad2_kernel(const bf16* __restrict__ dy,
+                              const bf16* __restrict__ x,
+                              float* __restrict__ dw,
+                              const bf16* __restrict__ zero,
+                              WgradShape sh, int grid_k, long l_per_z) {
+  constexpr int BMK = GKR * 64, BNW = GNC * 64, BL = 64;
+  constexpr int ACH = BMK * BL;            // elements per A buffer
+  constexpr int BCH = BNW * BL;
+  constexpr int NA = ACH / (256 * 8);      // A 16B chunks per thread
+  constexpr int NB = BCH / (256 * 8);
+  const int bk = blockIdx.x % grid_k;
+  const int bn = blockIdx.x / grid_k;
+  const int k0 = bk * BMK;
+  const int n0 = bn * BNW;
+  const long lz0 = (long)blockIdx.y * l_per_z;
+  const long lz1 = min(sh.L, lz0 + l_per_z);
+  if (lz0 >= lz1) return;
+
+  __shared__ __attribute__((aligned(16))) bf16 As2[2][ACH];
+  __shared__ __attribute__((aligned(16))) bf16 Bs2[2][BCH];
+
+  const int tid = threadIdx.x;
+  const int wid = tid >> 6, lane = tid & 63;
+  const int wr = wid / GNC, wc = wid % GNC;
+  const int l15 = lane & 15, l4 = lane >> 4;
+
+  // ---- per-slot staging state. Chunk ci = slot*256 + tid maps to
+  // (group g, l-row, half h): g = ci>>7, l = (ci&127)>>1, h = ci&1; dest
+  // element ci*8 (lane-linear for global_load_lds).
+  // A (dy): source = dy + (lz0+l)*K + k0 + g*16 + h*8 — pointer bump only.
+  long a_off[NA];
+  int a_l[NA];
+  bool a_chok[NA];
+#pragma unroll
+  for (int i = 0; i < NA; ++i) {
+    const int ci = i * 256 + tid;
+    const int g = ci >> 7, l = (ci & 127) >> 1, h = ci & 1;
+    const int ch = k0 + g * 16 + h * 8;
+    a_l[i] = l;
+    a_chok[i] = ch + 8 <= sh.K;
+    a_off[i] = (lz0 + l) * (long)sh.K + ch;
+  }
+  // B (x gather): nw = n0 + g*16 + h*8 -> fixed (r,s,c); pixel (n,p,q) from
+  // l = lz0 + l_fix, advanced by BL each step with precomputed carries.
+  int b_n[NB], b_p[NB], b_q[NB], b_ho[NB], b_wo[NB];
+  long b_coff[NB];
+  int b_l[NB];
+  bool b_chok[NB];
+  const int dn = (int)(BL / ((long)sh.P * sh.Q));
+  const int rem = (int)(BL % ((long)sh.P * sh.Q));
+  const int dp = rem / sh.Q, dq = rem % sh.Q;
+#pragma unroll
+  for (int i = 0; i < NB; ++i) {
+    const int ci = i * 256 + tid;
+    const int g = ci >> 7, l = (ci & 127) >> 1, h = ci & 1;
+    const int nw = n0 + g * 16 + h * 8;
+    b_l[i] = l;
+    b_chok[i] = nw + 8 <= sh.Nw;
+    const int c = nw % sh.C;
+    const int rs = nw / sh.C;
+    b_ho[i] = (rs / sh.S) - sh.pad;         // r - pad
+    b_wo[i] = (rs % sh.S) - sh.pad;         // s - pad
+    b_coff[i] = c;
+    const long m = lz0 + l;
+    b_q[i] = (int)(m % sh.Q);
+    long t = m / sh.Q;
+    b_p[i] = (int)(t % sh.P);
+    b_n[i] = (int)(t / sh.P);
+  }
+  auto b_advance = [&]() {
+#pragma unroll
+    for (int i = 0; i < NB; ++i) {
+      b_q[i] += dq;
+      if (b_q[i] >= sh.Q) { b_q[i] -= sh.Q; ++b_p[i]; }
+      b_p[i] += dp;
+      if (b_p[i] >= sh.P) { b_p[i] -= sh.P; ++b_n[i]; }
+      b_n[i] += dn;
+    }
+  };
+
+  auto stage = [&](int buf, long l0) {
+#pragma unroll
+    for (int i = 0; i < NA; ++i) {
+      const bf16* src = (a_chok[i] && l0 + a_l[i] < lz1) ? dy + a_off[i] : zero;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)src,
+          (__attribute__((address_space(3))) void*)(&As2[buf][(i * 256 + wid * 64) * 8]),
+          16, 0, 0);
+      a_off[i] += (long)BL * sh.K;
+    }
+#pragma unroll
+    for (int i = 0; i < NB; ++i) {
+      const int hh = b_p[i] * sh.stride + b_ho[i];
+      const int ww = b_q[i] * sh.stride + b_wo[i];
+      const bool ok = b_chok[i] && (l0 + b_l[i] < lz1) &&
+                      (unsigned)hh < (unsigned)sh.H && (unsigned)ww < (unsigned)sh.W;
+      const bf16* src = ok
+          ? x + (((long)b_n[i] * sh.H + hh) * sh.W + ww) * sh.C + b_coff[i]
+          : zero;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)src,
+          (__attribute__((address_space(3))) void*)(&Bs2[buf][(i * 256 + wid * 64) * 8]),
+          16, 0, 0);
+    }
+    b_advance();
+  };
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  // tr_b16 fragment read: two b64 transpose loads give lane l column l&15,
+  // l-rows (l>>4)*8 + 0..7 of a [64][16] group image — the MFMA operand.
+  auto trfrag = [&](const bf16* img, int group, int mc) -> bf16x8 {
+    const unsigned base = (unsigned)(unsigned long long)img +
+        (unsigned)((group * 1024 + mc * 32 * 16 + l4 * 8 * 16 + l15 * 4) * 2);
+    unsigned long long v0, v1;
+    asm volatile("ds_read_b64_tr_b16 %0, %2 offset:0\n\t"
+                 "ds_read_b64_tr_b16 %1, %2 offset:128"
+                 : "=v"(v0), "=v"(v1) : "v"(base));
+    union { unsigned long long q[2]; bf16x8 v; } u;
+    u.q[0] = v0;
+    u.q[1] = v1;
+    return u.v;
+  };
+
+  auto compute = [&](int buf) {
+#pragma unroll
+    for (int mc = 0; mc < 2; ++mc) {
+      bf16x8 afrag[4], bfrag[4];
+#pragma unroll
+      for (int f = 0; f < 4; ++f) {
+        afrag[f] = trfrag(As2[buf], wr * 4 + f, mc);
+        bfrag[f] = trfrag(Bs2[buf], wc * 4 + f, mc);
+      }
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag[mi], bfrag[ni], acc[mi][ni], 0, 0, 0);
+    }
+  };
+
+  stage(0, lz0);
+  int buf = 0;
+  for (long m0 = lz0; m0 < lz1; m0 += BL) {
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+    if (m0 + BL < lz1) stage(buf ^ 1, m0 + BL);
+    compute(buf);
+    buf ^= 1;
+  }
+
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi) {
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni) {
+      const int col = n0 + wc * 64 + ni * 16 + l15;
+      if (col >= sh.Nw) continue;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = k0 + wr * 64 + mi * 16 + l4 * 4 + r;
+        if (row < sh.K) atomicAdd(&dw[(long)row * sh.Nw + col], acc[mi][ni][r]);
+      }
+    }
+  }
+}
+
 // direct fallback: one dW element per thread, strided over L (safety net;
 // normal stems go through the packed-im2col MFMA path)
 __global__ void wgrad_direct_kernel(const bf16* __restrict__ dy,
@@ -243,7 +436,8 @@ __global__ void wgrad_direct_kernel(const bf16* __restrict__ dy,
   }
 }
 
-extern "C" void al_conv2d_wgrad(const void* dy, const void* x, float* dw, int N, int H,
+extern "C" void al_conv2d_wgrad(const void* dy, const void* x, float* dw,
+                                const void* zero_page, int N, int H,
                                 int W, int C, int K, int R, int S, int P, int Q,
                                 int stride, int pad, hipStream_t stream) {
   WgradShape sh;
@@ -274,7 +468,21 @@ extern "C" void al_conv2d_wgrad(const void* dy, const void* x, float* dw, int N,
     l_per_z = ((l_per_z + 63) / 64) * 64;
     z = (int)((sh.L + l_per_z - 1) / l_per_z);
     dim3 grid(tiles, z), block(256);
-    if (narrow_k)
+    static int v2 = -1;
+    if (v2 < 0) {
+      const char* e = getenv("AL_WGRAD_V2");
+      v2 = (e && e[0] == '0') ? 0 : 1;    // tr_b16 staging path (default)
+    }
+    if (v2) {
+      if (narrow_k)
+        hipLaunchKernelGGL((wgrad2_kernel<1, 4>), grid, block, 0, stream,
+                           (const bf16*)dy, (const bf16*)x, dw,
+                           (const bf16*)zero_page, sh, grid_k, l_per_z);
+      else
+        hipLaunchKernelGGL((wgrad2_kernel<2, 2>), grid, block, 0, stream,
+                           (const bf16*)dy, (const bf16*)x, dw,
+                           (const bf16*)zero_page, sh, grid_k, l_per_z);
+    } else if (narrow_k)
       hipLaunchKernelGGL((wgrad_kernel<1, 4>), grid, block, 0, stream,
                          (const bf16*)dy, (const bf16*)x, dw, sh, grid_k, l_per_z);
     else
