@@ -1029,7 +1029,9 @@ __global__ void im2col_pack_kernel(const bf16* __restrict__ x, bf16* __restrict_
     int r = rs / sh.S;
     s16x8 o[2] = {{0,0,0,0,0,0,0,0},{0,0,0,0,0,0,0,0}};
     short* os = (short*)o;
-#pragma unroll 1
+    // fully unrolled: the 16 gathers are independent — issue them all and
+    // let the carry chain (cheap VALU) overlap the loads
+#pragma unroll
     for (int j = 0; j < 16; ++j) {
       if (kd + j < sh.KD) {
         const int h = h0 + r, w = w0 + s;
