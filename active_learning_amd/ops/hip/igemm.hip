@@ -937,9 +937,14 @@ extern "C" void al_conv2d_mm(int mode, const void* A, const void* B, void* out,
     const char* g = getenv("AL_GEMM256_MIN_GRID");
     if (g) min_grid = atoi(g);
   }
+  // pure route: 1x1 s1 p0 only (contiguous rows). gather route: any stride
+  // for fwd (the per-slot state uses p*stride - pad directly — this picks up
+  // the 1x1-s2 downsamples and the stride-2 3x3s); bwd-data keeps stride 1
+  // (stride-2 bwd goes through the parity-class decomposition above).
   if (gemm256_on && (mode == MODE_FWD || mode == MODE_BWD_DATA) &&
-      stride == 1 && sh.Nout % 256 == 0 && sh.KD % 64 == 0 && sh.M >= 128) {
-    const bool pure = (R == 1 && S == 1 && pad == 0);
+      (mode == MODE_FWD || stride == 1) &&
+      sh.Nout % 256 == 0 && sh.KD % 64 == 0 && sh.M >= 128) {
+    const bool pure = (R == 1 && S == 1 && stride == 1 && pad == 0);
     const int fast = (mode == MODE_FWD) ? sh.C : sh.K;
     const int grid_m = (int)((sh.M + 255) / 256);
     const int grid_n = sh.Nout / 256;

@@ -238,7 +238,7 @@ __global__ void wgrad_kernel(const bf16* __restrict__ dy, const bf16* __restrict
 // (n,p,q) pixel state advanced by precomputed (dn,dp,dq) carries.
 // ---------------------------------------------------------------------------
 
-template <int GKR, int GNC>
+template <int GKR, int GNC, bool SCAL = false>
 __launch_bounds__(256)
 __global__ void wgrad2_kernel(const bf16* __restrict__ dy,
                               const bf16* __restrict__ x,
@@ -355,16 +355,25 @@ __global__ void wgrad2_kernel(const bf16* __restrict__ dy,
 
   // tr_b16 fragment read: two b64 transpose loads give lane l column l&15,
   // l-rows (l>>4)*8 + 0..7 of a [64][16] group image — the MFMA operand.
+  static_assert(true, "");
+  auto scalfrag = [&](const bf16* img, int group, int mc) -> bf16x8 {
+    // diagnostic fallback (AL_WGRAD_SCAL=1): same gather via compiler-
+    // managed scalar LDS reads
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      o[j] = img[group * 1024 + (mc * 32 + l4 * 8 + j) * 16 + l15];
+    return o;
+  };
+  typedef __attribute__((address_space(3))) s16x4* lds_v4p;
   auto trfrag = [&](const bf16* img, int group, int mc) -> bf16x8 {
-    const unsigned base = (unsigned)(unsigned long long)img +
-        (unsigned)((group * 1024 + mc * 32 * 16 + l4 * 8 * 16 + l15 * 4) * 2);
-    unsigned long long v0, v1;
-    asm volatile("ds_read_b64_tr_b16 %0, %2 offset:0\n\t"
-                 "ds_read_b64_tr_b16 %1, %2 offset:128"
-                 : "=v"(v0), "=v"(v1) : "v"(base));
-    union { unsigned long long q[2]; bf16x8 v; } u;
-    u.q[0] = v0;
-    u.q[1] = v1;
+    // compiler-modeled transpose loads (proper lgkmcnt tracking and
+    // scheduling — a hand-rolled asm version suffered occupancy-dependent
+    // reordering hazards)
+    const bf16* p = img + group * 1024 + mc * 32 * 16 + l4 * 8 * 16 + l15 * 4;
+    union { s16x4 h[2]; bf16x8 v; } u;
+    u.h[0] = __builtin_amdgcn_ds_read_tr16_b64_v4i16((lds_v4p)p);
+    u.h[1] = __builtin_amdgcn_ds_read_tr16_b64_v4i16((lds_v4p)(p + 64));
     return u.v;
   };
 
@@ -372,12 +381,19 @@ __global__ void wgrad2_kernel(const bf16* __restrict__ dy,
 #pragma unroll
     for (int mc = 0; mc < 2; ++mc) {
       bf16x8 afrag[4], bfrag[4];
+      if (SCAL) {
 #pragma unroll
-      for (int f = 0; f < 4; ++f) {
-        afrag[f] = trfrag(As2[buf], wr * 4 + f, mc);
-        bfrag[f] = trfrag(Bs2[buf], wc * 4 + f, mc);
+        for (int f = 0; f < 4; ++f) {
+          afrag[f] = scalfrag(As2[buf], wr * 4 + f, mc);
+          bfrag[f] = scalfrag(Bs2[buf], wc * 4 + f, mc);
+        }
+      } else {
+#pragma unroll
+        for (int f = 0; f < 4; ++f) {
+          afrag[f] = trfrag(As2[buf], wr * 4 + f, mc);
+          bfrag[f] = trfrag(Bs2[buf], wc * 4 + f, mc);
+        }
       }
-      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
 #pragma unroll
       for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
@@ -474,7 +490,21 @@ extern "C" void al_conv2d_wgrad(const void* dy, const void* x, float* dw,
       v2 = (e && e[0] == '0') ? 0 : 1;    // tr_b16 staging path (default)
     }
     if (v2) {
-      if (narrow_k)
+      static int scal = -1;
+      if (scal < 0) {
+        const char* e = getenv("AL_WGRAD_SCAL");
+        scal = (e && e[0] == '1') ? 1 : 0;
+      }
+      if (scal) {
+        if (narrow_k)
+          hipLaunchKernelGGL((wgrad2_kernel<1, 4, true>), grid, block, 0, stream,
+                             (const bf16*)dy, (const bf16*)x, dw,
+                             (const bf16*)zero_page, sh, grid_k, l_per_z);
+        else
+          hipLaunchKernelGGL((wgrad2_kernel<2, 2, true>), grid, block, 0, stream,
+                             (const bf16*)dy, (const bf16*)x, dw,
+                             (const bf16*)zero_page, sh, grid_k, l_per_z);
+      } else if (narrow_k)
         hipLaunchKernelGGL((wgrad2_kernel<1, 4>), grid, block, 0, stream,
                            (const bf16*)dy, (const bf16*)x, dw,
                            (const bf16*)zero_page, sh, grid_k, l_per_z);

@@ -47,6 +47,8 @@ CONV_CASES = [
     # gathered 256^2 route (grid >= 192 workgroups):
     (8, 56, 56, 64, 512, 3, 1, 1),    # fwd gather (grid 98x2)
     (16, 56, 56, 256, 64, 3, 1, 1),   # bwd-data gather (Nout=256, grid 196x1)
+    (16, 56, 56, 128, 256, 1, 2, 0),  # 1x1 s2 fwd gather (downsample shape)
+    (8, 56, 56, 128, 256, 3, 2, 1),   # 3x3 s2 fwd gather
 ]
 
 
