@@ -17,6 +17,9 @@ GPU dispatch goes to active_learning_amd._C (HIP/gfx950); CPU fallback uses
 plain PyTorch and defines the semantics the kernels are tested against.
 """
 
+import os
+
+import numpy as np
 import torch
 import torch.distributed as dist
 import torch.nn.functional as F
@@ -106,6 +109,65 @@ def _gpu_conv_wgrad_packed(ext, dy, apack, w_shape):
     return dwpad.view(K, kdpad)[:, :KD].reshape(K, R, S, C).contiguous()
 
 
+class _GradArena:
+    """One flat fp32 buffer for every conv weight gradient.
+
+    conv2d_bwd_weight's split-K accumulation needs a zeroed output, which
+    cost ~50 small fill launches per training step. The arena hands out
+    views of a single flat buffer instead, zeroed ONCE at the start of each
+    backward (the first take() after any forward marked the step dirty).
+    Keys are (weight data_ptr, shape); unknown keys fall back to the plain
+    self-allocating path for that call and join the arena at the next
+    rebuild. NOTE: assumes the standard step discipline (one backward per
+    forward, optimizer consumes grads before the next forward) — gradient
+    accumulation across multiple backwards must disable it
+    (AL_GRAD_ARENA=0), see PARITY.md.
+    """
+
+    def __init__(self):
+        self.enabled = os.environ.get("AL_GRAD_ARENA", "1") == "1"
+        self.views = {}
+        self.shapes = {}
+        self.pending = []
+        self.flat = None
+        self.dirty = False
+
+    def mark_step(self):
+        self.dirty = True
+
+    def take(self, key, shape, device):
+        if not self.enabled:
+            return None
+        if self.dirty:
+            if self.pending:
+                self._rebuild(device)
+            elif self.flat is not None:
+                self.flat.zero_()
+            self.dirty = False
+        v = self.views.get(key)
+        if v is not None:
+            return v
+        if (key, tuple(shape)) not in [(k, s) for k, s in self.pending]:
+            self.pending.append((key, tuple(shape)))
+        return None
+
+    def _rebuild(self, device):
+        for key, shape in self.pending:
+            self.shapes[key] = shape
+        self.pending = []
+        total = sum(int(np.prod(s)) for s in self.shapes.values())
+        self.flat = torch.zeros(total, dtype=torch.float32, device=device)
+        off = 0
+        self.views = {}
+        for key, shape in self.shapes.items():
+            n = int(np.prod(shape))
+            self.views[key] = self.flat.narrow(0, off, n).view(shape)
+            off += n
+
+
+_grad_arena = _GradArena()
+
+
 class Conv2dNHWC(Function):
     """y[N,P,Q,K] = conv(x[N,H,W,C], w[K,R,S,C]; stride, pad), no bias
     (ResNet convs carry no bias; BN follows)."""
@@ -119,6 +181,7 @@ class Conv2dNHWC(Function):
         ctx.weight_dtype = weight.dtype
         if x.is_cuda:
             ext = require_extension()
+            _grad_arena.mark_step()
             K, R, S, C = w_c.shape
             if _igemm_eligible(C, R * S * C):
                 ctx.apack = None
@@ -153,7 +216,14 @@ class Conv2dNHWC(Function):
                                              ctx.padding, x.shape[1], x.shape[2])
             if ctx.needs_input_grad[1]:
                 if _igemm_eligible(C, R * S * C):
-                    dw = ext.conv2d_bwd_weight(dy, x, R, S, ctx.stride, ctx.padding)
+                    buf = _grad_arena.take((w_c.data_ptr(), w_c.shape), w_c.shape,
+                                           dy.device)
+                    if buf is not None:
+                        dw = ext.conv2d_bwd_weight_into(dy, x, R, S, ctx.stride,
+                                                        ctx.padding, buf)
+                    else:
+                        dw = ext.conv2d_bwd_weight(dy, x, R, S, ctx.stride,
+                                                   ctx.padding)
                 else:
                     dw = _gpu_conv_wgrad_packed(ext, dy, ctx.apack, w_c.shape)
         else:
@@ -190,6 +260,7 @@ class Conv2dNHWCStats(Function):
         ctx.weight_dtype = weight.dtype
         if x.is_cuda:
             ext = require_extension()
+            _grad_arena.mark_step()
             K, R, S, C = w_c.shape
             if _igemm_eligible(C, R * S * C):
                 ctx.apack = None

@@ -132,6 +132,7 @@ __global__ void bn_norm2_kernel(const bf16* __restrict__ x, bf16* __restrict__ y
   }
   const long row0 = (long)blockIdx.x * rows_per_block + row_lane;
   const long step = (long)gridDim.x * rows_per_block;
+  #pragma unroll 2
   for (long r = row0; r < rows; r += step) {
     s16x8 v = ((const s16x8*)x)[r * C8 + c8];
     s16x8 rv;
@@ -287,6 +288,7 @@ __global__ void bn_bwd2_kernel(const bf16* __restrict__ dy, const bf16* __restri
   }
   const long row0 = (long)blockIdx.x * rows_per_block + row_lane;
   const long step = (long)gridDim.x * rows_per_block;
+  #pragma unroll 2
   for (long r = row0; r < rows; r += step) {
     const long i = r * C8 + c8;
     s16x8 gv = ((const s16x8*)dy)[i];

@@ -347,3 +347,39 @@ def test_train_forward_with_stats_fusion_matches_cpu():
     rm_c = net_c.encoder.bn1.running_mean
     rm_g = net_g.encoder.bn1.running_mean
     assert relerr(rm_g, rm_c) < 0.05
+
+
+def test_grad_arena_matches_plain():
+    """Arena-backed wgrad output (flat pre-zeroed buffer, accumulate-into)
+    must equal the self-allocating path across several steps."""
+    from active_learning_amd.ops import functional as AFN
+
+    w = (torch.randn(256, 3, 3, 128) * 0.05).cuda().requires_grad_(True)
+    xs = [torch.randn(2, 14, 14, 128).cuda().to(torch.bfloat16) for _ in range(3)]
+    dys = None
+
+    def run(enabled):
+        AFN._grad_arena.enabled = enabled
+        AFN._grad_arena.views = {}
+        AFN._grad_arena.shapes = {}
+        AFN._grad_arena.pending = []
+        AFN._grad_arena.flat = None
+        AFN._grad_arena.dirty = False
+        grads = []
+        nonlocal dys
+        mk = dys is None
+        if mk:
+            dys = []
+        for i, x in enumerate(xs):
+            w.grad = None
+            y = AFN.conv2d(x, w, 1, 1)
+            if mk:
+                dys.append(torch.randn_like(y))
+            y.backward(dys[i])
+            grads.append(w.grad.detach().clone())
+        return grads
+
+    g_plain = run(False)
+    g_arena = run(True)
+    for a, b in zip(g_arena, g_plain):
+        assert torch.equal(a, b), "arena grad mismatch"
