@@ -481,7 +481,10 @@ __global__ void igemm_kernel(const bf16* __restrict__ A, const bf16* __restrict_
 
 #define MODE_PURE 3
 
-template <int MODE>
+// BNT: output-tile width (256, or 128 for Nout%256!=0 layers e.g. the
+// 128-wide layer2 convs). A-side geometry is unchanged; B half-tiles are
+// (BNT/2) rows and the per-wave quadrant slice narrows to BNT/8 columns.
+template <int MODE, int BNT = 256>
 __launch_bounds__(512, 1)
 __global__ void gemm256_nt_kernel(const bf16* __restrict__ A,
                                   const bf16* __restrict__ B,
@@ -495,8 +498,11 @@ __global__ void gemm256_nt_kernel(const bf16* __restrict__ A,
                                   float* __restrict__ stat_sumsq) {
   const long M = sh.M;
   const int Nout = sh.Nout, KD = sh.KD;
-  constexpr int BM = 256, BN = 256, BK = 64;
-  constexpr int HALF = 128 * BK;        // bf16 elements per half-tile (16 KiB)
+  constexpr int BM = 256, BN = BNT, BK = 64;
+  constexpr int HALF = 128 * BK;          // A half-tile elements (16 KiB)
+  constexpr int HALF_B = (BNT / 2) * BK;  // B half-tile elements
+  constexpr int NLB = (BNT == 256) ? 2 : 1;  // B glds per thread per half
+  constexpr int NF = BNT / 128;           // per-wave N fragments
   // XCD-aware bijective block remap (T1)
   const int nwg = gridDim.x;
   int bid = blockIdx.x;
@@ -514,7 +520,7 @@ __global__ void gemm256_nt_kernel(const bf16* __restrict__ A,
   extern __shared__ __attribute__((aligned(16))) char smem[];
   // slot layout: A half-tiles [buf][half] then B half-tiles [buf][half]
   bf16* As = (bf16*)smem;               // 4 x HALF
-  bf16* Bs = As + 4 * HALF;             // 4 x HALF
+  bf16* Bs = As + 4 * HALF;             // 4 x HALF_B
 
   const int tid = threadIdx.x;
   const int wid = tid >> 6, lane = tid & 63;
@@ -624,11 +630,11 @@ __global__ void gemm256_nt_kernel(const bf16* __restrict__ A,
   };
   auto stage_b = [&](int h, int ktile, bf16* slot) {
 #pragma unroll
-    for (int li = 0; li < 2; ++li) {
+    for (int li = 0; li < NLB; ++li) {
       const int dc = li * 512 + wid * 64 + lane;
       const int drow = dc >> 3, du = dc & 7;
       const int su = du ^ (drow & 7);
-      const long row = n0 + h * 128 + drow;
+      const long row = n0 + h * (BNT / 2) + drow;
       const bf16* src = (ktile < KT && row < Nout)
                             ? B + row * KD + (long)ktile * BK + su * 8
                             : zero;
@@ -650,16 +656,16 @@ __global__ void gemm256_nt_kernel(const bf16* __restrict__ A,
         if (MODE != MODE_PURE) ga_advance(1);
         stage_a(1, g + 1, As + ((buf ^ 1) * 2 + 1) * HALF);
         break;
-      case 1: stage_b(1, g + 1, Bs + ((buf ^ 1) * 2 + 1) * HALF); break;
+      case 1: stage_b(1, g + 1, Bs + ((buf ^ 1) * 2 + 1) * HALF_B); break;
       case 2:
         if (MODE != MODE_PURE) ga_advance(0);
         stage_a(0, g + 2, As + (buf * 2) * HALF);
         break;
-      case 3: stage_b(0, g + 2, Bs + (buf * 2) * HALF); break;
+      case 3: stage_b(0, g + 2, Bs + (buf * 2) * HALF_B); break;
     }
   };
 
-  f32x4 acc[2][2][4][2];
+  f32x4 acc[2][2][4][NF];
 #pragma unroll
   for (int i = 0; i < 2; ++i)
 #pragma unroll
@@ -667,7 +673,7 @@ __global__ void gemm256_nt_kernel(const bf16* __restrict__ A,
 #pragma unroll
       for (int g = 0; g < 4; ++g)
 #pragma unroll
-        for (int n = 0; n < 2; ++n) acc[i][j][g][n] = f32x4{0.f, 0.f, 0.f, 0.f};
+        for (int n = 0; n < NF; ++n) acc[i][j][g][n] = f32x4{0.f, 0.f, 0.f, 0.f};
 
   // prologue: K-tile 0 fully + K-tile 1's h0 halves (6 half-tiles); K-tile
   // 1's h1 halves arrive via group 0's ph0/ph1 refills. Order matters for
@@ -676,22 +682,24 @@ __global__ void gemm256_nt_kernel(const bf16* __restrict__ A,
   stage_a(0, 0, As);
   stage_b(0, 0, Bs);
   stage_a(1, 0, As + HALF);
-  stage_b(1, 0, Bs + HALF);
+  stage_b(1, 0, Bs + HALF_B);
   if (MODE != MODE_PURE) ga_advance(0);
   stage_a(0, 1, As + 2 * HALF);
-  stage_b(0, 1, Bs + 2 * HALF);
+  stage_b(0, 1, Bs + 2 * HALF_B);
 
   for (int g = 0; g < KT; ++g) {
     const int buf = g & 1;
     // once per K-tile: the 2 half-tiles issued since this tile's last half
-    // (4 loads/wave) may stay in flight — never drain to vmcnt(0) (T3+T4)
-    asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    // (one A refill = 2 loads + one B refill = NLB) may stay in flight —
+    // never drain to vmcnt(0) (T3+T4)
+    if (NLB == 2) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    else asm volatile("s_waitcnt vmcnt(3)" ::: "memory");
 #pragma unroll
     for (int ph = 0; ph < 4; ++ph) {
       const int qm = ph >> 1, qn = ph & 1;
       const bf16* ah = As + (buf * 2 + qm) * HALF;
-      const bf16* bh = Bs + (buf * 2 + qn) * HALF;
-      bf16x8 af[4][2], bf[2][2];
+      const bf16* bh = Bs + (buf * 2 + qn) * HALF_B;
+      bf16x8 af[4][2], bf[NF][2];
 #pragma unroll
       for (int fg = 0; fg < 4; ++fg) {
         const int rowh = wrq * 64 + fg * 16 + l15;
@@ -700,8 +708,8 @@ __global__ void gemm256_nt_kernel(const bf16* __restrict__ A,
           af[fg][kc] = *(const bf16x8*)(ah + rowh * BK + (((kc * 4 + l4) ^ (rowh & 7)) * 8));
       }
 #pragma unroll
-      for (int ng = 0; ng < 2; ++ng) {
-        const int colh = wcq * 32 + ng * 16 + l15;
+      for (int ng = 0; ng < NF; ++ng) {
+        const int colh = wcq * (BNT / 8) + ng * 16 + l15;
 #pragma unroll
         for (int kc = 0; kc < 2; ++kc)
           bf[ng][kc] = *(const bf16x8*)(bh + colh * BK + (((kc * 4 + l4) ^ (colh & 7)) * 8));
@@ -715,7 +723,7 @@ __global__ void gemm256_nt_kernel(const bf16* __restrict__ A,
 #pragma unroll
         for (int fg = 0; fg < 4; ++fg)
 #pragma unroll
-          for (int ng = 0; ng < 2; ++ng)
+          for (int ng = 0; ng < NF; ++ng)
             acc[qm][qn][fg][ng] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                 af[fg][kc], bf[ng][kc], acc[qm][qn][fg][ng], 0, 0, 0);
       __builtin_amdgcn_s_setprio(0);
@@ -735,11 +743,12 @@ __global__ void gemm256_nt_kernel(const bf16* __restrict__ A,
 #pragma unroll
       for (int fg = 0; fg < 4; ++fg)
 #pragma unroll
-        for (int ng = 0; ng < 2; ++ng)
+        for (int ng = 0; ng < NF; ++ng)
 #pragma unroll
           for (int r = 0; r < 4; ++r)
             etile[(qm * 128 + wrq * 64 + fg * 16 + l4 * 4 + r) * BN +
-                  qn * 128 + wcq * 32 + ng * 16 + l15] = f2bf(acc[qm][qn][fg][ng][r]);
+                  qn * (BNT / 2) + wcq * (BNT / 8) + ng * 16 + l15] =
+                f2bf(acc[qm][qn][fg][ng][r]);
   __syncthreads();
 
   constexpr int CPR = BN / 8;          // 32 16B chunks per row
@@ -752,7 +761,7 @@ __global__ void gemm256_nt_kernel(const bf16* __restrict__ A,
     const int row = chunk / CPR, cc = chunk % CPR;
     const long m = m0 + row;
     if (m >= M) continue;
-    const int col0 = n0 + cc * 8;     // Nout % 256 == 0: no column tail
+    const int col0 = n0 + cc * 8;     // Nout % BNT == 0: no column tail
     const long off = m * Nout + col0;
     s16x8 v = *(const s16x8*)(etile + row * BN + cc * 8);
     if (stat_sum) {
@@ -777,16 +786,17 @@ __global__ void gemm256_nt_kernel(const bf16* __restrict__ A,
     *(s16x8*)(out + off) = v;
   }
   if (stat_sum) {
-    // lanes at stride CPR=32 share a column group: one xor fold, then the
+    // lanes at stride CPR share a column group: xor folds, then the
     // per-wave partials go through LDS (etile reads are done) and one global
     // atomic per column per block
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      st_s[j] += __shfl_xor(st_s[j], 32, 64);
-      st_q[j] += __shfl_xor(st_q[j], 32, 64);
-    }
+    for (int j = 0; j < 8; ++j)
+      for (int off = CPR; off < 64; off <<= 1) {
+        st_s[j] += __shfl_xor(st_s[j], off, 64);
+        st_q[j] += __shfl_xor(st_q[j], off, 64);
+      }
     __syncthreads();
-    float* sred = (float*)As;         // [8 waves][32 groups][16]
+    float* sred = (float*)As;         // [8 waves][CPR groups][16]
     const int cc0 = tid % CPR;
     if (lane < CPR) {
 #pragma unroll
@@ -943,32 +953,43 @@ extern "C" void al_conv2d_mm(int mode, const void* A, const void* B, void* out,
   // (stride-2 bwd goes through the parity-class decomposition above).
   if (gemm256_on && (mode == MODE_FWD || mode == MODE_BWD_DATA) &&
       (mode == MODE_FWD || stride == 1) &&
-      sh.Nout % 256 == 0 && sh.KD % 64 == 0 && sh.M >= 128) {
+      sh.Nout % 128 == 0 && sh.KD % 64 == 0 && sh.M >= 128) {
+    const int bnt = (sh.Nout % 256 == 0) ? 256 : 128;
     const bool pure = (R == 1 && S == 1 && stride == 1 && pad == 0);
     const int fast = (mode == MODE_FWD) ? sh.C : sh.K;
     const int grid_m = (int)((sh.M + 255) / 256);
-    const int grid_n = sh.Nout / 256;
+    const int grid_n = sh.Nout / bnt;
     const bool gather_ok = !pure && fast % 8 == 0 &&
                            grid_m * grid_n >= min_grid;
     if (pure || gather_ok) {
       static bool attr_set = false;
       if (!attr_set) {
-        (void)hipFuncSetAttribute((const void*)gemm256_nt_kernel<MODE_PURE>,
-                                  hipFuncAttributeMaxDynamicSharedMemorySize, 131072);
-        (void)hipFuncSetAttribute((const void*)gemm256_nt_kernel<MODE_FWD>,
-                                  hipFuncAttributeMaxDynamicSharedMemorySize, 131072);
-        (void)hipFuncSetAttribute((const void*)gemm256_nt_kernel<MODE_BWD_DATA>,
-                                  hipFuncAttributeMaxDynamicSharedMemorySize, 131072);
+#define SET_ATTR(MODE_, BNT_) (void)hipFuncSetAttribute( \
+        (const void*)gemm256_nt_kernel<MODE_, BNT_>, \
+        hipFuncAttributeMaxDynamicSharedMemorySize, 131072)
+        SET_ATTR(MODE_PURE, 256); SET_ATTR(MODE_FWD, 256);
+        SET_ATTR(MODE_BWD_DATA, 256);
+        SET_ATTR(MODE_PURE, 128); SET_ATTR(MODE_FWD, 128);
+        SET_ATTR(MODE_BWD_DATA, 128);
+#undef SET_ATTR
         attr_set = true;
       }
+      // LDS: A ring 64 KiB + B ring 4 * (bnt/2) * 64 * 2B
+      const size_t lds = 4 * 128 * 64 * 2 + 4 * (size_t)(bnt / 2) * 64 * 2;
       dim3 grid(grid_m * grid_n), block(512);
-#define LAUNCH256(MODE_) hipLaunchKernelGGL((gemm256_nt_kernel<MODE_>), grid, \
-      block, 131072, stream, (const bf16*)A, (const bf16*)B, (bf16*)out, \
+#define LAUNCH256(MODE_, BNT_) hipLaunchKernelGGL((gemm256_nt_kernel<MODE_, BNT_>), \
+      grid, block, lds, stream, (const bf16*)A, (const bf16*)B, (bf16*)out, \
       (const bf16*)zero_page, sh, grid_m, epi_scale, epi_shift, \
       (const bf16*)epi_res, epi_relu, stat_sum, stat_sumsq)
-      if (pure) LAUNCH256(MODE_PURE);
-      else if (mode == MODE_FWD) LAUNCH256(MODE_FWD);
-      else LAUNCH256(MODE_BWD_DATA);
+      if (bnt == 256) {
+        if (pure) LAUNCH256(MODE_PURE, 256);
+        else if (mode == MODE_FWD) LAUNCH256(MODE_FWD, 256);
+        else LAUNCH256(MODE_BWD_DATA, 256);
+      } else {
+        if (pure) LAUNCH256(MODE_PURE, 128);
+        else if (mode == MODE_FWD) LAUNCH256(MODE_FWD, 128);
+        else LAUNCH256(MODE_BWD_DATA, 128);
+      }
 #undef LAUNCH256
       return;
     }
