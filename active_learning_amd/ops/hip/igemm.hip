@@ -961,7 +961,12 @@ extern "C" void al_conv2d_mm(int mode, const void* A, const void* B, void* out,
     const int grid_n = sh.Nout / bnt;
     const bool gather_ok = !pure && fast % 8 == 0 &&
                            grid_m * grid_n >= min_grid;
-    if (pure || gather_ok) {
+    // measured (tools/ab_gemm256.py): the narrow tile only beats the 128^2
+    // igemm on the pure bwd-data shapes (+16% on l2.conv1); the gathered
+    // 3x3 at BNT=128 regresses ~30% (8 MFMA/phase cannot cover the phase
+    // overhead), so those keep the 2-barrier 128^2 kernel.
+    const bool bnt_ok = (bnt == 256) || (pure && mode == MODE_BWD_DATA);
+    if ((pure || gather_ok) && bnt_ok) {
       static bool attr_set = false;
       if (!attr_set) {
 #define SET_ATTR(MODE_, BNT_) (void)hipFuncSetAttribute( \

@@ -49,9 +49,9 @@ CONV_CASES = [
     (16, 56, 56, 256, 64, 3, 1, 1),   # bwd-data gather (Nout=256, grid 196x1)
     (16, 56, 56, 128, 256, 1, 2, 0),  # 1x1 s2 fwd gather (downsample shape)
     (8, 56, 56, 128, 256, 3, 2, 1),   # 3x3 s2 fwd gather
-    # 256x128 tile (Nout % 256 != 0):
-    (4, 28, 28, 256, 128, 1, 1, 0),   # pure 1x1, BNT=128
-    (64, 28, 28, 128, 128, 3, 1, 1),  # gathered 3x3, BNT=128 (grid 196)
+    # 256x128 tile (Nout % 256 != 0; routed for pure bwd-data only):
+    (4, 28, 28, 256, 128, 1, 1, 0),   # fwd falls back to 128^2
+    (4, 28, 28, 128, 256, 1, 1, 0),   # bwd-data routes BNT=128 (Nout=C=128)
 ]
 
 
