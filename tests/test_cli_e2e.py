@@ -65,3 +65,23 @@ def test_freeze_feature_linear_eval(tmp_path):
     args = _debug_args(tmp_path, ["--strategy", "RandomSampler", "--freeze_feature"])
     s = main(args)
     assert s.idxs_lb.sum() == 15
+
+
+def test_gen_jobs_commands_parse(capsys):
+    """Every command line gen_jobs emits must be accepted by our CLI parser
+    and name a registered strategy (reference gen_jobs.py parity)."""
+    import shlex
+    from active_learning_amd import gen_jobs
+    from active_learning_amd.cli import build_parser
+    from active_learning_amd.strategies import get_strategy
+
+    gen_jobs.linear_evaluation_imagenet_experiments(dataset_dir="/data")
+    gen_jobs.end_to_end_imagenet_experiments_pretrained(dataset_dir="/data")
+    gen_jobs.cifar10_experiments(dataset_dir="/data")
+    out = capsys.readouterr().out.strip().splitlines()
+    assert len(out) == 9 + 9 + 11
+    parser = build_parser()
+    for line in out:
+        args = parser.parse_args(shlex.split(line)[2:])  # drop "python main_al.py"
+        assert get_strategy(args.strategy) is not None
+        assert args.rounds > 0 and args.round_budget > 0
