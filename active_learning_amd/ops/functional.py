@@ -386,8 +386,14 @@ class BatchNormAct(Function):
 
         if x.is_cuda:
             ext = require_extension()
-            y = ext.bn_norm_fwd(x, mean, invstd, weight, bias, relu,
-                                residual if residual is not None else x.new_empty(0))
+            # want_mask: the backward re-streams a 1-bit relu mask instead of
+            # y. (grad mode is always off inside Function.forward, so gate on
+            # the inputs' requires_grad flags only.)
+            need_grad = (x.requires_grad or weight.requires_grad or
+                         (residual is not None and residual.requires_grad))
+            y, relu_mask = ext.bn_norm_fwd(
+                x, mean, invstd, weight, bias, relu,
+                residual if residual is not None else x.new_empty(0), need_grad)
         else:
             xf = x.float()
             y = (xf - mean) * invstd * weight + bias
@@ -396,7 +402,11 @@ class BatchNormAct(Function):
             if relu:
                 y = F.relu(y)
             y = y.to(x.dtype)
-        ctx.save_for_backward(x, y, weight, mean, invstd)
+            relu_mask = None
+        # GPU path saves the bit mask (backward never re-reads y); CPU path
+        # keeps y for its fp32 reference masks
+        ctx.save_for_backward(x, relu_mask if x.is_cuda else y, weight, mean,
+                              invstd)
         ctx.use_batch_stats = use_batch_stats
         ctx.relu = relu
         ctx.has_residual = residual is not None
@@ -406,12 +416,14 @@ class BatchNormAct(Function):
 
     @staticmethod
     def backward(ctx, dy):
-        x, y, weight, mean, invstd = ctx.saved_tensors
+        x, mask_or_y, weight, mean, invstd = ctx.saved_tensors
+        y = None if x.is_cuda else mask_or_y
         dy = dy.contiguous()
         if x.is_cuda:
             ext = require_extension()
             # reduce pass: per-channel sums of dy~ and dy~*xhat (dy~ = mask * dy)
-            sum_dy, sum_dy_xhat = ext.bn_bwd_reduce(dy, x, y, mean, invstd, ctx.relu)
+            sum_dy, sum_dy_xhat = ext.bn_bwd_reduce(dy, x, mask_or_y, mean, invstd,
+                                                    ctx.relu)
         else:
             dyf = dy.float()
             if ctx.relu:
@@ -438,9 +450,9 @@ class BatchNormAct(Function):
         dres = None
         if x.is_cuda:
             ext = require_extension()
-            dx, dres_t = ext.bn_bwd(dy, x, y, mean, invstd, weight, sum_dy, sum_dy_xhat,
-                                    n_global, ctx.use_batch_stats, ctx.relu,
-                                    ctx.has_residual)
+            dx, dres_t = ext.bn_bwd(dy, x, mask_or_y, mean, invstd, weight, sum_dy,
+                                    sum_dy_xhat, n_global, ctx.use_batch_stats,
+                                    ctx.relu, ctx.has_residual)
             if ctx.has_residual:
                 dres = dres_t
         else:

@@ -33,11 +33,12 @@ void al_bn_finalize(const float* part_s, const float* part_ss, float* mean,
                     void* stream);
 void al_bn_norm_fwd(const void* x, void* y, const float* mean, const float* invstd,
                     const float* gamma, const float* beta, const void* res, int relu,
-                    long rows, int C, void* stream);
-void al_bn_bwd_reduce(const void* dy, const void* x, const void* y, const float* mean,
+                    void* relu_mask, long rows, int C, void* stream);
+void al_bn_bwd_reduce(const void* dy, const void* x, const void* relu_mask,
+                      const float* mean,
                       const float* invstd, float* sum_dy, float* sum_dy_xhat, int relu,
                       long rows, int C, void* stream);
-void al_bn_bwd(const void* dy, const void* x, const void* y, const float* mean,
+void al_bn_bwd(const void* dy, const void* x, const void* relu_mask, const float* mean,
                const float* invstd, const float* gamma, const float* sum_dy,
                const float* sum_dy_xhat, float n, int use_batch_stats, int relu,
                int has_res, void* dx, void* dres, long rows, int C, void* stream);
@@ -262,9 +263,10 @@ std::vector<Tensor> bn_stats(const Tensor& x) {
   return {sum, sumsq};
 }
 
-Tensor bn_norm_fwd(const Tensor& x, const Tensor& mean, const Tensor& invstd,
-                   const Tensor& gamma, const Tensor& beta, bool relu,
-                   const Tensor& residual) {
+std::vector<Tensor> bn_norm_fwd(const Tensor& x, const Tensor& mean,
+                                const Tensor& invstd,
+                                const Tensor& gamma, const Tensor& beta, bool relu,
+                                const Tensor& residual, bool want_mask) {
   check_bf16_contig(x, "x");
   const int C = x.size(-1);
   const long rows = x.numel() / C;
@@ -272,18 +274,27 @@ Tensor bn_norm_fwd(const Tensor& x, const Tensor& mean, const Tensor& invstd,
   auto y = torch::empty_like(x);
   const bool has_res = residual.numel() > 0;
   if (has_res) check_bf16_contig(residual, "residual");
+  // relu mask: one bit per element, read by the backward passes in place of
+  // a full re-stream of y
+  Tensor mask = (relu && want_mask)
+                    ? torch::empty({rows, (long)C / 8},
+                                   x.options().dtype(torch::kUInt8))
+                    : torch::empty({0}, x.options().dtype(torch::kUInt8));
   al_bn_norm_fwd(x.data_ptr(), y.data_ptr(), mean.contiguous().data_ptr<float>(),
                  invstd.contiguous().data_ptr<float>(),
                  gamma.contiguous().data_ptr<float>(),
                  beta.contiguous().data_ptr<float>(),
-                 has_res ? residual.data_ptr() : nullptr, relu ? 1 : 0, rows, C,
+                 has_res ? residual.data_ptr() : nullptr, relu ? 1 : 0,
+                 mask.numel() ? mask.data_ptr() : nullptr, rows, C,
                  cur_stream());
-  return y;
+  return {y, mask};
 }
 
-std::vector<Tensor> bn_bwd_reduce(const Tensor& dy, const Tensor& x, const Tensor& y,
+std::vector<Tensor> bn_bwd_reduce(const Tensor& dy, const Tensor& x,
+                                  const Tensor& relu_mask,
                                   const Tensor& mean, const Tensor& invstd,
                                   bool relu) {
+  TORCH_CHECK(!relu || relu_mask.numel() > 0, "bn_bwd_reduce: relu needs mask");
   check_bf16_contig(dy, "dy");
   const int C = x.size(-1);
   const long rows = x.numel() / C;
@@ -291,7 +302,8 @@ std::vector<Tensor> bn_bwd_reduce(const Tensor& dy, const Tensor& x, const Tenso
   const int nb = al_bn_reduce_blocks(rows, C);
   auto part_s = torch::empty({nb, C}, opts);
   auto part_sx = torch::empty({nb, C}, opts);
-  al_bn_bwd_reduce(dy.data_ptr(), x.data_ptr(), y.data_ptr(),
+  al_bn_bwd_reduce(dy.data_ptr(), x.data_ptr(),
+                   relu_mask.numel() ? relu_mask.data_ptr() : nullptr,
                    mean.contiguous().data_ptr<float>(),
                    invstd.contiguous().data_ptr<float>(), part_s.data_ptr<float>(),
                    part_sx.data_ptr<float>(), relu ? 1 : 0, rows, C,
@@ -304,18 +316,21 @@ std::vector<Tensor> bn_bwd_reduce(const Tensor& dy, const Tensor& x, const Tenso
   return {sum_dy, sum_dy_xhat};
 }
 
-std::vector<Tensor> bn_bwd(const Tensor& dy, const Tensor& x, const Tensor& y,
+std::vector<Tensor> bn_bwd(const Tensor& dy, const Tensor& x,
+                           const Tensor& relu_mask,
                            const Tensor& mean, const Tensor& invstd,
                            const Tensor& gamma, const Tensor& sum_dy,
                            const Tensor& sum_dy_xhat, double n, bool use_batch_stats,
                            bool relu, bool has_res) {
   check_bf16_contig(dy, "dy");
+  TORCH_CHECK(!relu || relu_mask.numel() > 0, "bn_bwd: relu needs mask");
   const int C = x.size(-1);
   const long rows = x.numel() / C;
   auto dx = torch::empty_like(x);
   Tensor dres;
   if (has_res) dres = torch::empty_like(x);
-  al_bn_bwd(dy.data_ptr(), x.data_ptr(), y.data_ptr(),
+  al_bn_bwd(dy.data_ptr(), x.data_ptr(),
+            relu_mask.numel() ? relu_mask.data_ptr() : nullptr,
             mean.contiguous().data_ptr<float>(),
             invstd.contiguous().data_ptr<float>(),
             gamma.contiguous().data_ptr<float>(),

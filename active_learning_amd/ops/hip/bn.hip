@@ -115,8 +115,9 @@ __global__ void bn_norm2_kernel(const bf16* __restrict__ x, bf16* __restrict__ y
                                 const float* __restrict__ invstd,
                                 const float* __restrict__ gamma,
                                 const float* __restrict__ beta,
-                                const bf16* __restrict__ res, long rows, int C,
-                                int cg_per_block) {
+                                const bf16* __restrict__ res,
+                                unsigned char* __restrict__ relu_mask,
+                                long rows, int C, int cg_per_block) {
   const int cg_local = threadIdx.x % cg_per_block;
   const int row_lane = threadIdx.x / cg_per_block;
   const int rows_per_block = 256 / cg_per_block;
@@ -138,21 +139,29 @@ __global__ void bn_norm2_kernel(const bf16* __restrict__ x, bf16* __restrict__ y
     s16x8 rv;
     if (RES) rv = ((const s16x8*)res)[r * C8 + c8];
     s16x8 o;
+    unsigned char mb = 0;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       float f = bits2f(v[j]) * a[j] + b[j];
       if (RES) f += bits2f(rv[j]);
-      if (RELU) f = fmaxf(f, 0.f);
+      if (RELU) {
+        mb |= (unsigned char)(f > 0.f) << j;
+        f = fmaxf(f, 0.f);
+      }
       o[j] = f2bits(f);
     }
     ((s16x8*)y)[r * C8 + c8] = o;
+    // one mask BIT per element: the backward passes read this instead of
+    // re-streaming y (drops a full activation read from both)
+    if (RELU && relu_mask) relu_mask[r * C8 + c8] = mb;
   }
 }
 
 extern "C" void al_bn_norm_fwd(const void* x, void* y, const float* mean,
                                const float* invstd, const float* gamma,
                                const float* beta, const void* res, int relu,
-                               long rows, int C, hipStream_t stream) {
+                               void* relu_mask, long rows, int C,
+                               hipStream_t stream) {
   const bf16* r = (const bf16*)res;
   const int cg = bn_cg_per_block(C);
   const int rpb = 256 / cg;
@@ -161,7 +170,7 @@ extern "C" void al_bn_norm_fwd(const void* x, void* y, const float* mean,
 #define CASE(RELU_, RES_) \
   hipLaunchKernelGGL((bn_norm2_kernel<RELU_, RES_>), grid, block, 0, stream, \
                      (const bf16*)x, (bf16*)y, mean, invstd, gamma, beta, r, \
-                     rows, C, cg)
+                     (unsigned char*)relu_mask, rows, C, cg)
   if (relu) { if (r) CASE(true, true); else CASE(true, false); }
   else      { if (r) CASE(false, true); else CASE(false, false); }
 #undef CASE
@@ -174,7 +183,7 @@ extern "C" void al_bn_norm_fwd(const void* x, void* y, const float* mean,
 template <bool RELU>
 __global__ void bn_bwd_reduce_kernel(const bf16* __restrict__ dy,
                                      const bf16* __restrict__ x,
-                                     const bf16* __restrict__ y,
+                                     const unsigned char* __restrict__ relu_mask,
                                      const float* __restrict__ mean,
                                      const float* __restrict__ invstd,
                                      float* __restrict__ sum_dy,
@@ -199,12 +208,12 @@ __global__ void bn_bwd_reduce_kernel(const bf16* __restrict__ dy,
     const long off = r * C8 + c8;
     s16x8 gv = ((const s16x8*)dy)[off];
     s16x8 xv = ((const s16x8*)x)[off];
-    s16x8 yv;
-    if (RELU) yv = ((const s16x8*)y)[off];
+    unsigned char mb = 0xff;
+    if (RELU) mb = relu_mask[off];
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       float g = bits2f(gv[j]);
-      if (RELU) g = bits2f(yv[j]) > 0.f ? g : 0.f;
+      if (RELU) g = (mb >> j) & 1 ? g : 0.f;
       s[j] += g;
       sx[j] += g * (bits2f(xv[j]) - m[j]) * is[j];
     }
@@ -231,7 +240,7 @@ __global__ void bn_bwd_reduce_kernel(const bf16* __restrict__ dy,
   }
 }
 
-extern "C" void al_bn_bwd_reduce(const void* dy, const void* x, const void* y,
+extern "C" void al_bn_bwd_reduce(const void* dy, const void* x, const void* relu_mask,
                                  const float* mean, const float* invstd,
                                  float* sum_dy, float* sum_dy_xhat, int relu,
                                  long rows, int C, hipStream_t stream) {
@@ -242,11 +251,11 @@ extern "C" void al_bn_bwd_reduce(const void* dy, const void* x, const void* y,
   dim3 grid(row_blocks, (C / 8 + cg - 1) / cg);
   if (relu)
     hipLaunchKernelGGL((bn_bwd_reduce_kernel<true>), grid, block, 0, stream,
-                       (const bf16*)dy, (const bf16*)x, (const bf16*)y, mean, invstd,
+                       (const bf16*)dy, (const bf16*)x, (const unsigned char*)relu_mask, mean, invstd,
                        sum_dy, sum_dy_xhat, rows, C, cg);
   else
     hipLaunchKernelGGL((bn_bwd_reduce_kernel<false>), grid, block, 0, stream,
-                       (const bf16*)dy, (const bf16*)x, (const bf16*)y, mean, invstd,
+                       (const bf16*)dy, (const bf16*)x, (const unsigned char*)relu_mask, mean, invstd,
                        sum_dy, sum_dy_xhat, rows, C, cg);
 }
 
@@ -262,7 +271,7 @@ extern "C" void al_bn_bwd_reduce(const void* dy, const void* x, const void* y,
 // c0) hoisted into registers.
 template <bool RELU, bool BATCH, bool RES>
 __global__ void bn_bwd2_kernel(const bf16* __restrict__ dy, const bf16* __restrict__ x,
-                               const bf16* __restrict__ y,
+                               const unsigned char* __restrict__ relu_mask,
                                const float* __restrict__ mean,
                                const float* __restrict__ invstd,
                                const float* __restrict__ gamma,
@@ -292,14 +301,15 @@ __global__ void bn_bwd2_kernel(const bf16* __restrict__ dy, const bf16* __restri
   for (long r = row0; r < rows; r += step) {
     const long i = r * C8 + c8;
     s16x8 gv = ((const s16x8*)dy)[i];
-    s16x8 xv, yv;
+    s16x8 xv;
     if (BATCH) xv = ((const s16x8*)x)[i];
-    if (RELU) yv = ((const s16x8*)y)[i];
+    unsigned char mb = 0xff;
+    if (RELU) mb = relu_mask[i];
     s16x8 odx, ores;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       float g = bits2f(gv[j]);
-      if (RELU) g = bits2f(yv[j]) > 0.f ? g : 0.f;
+      if (RELU) g = (mb >> j) & 1 ? g : 0.f;
       if (RES) ores[j] = f2bits(g);
       float v;
       if (BATCH) v = gi[j] * g - t2[j] * bits2f(xv[j]) + c0[j];
@@ -311,7 +321,7 @@ __global__ void bn_bwd2_kernel(const bf16* __restrict__ dy, const bf16* __restri
   }
 }
 
-extern "C" void al_bn_bwd(const void* dy, const void* x, const void* y,
+extern "C" void al_bn_bwd(const void* dy, const void* x, const void* relu_mask,
                           const float* mean, const float* invstd, const float* gamma,
                           const float* sum_dy, const float* sum_dy_xhat, float n,
                           int use_batch_stats, int relu, int has_res, void* dx,
@@ -323,7 +333,7 @@ extern "C" void al_bn_bwd(const void* dy, const void* x, const void* y,
   float inv_n = 1.0f / n;
 #define CASE(RELU_, BATCH_, RES_) \
   hipLaunchKernelGGL((bn_bwd2_kernel<RELU_, BATCH_, RES_>), grid, block, 0, stream, \
-                     (const bf16*)dy, (const bf16*)x, (const bf16*)y, mean, invstd, \
+                     (const bf16*)dy, (const bf16*)x, (const unsigned char*)relu_mask, mean, invstd, \
                      gamma, sum_dy, sum_dy_xhat, inv_n, (bf16*)dx, (bf16*)dres, \
                      rows, C, cg)
   if (relu) {

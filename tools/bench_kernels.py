@@ -48,6 +48,8 @@ def main():
         x = torch.randn(rows, C, device=dev).to(torch.bfloat16).view(rows, 1, 1, C)
         dy = torch.randn_like(x)
         y = torch.randn_like(x)
+        mk = torch.randint(0, 256, (rows, C // 8), device=dev,
+                           dtype=torch.uint8)
         mean = torch.zeros(C, device=dev)
         invstd = torch.ones(C, device=dev)
         g = torch.ones(C, device=dev)
@@ -57,13 +59,13 @@ def main():
         report(f"bn_stats     r={rows} C={C}", timeit(lambda: ext.bn_stats(x), args.iters),
                byte=byte)
         report(f"bn_norm_fwd  r={rows} C={C}",
-               timeit(lambda: ext.bn_norm_fwd(x, mean, invstd, g, b, True, e),
+               timeit(lambda: ext.bn_norm_fwd(x, mean, invstd, g, b, True, e, True),
                       args.iters), byte=2 * byte)
         report(f"bn_bwd_reduce r={rows} C={C}",
-               timeit(lambda: ext.bn_bwd_reduce(dy, x, y, mean, invstd, True),
+               timeit(lambda: ext.bn_bwd_reduce(dy, x, mk, mean, invstd, True),
                       args.iters), byte=3 * byte)
         report(f"bn_bwd       r={rows} C={C}",
-               timeit(lambda: ext.bn_bwd(dy, x, y, mean, invstd, g, mean, mean,
+               timeit(lambda: ext.bn_bwd(dy, x, mk, mean, invstd, g, mean, mean,
                                          float(rows), True, True, False),
                       args.iters), byte=3 * byte)
 
