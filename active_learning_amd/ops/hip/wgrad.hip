@@ -428,6 +428,170 @@ __global__ void wgrad2_kernel(const bf16* __restrict__ dy,
   }
 }
 
+// ---------------------------------------------------------------------------
+// wgrad v3 (experimental, AL_WGRAD_V3): same row-major + tr_b16 structure as
+// v2 but 512 threads and a 3-buffer ring with counted vmcnt — the staging
+// for step s+2 is issued before step s's MFMAs and only vmcnt(4) (one stage
+// in flight) is waited at each step top, so no vmcnt(0) drain ever happens
+// in the loop. 96 KiB LDS -> 1 block/CU but still 8 waves (2/SIMD).
+// ---------------------------------------------------------------------------
+
+template <int GKR, int GNC>
+__launch_bounds__(512)
+__global__ void wgrad3_kernel(const bf16* __restrict__ dy,
+                              const bf16* __restrict__ x,
+                              float* __restrict__ dw,
+                              const bf16* __restrict__ zero,
+                              WgradShape sh, int grid_k, long l_per_z) {
+  constexpr int BMK = GKR * 64, BNW = GNC * 64, BL = 64;
+  constexpr int ACH = BMK * BL;
+  constexpr int BCH = BNW * BL;
+  constexpr int NA = ACH / (512 * 8);      // 16B chunks per thread
+  constexpr int NB = BCH / (512 * 8);
+  const int bk = blockIdx.x % grid_k;
+  const int bn = blockIdx.x / grid_k;
+  const int k0 = bk * BMK;
+  const int n0 = bn * BNW;
+  const long lz0 = (long)blockIdx.y * l_per_z;
+  const long lz1 = min(sh.L, lz0 + l_per_z);
+  if (lz0 >= lz1) return;
+
+  __shared__ __attribute__((aligned(16))) bf16 As3[3][ACH];
+  __shared__ __attribute__((aligned(16))) bf16 Bs3[3][BCH];
+
+  const int tid = threadIdx.x;
+  const int wid = tid >> 6, lane = tid & 63;
+  const int wr = wid >> 2, wc = wid & 3;   // 2x4 wave grid, 64x32 each
+  const int l15 = lane & 15, l4 = lane >> 4;
+
+  long a_off[NA];
+  int a_l[NA];
+  bool a_chok[NA];
+#pragma unroll
+  for (int i = 0; i < NA; ++i) {
+    const int ci = i * 512 + tid;
+    const int g = ci >> 7, l = (ci & 127) >> 1, h = ci & 1;
+    const int ch = k0 + g * 16 + h * 8;
+    a_l[i] = l;
+    a_chok[i] = ch + 8 <= sh.K;
+    a_off[i] = (lz0 + l) * (long)sh.K + ch;
+  }
+  int b_n[NB], b_p[NB], b_q[NB], b_ho[NB], b_wo[NB];
+  long b_coff[NB];
+  int b_l[NB];
+  bool b_chok[NB];
+  const int dn3 = (int)(BL / ((long)sh.P * sh.Q));
+  const int rem3 = (int)(BL % ((long)sh.P * sh.Q));
+  const int dp3 = rem3 / sh.Q, dq3 = rem3 % sh.Q;
+#pragma unroll
+  for (int i = 0; i < NB; ++i) {
+    const int ci = i * 512 + tid;
+    const int g = ci >> 7, l = (ci & 127) >> 1, h = ci & 1;
+    const int nw = n0 + g * 16 + h * 8;
+    b_l[i] = l;
+    b_chok[i] = nw + 8 <= sh.Nw;
+    const int c = nw % sh.C;
+    const int rs = nw / sh.C;
+    b_ho[i] = (rs / sh.S) - sh.pad;
+    b_wo[i] = (rs % sh.S) - sh.pad;
+    b_coff[i] = c;
+    const long m = lz0 + l;
+    b_q[i] = (int)(m % sh.Q);
+    long t = m / sh.Q;
+    b_p[i] = (int)(t % sh.P);
+    b_n[i] = (int)(t / sh.P);
+  }
+
+  auto stage = [&](int buf, long l0, bool live) {
+#pragma unroll
+    for (int i = 0; i < NA; ++i) {
+      const bf16* src = (live && a_chok[i] && l0 + a_l[i] < lz1)
+                            ? dy + a_off[i] : zero;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)src,
+          (__attribute__((address_space(3))) void*)(&As3[buf][(i * 512 + wid * 64) * 8]),
+          16, 0, 0);
+      a_off[i] += (long)BL * sh.K;
+    }
+#pragma unroll
+    for (int i = 0; i < NB; ++i) {
+      const int hh = b_p[i] * sh.stride + b_ho[i];
+      const int ww = b_q[i] * sh.stride + b_wo[i];
+      const bool ok = live && b_chok[i] && (l0 + b_l[i] < lz1) &&
+                      (unsigned)hh < (unsigned)sh.H && (unsigned)ww < (unsigned)sh.W;
+      const bf16* src = ok
+          ? x + (((long)b_n[i] * sh.H + hh) * sh.W + ww) * sh.C + b_coff[i]
+          : zero;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)src,
+          (__attribute__((address_space(3))) void*)(&Bs3[buf][(i * 512 + wid * 64) * 8]),
+          16, 0, 0);
+      b_q[i] += dq3;
+      if (b_q[i] >= sh.Q) { b_q[i] -= sh.Q; ++b_p[i]; }
+      b_p[i] += dp3;
+      if (b_p[i] >= sh.P) { b_p[i] -= sh.P; ++b_n[i]; }
+      b_n[i] += dn3;
+    }
+  };
+
+  f32x4 acc[4][2];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  typedef __attribute__((address_space(3))) s16x4* lds3_v4p;
+  auto trfrag = [&](const bf16* img, int group, int mc) -> bf16x8 {
+    const bf16* p = img + group * 1024 + mc * 32 * 16 + l4 * 8 * 16 + l15 * 4;
+    union { s16x4 h[2]; bf16x8 v; } u;
+    u.h[0] = __builtin_amdgcn_ds_read_tr16_b64_v4i16((lds3_v4p)p);
+    u.h[1] = __builtin_amdgcn_ds_read_tr16_b64_v4i16((lds3_v4p)(p + 64));
+    return u.v;
+  };
+
+  auto compute = [&](int buf) {
+#pragma unroll
+    for (int mc = 0; mc < 2; ++mc) {
+      bf16x8 afrag[4], bfrag[2];
+#pragma unroll
+      for (int f = 0; f < 4; ++f) afrag[f] = trfrag(As3[buf], wr * 4 + f, mc);
+#pragma unroll
+      for (int f = 0; f < 2; ++f) bfrag[f] = trfrag(Bs3[buf], wc * 2 + f, mc);
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 2; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag[mi], bfrag[ni], acc[mi][ni], 0, 0, 0);
+    }
+  };
+
+  const long steps = (lz1 - lz0 + BL - 1) / BL;
+  stage(0, lz0, true);
+  stage(1, lz0 + BL, 1 < steps);
+  for (long s = 0; s < steps; ++s) {
+    // counted wait: only the NEXT stage's 4 loads may stay in flight
+    asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    __syncthreads();
+    stage((int)((s + 2) % 3), lz0 + (s + 2) * BL, s + 2 < steps);
+    compute((int)(s % 3));
+  }
+
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi) {
+#pragma unroll
+    for (int ni = 0; ni < 2; ++ni) {
+      const int col = n0 + wc * 32 + ni * 16 + l15;
+      if (col >= sh.Nw) continue;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = k0 + wr * 64 + mi * 16 + l4 * 4 + r;
+        if (row < sh.K) atomicAdd(&dw[(long)row * sh.Nw + col], acc[mi][ni][r]);
+      }
+    }
+  }
+}
+
 // direct fallback: one dW element per thread, strided over L (safety net;
 // normal stems go through the packed-im2col MFMA path)
 __global__ void wgrad_direct_kernel(const bf16* __restrict__ dy,
@@ -495,7 +659,16 @@ extern "C" void al_conv2d_wgrad(const void* dy, const void* x, float* dw,
         const char* e = getenv("AL_WGRAD_SCAL");
         scal = (e && e[0] == '1') ? 1 : 0;
       }
-      if (scal) {
+      static int v3 = -1;
+      if (v3 < 0) {
+        const char* e = getenv("AL_WGRAD_V3");
+        v3 = (e && e[0] == '1') ? 1 : 0;
+      }
+      if (v3 && !narrow_k) {
+        hipLaunchKernelGGL((wgrad3_kernel<2, 2>), grid, dim3(512), 0, stream,
+                           (const bf16*)dy, (const bf16*)x, dw,
+                           (const bf16*)zero_page, sh, grid_k, l_per_z);
+      } else if (scal) {
         if (narrow_k)
           hipLaunchKernelGGL((wgrad2_kernel<1, 4, true>), grid, block, 0, stream,
                              (const bf16*)dy, (const bf16*)x, dw,
