@@ -32,6 +32,17 @@ def _cpu_conv_args(x_nhwc, w_krsc):
     return x_nhwc.permute(0, 3, 1, 2), w_krsc.permute(0, 3, 1, 2)
 
 
+def bump_tick(t):
+    """In-kernel updates mutate storage without bumping torch's _version;
+    derived caches (the folded-BN eval scale/shift, ops/fused.py) key on
+    this counter as well."""
+    if t is not None:
+        try:
+            t._al_tick = getattr(t, "_al_tick", 0) + 1
+        except Exception:
+            pass
+
+
 def cast_cached(weight, dtype):
     """Per-version cached bf16 copy of an fp32 master weight (saves a cast +
     copy per layer per step; invalidated by optimizers via
@@ -334,6 +345,8 @@ class BatchNormAct(Function):
                         mean, invstd = _re().bn_finalize(
                             s, ss, running_mean, running_var, float(n_local),
                             momentum, eps, True)
+                        bump_tick(running_mean)
+                        bump_tick(running_var)
                     n = float(n_local)
                 else:
                     n = float(n_local)
@@ -354,6 +367,8 @@ class BatchNormAct(Function):
                     mean, invstd = ext.bn_stats_finalize(
                         x, running_mean, running_var, momentum, eps,
                         running_mean is not None)
+                    bump_tick(running_mean)
+                    bump_tick(running_var)
                 n = float(n_local)
             else:
                 if x.is_cuda:

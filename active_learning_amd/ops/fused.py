@@ -19,9 +19,17 @@ _STATS_DISABLED = os.environ.get("AL_AMD_DISABLE_CONV_STATS") == "1"
 from .functional import _igemm_eligible, _wpad_cached, cast_cached
 
 
+def _tick(t):
+    return getattr(t, "_al_tick", 0)
+
+
 def _bn_fold_cached(bn):
+    # key includes _al_tick: the fused optimizers and the BN finalize kernel
+    # update these tensors in-kernel, which does NOT bump torch's _version
+    # (a stale fold here showed up as one-epoch-old validation accuracy)
     key = (bn.weight._version, bn.bias._version, bn.running_mean._version,
-           bn.running_var._version)
+           bn.running_var._version, _tick(bn.weight), _tick(bn.bias),
+           _tick(bn.running_mean), _tick(bn.running_var))
     cache = getattr(bn, "_al_fold", None)
     if cache is not None and cache[0] == key:
         return cache[1], cache[2]

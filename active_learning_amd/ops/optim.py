@@ -16,6 +16,7 @@ import math
 import torch
 
 from .extension import extension_available, load_extension
+from .functional import bump_tick
 
 
 def _invalidate(p):
@@ -67,6 +68,7 @@ class FusedSGD(torch.optim.Optimizer):
                         p._al_cast = (p._version, cache[1])
                     else:
                         _invalidate(p)
+                    bump_tick(p)  # in-kernel update: no _version bump
                 else:
                     gf = g.float()
                     if wd != 0:
@@ -108,6 +110,7 @@ class FusedAdam(torch.optim.Optimizer):
                     ext.adam_step(p, p.grad.to(torch.float32), state["exp_avg"],
                                   state["exp_avg_sq"], lr, beta1, beta2, eps, wd, bc1, bc2)
                     _invalidate(p)
+                    bump_tick(p)  # in-kernel update: no _version bump
                 else:
                     gf = p.grad.float()
                     if wd != 0:

@@ -388,3 +388,34 @@ def test_grad_arena_matches_plain():
         # split-K atomic order is nondeterministic run-to-run: numerically
         # identical, not bitwise
         assert torch.allclose(a, b, rtol=1e-4, atol=1e-2), "arena grad mismatch"
+
+
+def test_bn_fold_refreshes_after_inkernel_update():
+    """The folded-BN eval cache must key on in-kernel updates: fused SGD /
+    Adam and the BN finalize kernel mutate storage without bumping torch's
+    _version, so they bump _al_tick and the fold key includes it."""
+    from active_learning_amd.models.layers import Conv2dNHWC as ConvM
+    from active_learning_amd.models.layers import BatchNormAct2d
+    from active_learning_amd.ops.fused import conv_bn_act
+    from active_learning_amd.ops.functional import bump_tick
+
+    torch.manual_seed(0)
+    conv = ConvM(64, 64, 3, stride=1, padding=1).cuda()
+    bn = BatchNormAct2d(64, relu=True).cuda()
+    bn.eval()
+    conv.eval()
+    x = torch.randn(2, 8, 8, 64, device="cuda").to(torch.bfloat16)
+    with torch.no_grad():
+        y0 = conv_bn_act(conv, bn, x)  # fold cached
+        # mutate the weight storage WITHOUT a _version bump (what the fused
+        # optimizer kernels do): alias the storage through a fresh TensorImpl
+        alias = torch.tensor([], dtype=bn.weight.dtype, device=bn.weight.device)
+        alias.set_(bn.weight.untyped_storage(), 0, bn.weight.shape)
+        alias.mul_(2.0)
+        bump_tick(bn.weight)  # the optimizer contract
+        torch.cuda.synchronize()
+        y1 = conv_bn_act(conv, bn, x)
+        bn._al_fold = None
+        y_fresh = conv_bn_act(conv, bn, x)
+    assert torch.equal(y1, y_fresh), "fold cache served stale scale/shift"
+    assert not torch.equal(y0, y1)
