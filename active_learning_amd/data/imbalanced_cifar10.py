@@ -7,7 +7,7 @@ per-class shuffle + truncate (:45-61).
 
 import numpy as np
 
-from .cifar10 import CustomCIFAR10, get_data_cifar10
+from .cifar10 import CustomCIFAR10
 from .transforms import cifar_transforms
 
 

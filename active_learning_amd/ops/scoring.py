@@ -22,7 +22,6 @@ Reference counterparts:
     argmax/sampling all stay on device (HBM-resident per BASELINE.json).
 """
 
-import numpy as np
 import torch
 import torch.nn.functional as F
 

@@ -11,7 +11,6 @@ Selection = randomized greedy k-center (k-means++ seeding, :72-73).
 """
 
 import numpy as np
-import torch
 
 from ..ops.scoring import badge_pairwise_sqdist, badge_vectors
 from .common import forward_pool

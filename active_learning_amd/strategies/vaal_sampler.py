@@ -12,7 +12,6 @@ VAE step = labeled recon + transductive recon + adversarial BCE with
 
 import numpy as np
 import torch
-import torch.distributed as dist
 import torch.nn as nn
 from torch.utils.data import DataLoader, Subset
 
