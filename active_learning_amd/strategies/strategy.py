@@ -337,8 +337,9 @@ class Strategy:
             self.net = self.net.module
 
     def _train(self, rank, epoch, loader_tr, optimizer, criterion, step):
-        """One training epoch — the hot loop (strategy.py:249-284)."""
-        total_loss = 0.0
+        """One training epoch — the hot loop (strategy.py:249-284). The loss
+        stays on-device between log points: a per-batch .cpu() would force a
+        full pipeline sync every iteration."""
         for batch_idx, (x, y, _idxs) in enumerate(loader_tr):
             x = x.to(self.device, non_blocking=True)
             y = y.to(self.device, non_blocking=True)
@@ -349,9 +350,8 @@ class Strategy:
             if isinstance(self.net, BucketedDDP):
                 self.net.finalize_grads()
             optimizer.step()
-            cur_loss = loss.detach().float().cpu()
-            total_loss += float(cur_loss)
-            if batch_idx % 25 == 0:
+            if batch_idx % 25 == 0:  # reference cadence (strategy.py:276-279)
+                cur_loss = loss.detach().float().cpu()
                 msg = (f"\tRound {self.round}, Epoch {epoch}, batch "
                        f"{batch_idx}/{len(loader_tr)}, loss is {cur_loss} on worker "
                        f"rank {rank}")
