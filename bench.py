@@ -76,6 +76,10 @@ def main():
     net = ResNetSimCLR(base, num_classes=args.classes).to(device)
     if world > 1:
         from active_learning_amd.parallel import BucketedDDP, convert_sync_batchnorm
+        # the reference trains with SyncBatchNorm under DDP (strategy.py:292);
+        # keep the same work in the scaling bench. AL_BENCH_SYNCBN=0 disables.
+        if os.environ.get("AL_BENCH_SYNCBN", "1") == "1":
+            convert_sync_batchnorm(net)
         net = BucketedDDP(net, bucket_cap_mb=args.bucket_mb)
     opt = FusedSGD(net.parameters(), lr=0.1, momentum=0.9, weight_decay=1e-4)
 
@@ -160,7 +164,8 @@ def main():
             "data": "synthetic",
             "config": {"model": args.model, "global_batch": args.batch * n_gpus,
                        "seq_len": None, "img": args.img,
-                       "parallelism": f"dp{n_gpus}"},
+                       "parallelism": f"dp{n_gpus}" + ("+syncbn" if n_gpus > 1 and
+                           os.environ.get("AL_BENCH_SYNCBN", "1") == "1" else "")},
         }))
     if dist:
         dist.destroy_process_group()
