@@ -85,3 +85,13 @@ def test_gen_jobs_commands_parse(capsys):
         args = parser.parse_args(shlex.split(line)[2:])  # drop "python main_al.py"
         assert get_strategy(args.strategy) is not None
         assert args.rounds > 0 and args.round_budget > 0
+
+
+def test_compute_dtype_fp32_round(tmp_path):
+    """--compute_dtype fp32 disables the bf16 cast (CPU runs full fp32;
+    PARITY.md documents the GPU path as bf16-with-fp32-master)."""
+    args = _debug_args(tmp_path, ["--strategy", "RandomSampler",
+                                  "--compute_dtype", "fp32"])
+    s = main(args)
+    assert s.net.encoder.compute_dtype is None
+    assert s.round == 1
