@@ -482,9 +482,10 @@ __global__ void igemm_kernel(const bf16* __restrict__ A, const bf16* __restrict_
 #define MODE_PURE 3
 
 // BNT: output-tile width (256, or 128 for Nout%256!=0 layers e.g. the
-// 128-wide layer2 convs). A-side geometry is unchanged; B half-tiles are
-// (BNT/2) rows and the per-wave quadrant slice narrows to BNT/8 columns.
-template <int MODE, int BNT = 256>
+// 128-wide layer2 convs). BMT: output-tile height; 512x128 restores the
+// 256²-tile's MFMA density (128 block-MFMAs per phase) for gathered
+// 128-wide layers, where 256x128's 64 could not cover the phase overhead.
+template <int MODE, int BNT = 256, int BMT = 256>
 __launch_bounds__(512, 1)
 __global__ void gemm256_nt_kernel(const bf16* __restrict__ A,
                                   const bf16* __restrict__ B,
@@ -498,11 +499,14 @@ __global__ void gemm256_nt_kernel(const bf16* __restrict__ A,
                                   float* __restrict__ stat_sumsq) {
   const long M = sh.M;
   const int Nout = sh.Nout, KD = sh.KD;
-  constexpr int BM = 256, BN = BNT, BK = 64;
-  constexpr int HALF = 128 * BK;          // A half-tile elements (16 KiB)
+  constexpr int BM = BMT, BN = BNT, BK = 64;
+  constexpr int HALF = (BMT / 2) * BK;    // A half-tile elements
   constexpr int HALF_B = (BNT / 2) * BK;  // B half-tile elements
+  constexpr int NLA = BMT / 128;          // A glds per thread per half
   constexpr int NLB = (BNT == 256) ? 2 : 1;  // B glds per thread per half
   constexpr int NF = BNT / 128;           // per-wave N fragments
+  constexpr int MF = BMT / 64;            // per-wave M fragments
+  constexpr int NSL = 2 * NLA;            // A gather slots (2 halves)
   // XCD-aware bijective block remap (T1)
   const int nwg = gridDim.x;
   int bid = blockIdx.x;
@@ -514,7 +518,7 @@ __global__ void gemm256_nt_kernel(const bf16* __restrict__ A,
       bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + pos;
   }
   const int bm = bid % grid_m, bn = bid / grid_m;
-  const long m0 = (long)bm * BM;
+  const long m0 = (long)bm * BMT;
   const int n0 = bn * BN;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
@@ -534,14 +538,14 @@ __global__ void gemm256_nt_kernel(const bf16* __restrict__ A,
   // coordinate (r, s, cf) advances by +BK per refill — same carry scheme as
   // igemm_kernel, four independent slot states.
   const int fastC = (MODE == MODE_FWD) ? sh.C : sh.K;
-  int ga_r[4], ga_s[4], ga_cf[4];
-  long ga_pix[4];
-  int ga_p[4], ga_q[4];
-  bool ga_ok[4];
+  int ga_r[NSL], ga_s[NSL], ga_cf[NSL];
+  long ga_pix[NSL];
+  int ga_p[NSL], ga_q[NSL];
+  bool ga_ok[NSL];
   if (MODE != MODE_PURE) {
 #pragma unroll
-    for (int sl = 0; sl < 4; ++sl) {
-      const int h = sl >> 1, li = sl & 1;
+    for (int sl = 0; sl < NSL; ++sl) {
+      const int h = sl / NLA, li = sl % NLA;
       const int dc = li * 512 + wid * 64 + lane;
       const int drow = dc >> 3, du = dc & 7;
       const int su = du ^ (drow & 7);
@@ -550,7 +554,7 @@ __global__ void gemm256_nt_kernel(const bf16* __restrict__ A,
       const int rs = kd0 / fastC;
       ga_s[sl] = rs % sh.S;
       ga_r[sl] = rs / sh.S;
-      const long m = m0 + h * 128 + drow;
+      const long m = m0 + h * (BMT / 2) + drow;
       ga_ok[sl] = m < M;
       if (ga_ok[sl]) {
         if (MODE == MODE_FWD) {
@@ -592,10 +596,10 @@ __global__ void gemm256_nt_kernel(const bf16* __restrict__ A,
       return A + (ga_pix[sl] + (long)p * sh.Q + q) * sh.K + ga_cf[sl];
     }
   };
-  auto ga_advance = [&](int h) {      // +BK for the two slots of half h
+  auto ga_advance = [&](int h) {      // +BK for the slots of half h
 #pragma unroll
-    for (int li = 0; li < 2; ++li) {
-      const int sl = h * 2 + li;
+    for (int li = 0; li < NLA; ++li) {
+      const int sl = h * NLA + li;
       ga_cf[sl] += BK;
       while (ga_cf[sl] >= fastC) {
         ga_cf[sl] -= fastC;
@@ -610,17 +614,17 @@ __global__ void gemm256_nt_kernel(const bf16* __restrict__ A,
   // A-halves of gather modes read through the slot state (advance first!).
   auto stage_a = [&](int h, int ktile, bf16* slot) {
 #pragma unroll
-    for (int li = 0; li < 2; ++li) {
+    for (int li = 0; li < NLA; ++li) {
       const bf16* src;
       if (MODE == MODE_PURE) {
         const int dc = li * 512 + wid * 64 + lane;
         const int drow = dc >> 3, du = dc & 7;
         const int su = du ^ (drow & 7);
-        const long row = m0 + h * 128 + drow;
+        const long row = m0 + h * (BMT / 2) + drow;
         src = (ktile < KT && row < M) ? A + row * KD + (long)ktile * BK + su * 8
                                       : zero;
       } else {
-        src = (ktile < KT) ? ga_src(h * 2 + li) : zero;
+        src = (ktile < KT) ? ga_src(h * NLA + li) : zero;
       }
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) void*)src,
@@ -665,13 +669,13 @@ __global__ void gemm256_nt_kernel(const bf16* __restrict__ A,
     }
   };
 
-  f32x4 acc[2][2][4][NF];
+  f32x4 acc[2][2][MF][NF];
 #pragma unroll
   for (int i = 0; i < 2; ++i)
 #pragma unroll
     for (int j = 0; j < 2; ++j)
 #pragma unroll
-      for (int g = 0; g < 4; ++g)
+      for (int g = 0; g < MF; ++g)
 #pragma unroll
         for (int n = 0; n < NF; ++n) acc[i][j][g][n] = f32x4{0.f, 0.f, 0.f, 0.f};
 
@@ -690,44 +694,76 @@ __global__ void gemm256_nt_kernel(const bf16* __restrict__ A,
   for (int g = 0; g < KT; ++g) {
     const int buf = g & 1;
     // once per K-tile: the 2 half-tiles issued since this tile's last half
-    // (one A refill = 2 loads + one B refill = NLB) may stay in flight —
+    // (one A refill = NLA loads + one B refill = NLB) may stay in flight —
     // never drain to vmcnt(0) (T3+T4)
-    if (NLB == 2) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    if (NLA + NLB == 4) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    else if (NLA + NLB == 5) asm volatile("s_waitcnt vmcnt(5)" ::: "memory");
     else asm volatile("s_waitcnt vmcnt(3)" ::: "memory");
 #pragma unroll
     for (int ph = 0; ph < 4; ++ph) {
       const int qm = ph >> 1, qn = ph & 1;
       const bf16* ah = As + (buf * 2 + qm) * HALF;
       const bf16* bh = Bs + (buf * 2 + qn) * HALF_B;
-      bf16x8 af[4][2], bf[NF][2];
+      if constexpr (MF <= 4) {
+        bf16x8 af[MF][2], bf[NF][2];
 #pragma unroll
-      for (int fg = 0; fg < 4; ++fg) {
-        const int rowh = wrq * 64 + fg * 16 + l15;
+        for (int fg = 0; fg < MF; ++fg) {
+          const int rowh = wrq * (BMT / 4) + fg * 16 + l15;
+#pragma unroll
+          for (int kc = 0; kc < 2; ++kc)
+            af[fg][kc] = *(const bf16x8*)(ah + rowh * BK + (((kc * 4 + l4) ^ (rowh & 7)) * 8));
+        }
+#pragma unroll
+        for (int ng = 0; ng < NF; ++ng) {
+          const int colh = wcq * (BNT / 8) + ng * 16 + l15;
+#pragma unroll
+          for (int kc = 0; kc < 2; ++kc)
+            bf[ng][kc] = *(const bf16x8*)(bh + colh * BK + (((kc * 4 + l4) ^ (colh & 7)) * 8));
+        }
+        refill(g, ph);
+        __builtin_amdgcn_s_barrier();
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_setprio(1);
 #pragma unroll
         for (int kc = 0; kc < 2; ++kc)
-          af[fg][kc] = *(const bf16x8*)(ah + rowh * BK + (((kc * 4 + l4) ^ (rowh & 7)) * 8));
+#pragma unroll
+          for (int fg = 0; fg < MF; ++fg)
+#pragma unroll
+            for (int ng = 0; ng < NF; ++ng)
+              acc[qm][qn][fg][ng] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  af[fg][kc], bf[ng][kc], acc[qm][qn][fg][ng], 0, 0, 0);
+        __builtin_amdgcn_s_setprio(0);
+        __builtin_amdgcn_s_barrier();
+      } else {
+        // tall tile (MF=8): read fragments per 32-deep chunk so only MF+NF
+        // frags are live at once (af[MF][2] would spill at 512 rows)
+        refill(g, ph);
+        __builtin_amdgcn_s_barrier();
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int kc = 0; kc < 2; ++kc) {
+          bf16x8 af[MF], bf[NF];
+#pragma unroll
+          for (int fg = 0; fg < MF; ++fg) {
+            const int rowh = wrq * (BMT / 4) + fg * 16 + l15;
+            af[fg] = *(const bf16x8*)(ah + rowh * BK + (((kc * 4 + l4) ^ (rowh & 7)) * 8));
+          }
+#pragma unroll
+          for (int ng = 0; ng < NF; ++ng) {
+            const int colh = wcq * (BNT / 8) + ng * 16 + l15;
+            bf[ng] = *(const bf16x8*)(bh + colh * BK + (((kc * 4 + l4) ^ (colh & 7)) * 8));
+          }
+#pragma unroll
+          for (int fg = 0; fg < MF; ++fg)
+#pragma unroll
+            for (int ng = 0; ng < NF; ++ng)
+              acc[qm][qn][fg][ng] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  af[fg], bf[ng], acc[qm][qn][fg][ng], 0, 0, 0);
+        }
+        __builtin_amdgcn_s_setprio(0);
+        __builtin_amdgcn_s_barrier();
       }
-#pragma unroll
-      for (int ng = 0; ng < NF; ++ng) {
-        const int colh = wcq * (BNT / 8) + ng * 16 + l15;
-#pragma unroll
-        for (int kc = 0; kc < 2; ++kc)
-          bf[ng][kc] = *(const bf16x8*)(bh + colh * BK + (((kc * 4 + l4) ^ (colh & 7)) * 8));
-      }
-      refill(g, ph);
-      __builtin_amdgcn_s_barrier();
-      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-      __builtin_amdgcn_s_setprio(1);
-#pragma unroll
-      for (int kc = 0; kc < 2; ++kc)
-#pragma unroll
-        for (int fg = 0; fg < 4; ++fg)
-#pragma unroll
-          for (int ng = 0; ng < NF; ++ng)
-            acc[qm][qn][fg][ng] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                af[fg][kc], bf[ng][kc], acc[qm][qn][fg][ng], 0, 0, 0);
-      __builtin_amdgcn_s_setprio(0);
-      __builtin_amdgcn_s_barrier();
     }
   }
 
@@ -741,12 +777,12 @@ __global__ void gemm256_nt_kernel(const bf16* __restrict__ A,
 #pragma unroll
     for (int qn = 0; qn < 2; ++qn)
 #pragma unroll
-      for (int fg = 0; fg < 4; ++fg)
+      for (int fg = 0; fg < MF; ++fg)
 #pragma unroll
         for (int ng = 0; ng < NF; ++ng)
 #pragma unroll
           for (int r = 0; r < 4; ++r)
-            etile[(qm * 128 + wrq * 64 + fg * 16 + l4 * 4 + r) * BN +
+            etile[(qm * (BMT / 2) + wrq * (BMT / 4) + fg * 16 + l4 * 4 + r) * BN +
                   qn * (BNT / 2) + wcq * (BNT / 8) + ng * 16 + l15] =
                 f2bf(acc[qm][qn][fg][ng][r]);
   __syncthreads();
@@ -957,43 +993,57 @@ extern "C" void al_conv2d_mm(int mode, const void* A, const void* B, void* out,
     const int bnt = (sh.Nout % 256 == 0) ? 256 : 128;
     const bool pure = (R == 1 && S == 1 && stride == 1 && pad == 0);
     const int fast = (mode == MODE_FWD) ? sh.C : sh.K;
-    const int grid_m = (int)((sh.M + 255) / 256);
+    // tile height: the 512-row tall tile restores the 256² tile's per-phase
+    // MFMA density for gathered 128-wide layers, but MEASURED ~20% slower
+    // than the 2-barrier 128² kernel anyway (per-kc fragment re-reads double
+    // the LDS traffic and the gather state grows to 8 slots) — kept for
+    // round-2 tuning behind AL_GEMM256_TALL=1. Pure bwd-data at 128 wide
+    // keeps the 256-row tile (measured +16%).
+    static int tall = -1;
+    if (tall < 0) {
+      const char* e = getenv("AL_GEMM256_TALL");
+      tall = (e && e[0] == '1') ? 1 : 0;
+    }
+    const int bmt = (tall && bnt == 128 && !pure) ? 512 : 256;
+    const int grid_m = (int)((sh.M + bmt - 1) / bmt);
     const int grid_n = sh.Nout / bnt;
     const bool gather_ok = !pure && fast % 8 == 0 &&
-                           grid_m * grid_n >= min_grid;
-    // measured (tools/ab_gemm256.py): the narrow tile only beats the 128^2
-    // igemm on the pure bwd-data shapes (+16% on l2.conv1); the gathered
-    // 3x3 at BNT=128 regresses ~30% (8 MFMA/phase cannot cover the phase
-    // overhead), so those keep the 2-barrier 128^2 kernel.
-    const bool bnt_ok = (bnt == 256) || (pure && mode == MODE_BWD_DATA);
+                           grid_m * grid_n >= min_grid / (bmt / 256);
+    const bool bnt_ok = (bnt == 256) ||
+                        (pure && mode == MODE_BWD_DATA) ||
+                        (tall && !pure);
     if ((pure || gather_ok) && bnt_ok) {
       static bool attr_set = false;
       if (!attr_set) {
-#define SET_ATTR(MODE_, BNT_) (void)hipFuncSetAttribute( \
-        (const void*)gemm256_nt_kernel<MODE_, BNT_>, \
-        hipFuncAttributeMaxDynamicSharedMemorySize, 131072)
-        SET_ATTR(MODE_PURE, 256); SET_ATTR(MODE_FWD, 256);
-        SET_ATTR(MODE_BWD_DATA, 256);
-        SET_ATTR(MODE_PURE, 128); SET_ATTR(MODE_FWD, 128);
-        SET_ATTR(MODE_BWD_DATA, 128);
+#define SET_ATTR(MODE_, BNT_, BMT_) (void)hipFuncSetAttribute( \
+        (const void*)gemm256_nt_kernel<MODE_, BNT_, BMT_>, \
+        hipFuncAttributeMaxDynamicSharedMemorySize, 163840)
+        SET_ATTR(MODE_PURE, 256, 256); SET_ATTR(MODE_FWD, 256, 256);
+        SET_ATTR(MODE_BWD_DATA, 256, 256);
+        SET_ATTR(MODE_PURE, 128, 256);
+        SET_ATTR(MODE_FWD, 128, 512); SET_ATTR(MODE_BWD_DATA, 128, 512);
 #undef SET_ATTR
         attr_set = true;
       }
-      // LDS: A ring 64 KiB + B ring 4 * (bnt/2) * 64 * 2B
-      const size_t lds = 4 * 128 * 64 * 2 + 4 * (size_t)(bnt / 2) * 64 * 2;
+      // LDS: A ring 4 * (bmt/2) * 64 * 2B + B ring 4 * (bnt/2) * 64 * 2B
+      const size_t lds = 4 * (size_t)(bmt / 2) * 64 * 2 +
+                         4 * (size_t)(bnt / 2) * 64 * 2;
       dim3 grid(grid_m * grid_n), block(512);
-#define LAUNCH256(MODE_, BNT_) hipLaunchKernelGGL((gemm256_nt_kernel<MODE_, BNT_>), \
+#define LAUNCH256(MODE_, BNT_, BMT_) hipLaunchKernelGGL( \
+      (gemm256_nt_kernel<MODE_, BNT_, BMT_>), \
       grid, block, lds, stream, (const bf16*)A, (const bf16*)B, (bf16*)out, \
       (const bf16*)zero_page, sh, grid_m, epi_scale, epi_shift, \
       (const bf16*)epi_res, epi_relu, stat_sum, stat_sumsq)
       if (bnt == 256) {
-        if (pure) LAUNCH256(MODE_PURE, 256);
-        else if (mode == MODE_FWD) LAUNCH256(MODE_FWD, 256);
-        else LAUNCH256(MODE_BWD_DATA, 256);
+        if (pure) LAUNCH256(MODE_PURE, 256, 256);
+        else if (mode == MODE_FWD) LAUNCH256(MODE_FWD, 256, 256);
+        else LAUNCH256(MODE_BWD_DATA, 256, 256);
+      } else if (pure) {
+        LAUNCH256(MODE_PURE, 128, 256);
+      } else if (mode == MODE_FWD) {
+        LAUNCH256(MODE_FWD, 128, 512);
       } else {
-        if (pure) LAUNCH256(MODE_PURE, 128);
-        else if (mode == MODE_FWD) LAUNCH256(MODE_FWD, 128);
-        else LAUNCH256(MODE_BWD_DATA, 128);
+        LAUNCH256(MODE_BWD_DATA, 128, 512);
       }
 #undef LAUNCH256
       return;

@@ -52,6 +52,10 @@ CONV_CASES = [
     # 256x128 tile (Nout % 256 != 0; routed for pure bwd-data only):
     (4, 28, 28, 256, 128, 1, 1, 0),   # fwd falls back to 128^2
     (4, 28, 28, 128, 256, 1, 1, 0),   # bwd-data routes BNT=128 (Nout=C=128)
+    # 512x128 tall tile shapes (AL_GEMM256_TALL opt-in; default takes 128^2 —
+    # both paths must be numerically correct):
+    (16, 56, 56, 64, 128, 3, 1, 1),
+    (16, 56, 56, 128, 64, 3, 1, 1),
 ]
 
 

@@ -36,7 +36,7 @@ SHAPES = [
     ("l3.conv2", 256, 14, 256, 256, 3),   # gathered 256^2 (grid 196)
     ("l4.conv2", 256, 7, 512, 512, 3),    # grid 98 < 192: stays 128^2
     ("l2.conv1", 256, 28, 256, 128, 1),   # 256x128 tile (pure)
-    ("l2.conv2", 256, 28, 128, 128, 3),   # 256x128 tile (gathered)
+    ("l2.conv2", 256, 28, 128, 128, 3),   # 512x128 tall tile (gathered)
 ]
 
 
