@@ -108,3 +108,19 @@ def test_imbalanced_e2e_round(tmp_path, cifar_dir):
     assert s.idxs_lb.sum() == 5
     w = s.generate_imbalanced_training_weights()
     assert abs(w.sum().item() - 1.0) < 1e-5
+
+
+def test_synthetic_dataset_with_loader_workers():
+    """Datasets must survive DataLoader worker processes (the real arg pools
+    use num_workers=12; synthetic sets must pickle into workers too)."""
+    import torch
+    from active_learning_amd.data.synthetic import get_data_synthetic
+
+    train, test, al = get_data_synthetic(10, 64, 16, (3, 8, 8), seed=3)
+    loader = torch.utils.data.DataLoader(train, batch_size=16, num_workers=2,
+                                         persistent_workers=False)
+    seen = 0
+    for x, y, idx in loader:
+        seen += x.shape[0]
+        assert x.shape[1:] == (3, 8, 8)
+    assert seen == 64
