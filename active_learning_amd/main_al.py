@@ -108,26 +108,36 @@ def main(args):
         strategy.round = rd
         logger.info(f"Active Learning Round {rd} start.")
 
+        phase_times = {}
         al_round_0 = rd == 0 and init_pool_size == 0
         if rd > 0 or al_round_0:
             if al_round_0:
                 strategy.init_network_weights()
             t0 = time()
             labeled_idxs, cur_cost = strategy.query(args.round_budget)
-            print(f"Rd {rd} query_time is {time() - t0}")
+            phase_times[f"rd_{rd}_query_time_s"] = time() - t0
+            print(f"Rd {rd} query_time is {phase_times[f'rd_{rd}_query_time_s']}")
             strategy.update(labeled_idxs, cur_cost)
 
         t0 = time()
         strategy.init_network_weights()
-        print(f"Rd {rd} init_network_weights_time is {time() - t0}")
+        phase_times[f"rd_{rd}_init_weights_time_s"] = time() - t0
+        print(f"Rd {rd} init_network_weights_time is "
+              f"{phase_times[f'rd_{rd}_init_weights_time_s']}")
 
         t0 = time()
         strategy.train()
-        print(f"Rd {rd} train_time is {time() - t0}")
+        phase_times[f"rd_{rd}_train_time_s"] = time() - t0
+        print(f"Rd {rd} train_time is {phase_times[f'rd_{rd}_train_time_s']}")
 
         t0 = time()
         strategy.load_best_ckpt()
-        print(f"Rd {rd} load_best_ckpt_time is {time() - t0}")
+        phase_times[f"rd_{rd}_load_best_ckpt_time_s"] = time() - t0
+        print(f"Rd {rd} load_best_ckpt_time is "
+              f"{phase_times[f'rd_{rd}_load_best_ckpt_time_s']}")
+        # the reference only prints these spans (main_al.py:160-178); logging
+        # them to the tracker makes round wall-clock auditable after the fact
+        strategy.comet_experiment.log_metrics(phase_times, step=rd)
 
         t0 = time()
         strategy.test()

@@ -56,3 +56,20 @@ def test_disabled_writes_nothing(tmp_path):
     exp.log_metric("x", 1)
     exp.log_asset_data("d", name="a")
     assert not os.path.exists(os.path.join(str(tmp_path), f"metrics_{exp.key}.jsonl"))
+
+
+def test_phase_times_logged(tmp_path):
+    """main_al logs per-phase wall-clock spans to the tracker each round."""
+    from active_learning_amd.cli import get_args
+    from active_learning_amd.main_al import main
+
+    args = get_args([
+        "--dataset", "synthetic_cifar10", "--rounds", "2", "--round_budget", "10",
+        "--n_epoch", "1", "--early_stop_patience", "1", "--debug_mode",
+        "--ckpt_path", str(tmp_path / "c"), "--log_dir", str(tmp_path / "l"),
+        "--model", "SSLResNet18"])
+    s = main(args)
+    recs = _records(str(tmp_path / "l"), s.comet_experiment.key)
+    names = {r.get("name") for r in recs if r.get("kind") == "metric"}
+    assert {"rd_0_train_time_s", "rd_0_load_best_ckpt_time_s",
+            "rd_1_query_time_s", "rd_1_train_time_s"} <= names
