@@ -72,8 +72,12 @@ def main(args):
         args.world_size = max(1, n_devices)
 
     if not args.resume_training:
-        experiment = Experiment(project_name=args.project_name,
-                                disabled=not args.enable_comet, log_dir=args.log_dir)
+        # the local JSONL sink is always on (the reference's Comet metrics
+        # contract backed locally); --enable_comet additionally mirrors to a
+        # real Comet experiment when comet_ml is importable
+        experiment = Experiment(project_name=args.project_name, disabled=False,
+                                mirror_comet=args.enable_comet,
+                                log_dir=args.log_dir)
         experiment.add_tag(args.exp_name)
         experiment.add_tag(args.strategy)
         exp_hash = os.path.basename(os.path.normpath(experiment.url))[:9]

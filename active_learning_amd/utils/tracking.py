@@ -30,7 +30,7 @@ class Experiment:
     add_tag, set_name, get_key, .url)."""
 
     def __init__(self, project_name="active-learning", disabled=False, log_dir="./logs",
-                 experiment_key=None, **_ignored):
+                 experiment_key=None, mirror_comet=True, **_ignored):
         self.project_name = project_name
         self.disabled = disabled
         self.log_dir = log_dir
@@ -39,7 +39,7 @@ class Experiment:
         self.tags = []
         self._fh = None
         self._comet = None
-        if not disabled and _HAS_COMET:
+        if not disabled and mirror_comet and _HAS_COMET:
             try:  # pragma: no cover — comet not present in CI
                 self._comet = comet_ml.Experiment(project_name=project_name,
                                                   auto_param_logging=False,
