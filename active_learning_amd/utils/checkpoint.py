@@ -154,7 +154,8 @@ def load_experiment(args, check_args_match=True):
     with open(f"{prefix}/strategy.pick", "rb") as fh:
         strategy = pickle.load(fh)
     experiment = ExistingExperiment(previous_experiment=status["comet_exp_key"],
-                                    log_dir=args.log_dir)
+                                    log_dir=args.log_dir,
+                                    mirror_comet=getattr(args, "enable_comet", False))
     experiment.add_tag(args.exp_name)
     experiment.add_tag(args.strategy)
     strategy.comet_experiment = experiment
