@@ -251,3 +251,26 @@ def test_shard_gating(monkeypatch, tmp_path):
     assert not _should_shard(s, idxs[:10])   # too small to amortize the spawn
     monkeypatch.setenv("AL_SHARD_QUERY", "0")
     assert not _should_shard(s, idxs)        # kill switch
+
+
+def test_query_deterministic_under_seed(tmp_path):
+    """Same seeds -> same selection (reproducible experiments; the reference
+    relies on seeded pool init + torch/np RNG for repeatability)."""
+    import random
+    import numpy as np
+    import torch
+    from active_learning_amd.strategies import MarginSampler
+    from helpers import make_strategy
+
+    def one_run():
+        random.seed(7)
+        np.random.seed(7)
+        torch.manual_seed(7)
+        s = make_strategy(MarginSampler, ckpt_path=str(tmp_path))
+        s.update(np.arange(10), 10)
+        idxs, cost = s.query(8)
+        return list(np.asarray(idxs).ravel()), cost
+
+    a, ca = one_run()
+    b, cb = one_run()
+    assert a == b and ca == cb
