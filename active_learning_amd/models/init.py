@@ -27,11 +27,10 @@ def init_params(net):
             if m.bias is not None:
                 m.bias.zero_()
         elif isinstance(m, (BatchNormAct2d, nn.BatchNorm2d, nn.BatchNorm1d)):
+            # reference init_params touches only BN weight/bias, NOT the
+            # running stats (src/models/utils.py:10-13) — keep that behavior
             m.weight.fill_(1.0)
             m.bias.zero_()
-            if hasattr(m, "running_mean"):
-                m.running_mean.zero_()
-                m.running_var.fill_(1.0)
         elif isinstance(m, nn.Linear):
             m.weight.normal_(0, 1e-3)
             if m.bias is not None:

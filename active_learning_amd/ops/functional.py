@@ -142,6 +142,7 @@ class _GradArena:
         self.pending = []
         self.flat = None
         self.dirty = False
+        self.touched = set()  # keys taken since the last rebuild/zero
 
     def mark_step(self):
         self.dirty = True
@@ -154,7 +155,9 @@ class _GradArena:
                 self._rebuild(device)
             elif self.flat is not None:
                 self.flat.zero_()
+                self.touched = set()
             self.dirty = False
+        self.touched.add(key)
         v = self.views.get(key)
         if v is not None:
             return v
@@ -163,9 +166,15 @@ class _GradArena:
         return None
 
     def _rebuild(self, device):
+        # evict keys not used since the previous rebuild/zero: per-round
+        # weight re-init allocates fresh bf16 weight copies (new data_ptr
+        # keys), so without eviction the flat buffer grows every AL round
+        live = self.touched | {k for k, _ in self.pending}
+        self.shapes = {k: s for k, s in self.shapes.items() if k in live}
         for key, shape in self.pending:
             self.shapes[key] = shape
         self.pending = []
+        self.touched = set()
         total = sum(int(np.prod(s)) for s in self.shapes.values())
         self.flat = torch.zeros(total, dtype=torch.float32, device=device)
         off = 0

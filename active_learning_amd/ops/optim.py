@@ -31,6 +31,8 @@ class FusedSGD(torch.optim.Optimizer):
                  nesterov=False):
         if nesterov:
             raise ValueError("nesterov not supported")
+        if dampening != 0:
+            raise ValueError("dampening not supported")
         defaults = dict(lr=lr, momentum=momentum, weight_decay=weight_decay,
                         dampening=dampening)
         super().__init__(params, defaults)

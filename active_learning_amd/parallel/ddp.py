@@ -7,16 +7,25 @@ Replaces the reference's torch DistributedDataParallel usage
   RCCL on ROCm; tests use "gloo" on CPU);
 * parameters/buffers broadcast from rank 0 at construction (the reference's
   implicit DDP ctor broadcast, SURVEY.md §2.5 row 3);
-* gradients are packed into flat fp32 buckets in reverse registration order
-  (the order backward produces them) and each bucket's all-reduce is launched
-  asynchronously as soon as its last grad lands — overlapping communication
-  with the rest of backward. xGMI is 7 point-to-point links x ~153 GB/s per
-  GPU; several medium buckets in flight keep RCCL's channels busy, so the
-  default bucket is 16 MB rather than NCCL-on-NVSwitch's 25 MB.
+* **flat-gradient buckets, reduced in place**: before backward each param's
+  ``.grad`` is installed as a view into a flat fp32 bucket buffer, so
+  backward accumulates directly into the communication buffer and the
+  optimizer reads the averaged gradient from the same storage — zero
+  copy-in/copy-out passes (torch DDP copies grads into buckets and back).
+  RCCL's ProcessGroupNCCL launches each async all-reduce on its own internal
+  HIP stream, so communication overlaps the remainder of backward by
+  construction; ``Work.wait()`` in finalize only blocks the compute stream
+  when a bucket is still in flight.
+* bucket boundaries are rebuilt after the first backward from the ORDER the
+  gradients actually arrived (autograd's execution order), so each bucket's
+  all-reduce launches as early as possible. xGMI is 7 point-to-point links
+  x ~153 GB/s per GPU; several medium buckets in flight keep RCCL's channels
+  busy, so the default bucket is 16 MB rather than NCCL-on-NVSwitch's 25 MB.
 * parameters that never receive a gradient (e.g. a frozen backbone under
-  --freeze_feature, where the embedding is detached) are handled correctly by
-  construction: their bucket slots are zero-filled at finalize time instead
-  of waiting on hooks that never fire — no find_unused_parameters hang.
+  --freeze_feature, where the embedding is detached) contribute zeros to the
+  collective (buffers are zeroed at reset) and get ``p.grad = None`` restored
+  at finalize, matching torch DDP find_unused_parameters semantics — the
+  optimizer must skip them, not apply weight-decay/momentum to a zero grad.
 
 The wrapped module is exposed as ``self.module`` so state_dict keys carry the
 "module." prefix, matching the reference's checkpoint format
@@ -29,7 +38,8 @@ import torch.nn as nn
 
 
 class _Bucket:
-    __slots__ = ("params", "buffer", "views", "pending", "work", "launched")
+    __slots__ = ("params", "buffer", "views", "pending", "ready", "work",
+                 "launched")
 
     def __init__(self, params, device):
         self.params = params
@@ -41,11 +51,14 @@ class _Bucket:
             self.views.append(self.buffer[off:off + p.numel()].view(p.shape))
             off += p.numel()
         self.pending = set()
+        self.ready = set()
         self.work = None
         self.launched = False
 
     def reset(self):
+        self.buffer.zero_()
         self.pending = set(range(len(self.params)))
+        self.ready = set()
         self.work = None
         self.launched = False
 
@@ -57,15 +70,18 @@ class BucketedDDP(nn.Module):
         self.module = module
         self.pg = process_group  # None -> default group
         self.world_size = dist.get_world_size(process_group) if dist.is_initialized() else 1
+        self.bucket_cap_mb = bucket_cap_mb
         self._hooks = []
         self._params = [p for p in module.parameters() if p.requires_grad]
         self._param_to_loc = {}
         self.buckets = []
         if self.world_size > 1 and broadcast_params:
             self._broadcast_module()
-        self._build_buckets(bucket_cap_mb)
+        self._build_buckets(self._params, bucket_cap_mb)
         self._register_hooks()
         self._in_backward = False
+        self._ready_order = []      # params in observed backward order
+        self._order_final = False   # buckets already match backward order
 
     # ------------------------------------------------------------------ #
     def _broadcast_module(self):
@@ -76,11 +92,13 @@ class BucketedDDP(nn.Module):
                                                  torch.int64, torch.int32):
                     dist.broadcast(t.data, src=0, group=self.pg)
 
-    def _build_buckets(self, cap_mb):
+    def _build_buckets(self, ordered_params, cap_mb):
         cap = int(cap_mb * 1024 * 1024 / 4)
         device = self._params[0].device if self._params else torch.device("cpu")
+        self.buckets = []
+        self._param_to_loc = {}
         group, size = [], 0
-        for p in reversed(self._params):  # backward order approximation
+        for p in ordered_params:
             group.append(p)
             size += p.numel()
             if size >= cap:
@@ -99,6 +117,15 @@ class BucketedDDP(nn.Module):
             h = p.register_post_accumulate_grad_hook(self._on_grad_ready)
             self._hooks.append(h)
 
+    def _install_grad_views(self):
+        """Point every param's .grad at its slot in the (zeroed) flat bucket:
+        backward's AccumulateGrad adds into the view in place, so the bucket
+        IS the gradient storage end to end."""
+        for b in self.buckets:
+            b.reset()
+            for p, v in zip(b.params, b.views):
+                p.grad = v
+
     # ------------------------------------------------------------------ #
     def _on_grad_ready(self, p):
         if not self._in_backward:
@@ -106,8 +133,14 @@ class BucketedDDP(nn.Module):
         bi, pi = self._param_to_loc[id(p)]
         bucket = self.buckets[bi]
         if pi in bucket.pending:
-            bucket.views[pi].copy_(p.grad.detach().to(torch.float32))
             bucket.pending.discard(pi)
+            bucket.ready.add(pi)
+            if not self._order_final:
+                self._ready_order.append(p)
+            if p.grad is not bucket.views[pi]:
+                # a hook/user replaced .grad with fresh storage; fold it back
+                bucket.views[pi].copy_(p.grad.detach().to(torch.float32))
+                p.grad = bucket.views[pi]
         if not bucket.pending and not bucket.launched:
             self._launch(bucket)
 
@@ -119,15 +152,16 @@ class BucketedDDP(nn.Module):
     # ------------------------------------------------------------------ #
     def forward(self, *args, **kwargs):
         if self.world_size > 1 and torch.is_grad_enabled() and self.training:
-            for b in self.buckets:
-                b.reset()
+            self._install_grad_views()
             self._in_backward = True
         return self.module(*args, **kwargs)
 
     def finalize_grads(self):
-        """Wait for in-flight buckets, launch any stragglers (zero-filling
-        slots of params that produced no grad), and write the averaged
-        gradients back. Call between loss.backward() and optimizer.step()."""
+        """Wait for in-flight buckets and launch any stragglers (their unready
+        slots hold zeros, so the collective stays uniform across ranks even if
+        rank participation of a param differs). Params that produced no grad
+        anywhere get .grad = None so the optimizer skips them. Call between
+        loss.backward() and optimizer.step()."""
         if self.world_size <= 1:
             return
         if not self._in_backward:
@@ -135,18 +169,30 @@ class BucketedDDP(nn.Module):
         self._in_backward = False
         for bucket in self.buckets:
             if not bucket.launched:
-                for pi in list(bucket.pending):
-                    bucket.views[pi].zero_()
-                bucket.pending.clear()
                 self._launch(bucket)
         for bucket in self.buckets:
             if bucket.work is not None:
                 bucket.work.wait()
-            for p, v in zip(bucket.params, bucket.views):
-                if p.grad is None:
-                    p.grad = v.clone().to(p.dtype)
-                else:
-                    p.grad.detach().copy_(v.to(p.grad.dtype))
+            for pi in bucket.pending:
+                # no local grad: leave the param out of the step entirely
+                bucket.params[pi].grad = None
+        if not self._order_final:
+            self._rebuild_from_order()
+
+    def _rebuild_from_order(self):
+        """Re-chunk buckets to the order backward actually produced grads, so
+        from the second iteration on each bucket fills (and its all-reduce
+        launches) as early as possible."""
+        self._order_final = True
+        if not self._ready_order:
+            return
+        seen = {id(p) for p in self._ready_order}
+        ordered = list(self._ready_order) + [p for p in self._params
+                                             if id(p) not in seen]
+        # preserve this step's reduced grads: they live in the OLD buckets'
+        # storage via p.grad views, which survive the rebuild untouched.
+        self._build_buckets(ordered, self.bucket_cap_mb)
+        self._ready_order = []
 
     # passthrough conveniences -------------------------------------------------
     def train(self, mode=True):

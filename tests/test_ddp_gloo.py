@@ -55,31 +55,37 @@ def _ddp_grad_check(rank, world):
     model = torch.nn.Sequential(torch.nn.Linear(10, 32), torch.nn.ReLU(),
                                 torch.nn.Linear(32, 4))
     ddp = BucketedDDP(model, bucket_cap_mb=0.0001)  # force several buckets
-    torch.manual_seed(100 + rank)
-    x = torch.randn(8, 10)
-    y = torch.randint(0, 4, (8,))
-    out = ddp(x)
-    loss = torch.nn.functional.cross_entropy(out, y)
-    loss.backward()
-    ddp.finalize_grads()
-
-    # reference: average of per-rank grads on a replica
-    torch.manual_seed(7)
-    ref = torch.nn.Sequential(torch.nn.Linear(10, 32), torch.nn.ReLU(),
-                              torch.nn.Linear(32, 4))
-    grads_accum = [torch.zeros_like(p) for p in ref.parameters()]
-    for r in range(world):
-        torch.manual_seed(100 + r)
-        xr = torch.randn(8, 10)
-        yr = torch.randint(0, 4, (8,))
-        for p in ref.parameters():
+    # two iterations: the second runs on buckets rebuilt in observed
+    # backward order (flat-grad views reinstalled after the rebuild)
+    for it in range(2):
+        torch.manual_seed(100 + 10 * it + rank)
+        x = torch.randn(8, 10)
+        y = torch.randint(0, 4, (8,))
+        for p in ddp.parameters():
             p.grad = None
-        lr_ = torch.nn.functional.cross_entropy(ref(xr), yr)
-        lr_.backward()
-        for g, p in zip(grads_accum, ref.parameters()):
-            g += p.grad / world
-    for p, g in zip(ddp.module.parameters(), grads_accum):
-        assert torch.allclose(p.grad, g, atol=1e-6), "DDP grad mismatch"
+        out = ddp(x)
+        loss = torch.nn.functional.cross_entropy(out, y)
+        loss.backward()
+        ddp.finalize_grads()
+
+        # reference: average of per-rank grads on a replica
+        torch.manual_seed(7)
+        ref = torch.nn.Sequential(torch.nn.Linear(10, 32), torch.nn.ReLU(),
+                                  torch.nn.Linear(32, 4))
+        grads_accum = [torch.zeros_like(p) for p in ref.parameters()]
+        for r in range(world):
+            torch.manual_seed(100 + 10 * it + r)
+            xr = torch.randn(8, 10)
+            yr = torch.randint(0, 4, (8,))
+            for p in ref.parameters():
+                p.grad = None
+            lr_ = torch.nn.functional.cross_entropy(ref(xr), yr)
+            lr_.backward()
+            for g, p in zip(grads_accum, ref.parameters()):
+                g += p.grad / world
+        for p, g in zip(ddp.module.parameters(), grads_accum):
+            assert torch.allclose(p.grad, g, atol=1e-6), \
+                f"DDP grad mismatch (iter {it})"
 
 
 def test_bucketed_ddp_grad_averaging():
@@ -101,10 +107,18 @@ def _ddp_unused_param_check(rank, world):
     torch.manual_seed(3)
     m = Partial()
     ddp = BucketedDDP(m, bucket_cap_mb=0.0001)
-    x = torch.randn(4, 5)
-    ddp(x).sum().backward()
-    ddp.finalize_grads()  # must not hang; unused slots zero-filled
-    assert m.used.weight.grad is not None
+    for _ in range(2):  # second iter runs on order-rebuilt buckets
+        for p in ddp.parameters():
+            p.grad = None
+        x = torch.randn(4, 5)
+        ddp(x).sum().backward()
+        ddp.finalize_grads()  # must not hang
+        assert m.used.weight.grad is not None
+        # params that produced no grad must be LEFT OUT of the step
+        # (grad None), not zero-filled — a frozen backbone under
+        # --freeze_feature must not receive weight-decay/momentum updates
+        assert m.unused.weight.grad is None
+        assert m.unused.bias.grad is None
 
 
 def test_bucketed_ddp_unused_params():
