@@ -1,0 +1,127 @@
+"""hipGraph capture of the training step for the REAL trainer.
+
+The reference's metric is round wall-clock (main_al.py:160-178); the eager
+epoch loop pays a launch gap per kernel (~200 launches/step on ResNet-50).
+GraphedTrainStep captures forward + CE + backward + fused-SGD update into one
+hipGraph after a few eager warmup steps and replays it per batch, with:
+
+* static input/target buffers (H2D copy lands outside the graph);
+* the LR schedule routed through a device hyper buffer
+  (FusedSGD.enable_device_hyper) so scheduler.step() never forces re-capture;
+* an eager fallback for odd-sized batches (the non-drop_last tail) and for
+  anything that fails capture — capture failure can never lose training;
+* post-replay cache upkeep: in-graph kernels update BN running stats and
+  bf16 weight shadows in place without bumping the python-side version
+  ticks, so replay bumps them (the folded-BN eval cache keys on _al_tick,
+  ops/fused.py:26-43).
+
+Single-process only: with BucketedDDP the RCCL all-reduce ordering is driven
+by autograd hooks on the host, which a replay skips — world_size > 1 keeps
+the eager loop (its collectives already overlap backward).
+"""
+
+import logging
+
+import torch
+
+from .functional import bump_tick
+
+logger = logging.getLogger("ActiveLearning")
+
+
+class GraphedTrainStep:
+    """Wraps (net, optimizer, criterion) into a capturable step:
+    loss = step(x, y). Capture happens lazily after `warmup` eager calls."""
+
+    def __init__(self, net, optimizer, criterion, device, warmup=3):
+        self.net = net
+        self.opt = optimizer
+        self.crit = criterion
+        self.device = device
+        self.warmup = warmup
+        self._calls = 0
+        self._graph = None
+        self._failed = False
+        self.x_static = None
+        self.y_static = None
+        self.loss_static = None
+        self._bn_tensors = []
+        for m in net.modules():
+            for name in ("weight", "bias", "running_mean", "running_var"):
+                t = getattr(m, name, None)
+                if t is not None and hasattr(m, "running_mean"):
+                    self._bn_tensors.append(t)
+        can_hyper = hasattr(optimizer, "enable_device_hyper") and \
+            len(optimizer.param_groups) == 1
+        if can_hyper:
+            optimizer.enable_device_hyper(device)
+
+    # ------------------------------------------------------------------ #
+    def _eager(self, x, y):
+        if getattr(self.opt, "_hyper_dev", None) is not None:
+            self.opt.sync_hyper()
+        x = x.to(self.device, non_blocking=True)
+        y = y.to(self.device, non_blocking=True)
+        self.opt.zero_grad(set_to_none=True)
+        out = self.net(x)
+        loss = self.crit(out, y)
+        loss.backward()
+        self.opt.step()
+        return loss
+
+    def _capture(self):
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            self.opt.zero_grad(set_to_none=True)
+            out = self.net(self.x_static)
+            loss = self.crit(out, self.y_static)
+            loss.backward()
+            self.opt.step()
+            self.loss_static = loss
+        self._graph = g
+
+    def _post_replay(self):
+        # in-graph kernels updated BN stats + weight shadows in place; bump
+        # the python-side ticks so folded-eval caches refold (fused.py)
+        for t in self._bn_tensors:
+            bump_tick(t)
+
+    # ------------------------------------------------------------------ #
+    def step(self, x, y):
+        """x, y on CPU (pinned loader) or device. Returns the loss tensor."""
+        if self._failed or self.device.type != "cuda":
+            return self._eager(x, y)
+        if self.x_static is not None and x.shape != self.x_static.shape:
+            return self._eager(x, y)  # tail batch
+        if self._graph is None:
+            if self.x_static is None:
+                self.x_static = torch.empty(x.shape, dtype=x.dtype,
+                                            device=self.device)
+                self.y_static = torch.empty(y.shape, dtype=y.dtype,
+                                            device=self.device)
+            self.x_static.copy_(x)
+            self.y_static.copy_(y)
+            if self._calls < self.warmup:
+                self._calls += 1
+                return self._eager(self.x_static, self.y_static)
+            try:
+                torch.cuda.synchronize()
+                self._capture()
+                logger.info("hipGraph-captured the training step "
+                            f"(batch {tuple(x.shape)})")
+            except Exception as e:
+                logger.warning(f"hipGraph capture unavailable ({e!r}); "
+                               "training eagerly")
+                self._failed = True
+                self._graph = None
+                return self._eager(self.x_static, self.y_static)
+            self._graph.replay()
+            self._post_replay()
+            return self.loss_static
+        if getattr(self.opt, "_hyper_dev", None) is not None:
+            self.opt.sync_hyper()  # track the LR schedule between replays
+        self.x_static.copy_(x, non_blocking=True)
+        self.y_static.copy_(y, non_blocking=True)
+        self._graph.replay()
+        self._post_replay()
+        return self.loss_static

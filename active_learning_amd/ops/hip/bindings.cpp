@@ -60,6 +60,10 @@ void al_sgd_step(float* p, const float* g, float* buf, float lr, float momentum,
 void al_adam_step(float* p, const float* g, float* m, float* v, float lr, float b1,
                   float b2, float eps, float wd, float bc1, float bc2, long n,
                   void* stream);
+void al_sgd_step_multi(const void* table, int nchunks, float lr, float momentum,
+                       float wd, void* stream);
+void al_sgd_step_multi_dev(const void* table, int nchunks, const float* hyper,
+                           void* stream);
 }
 
 namespace {
@@ -419,6 +423,26 @@ void sgd_step(Tensor& p, const Tensor& g, Tensor& buf, double lr, double momentu
               (float)momentum, (float)wd, p.numel(), sh, cur_stream());
 }
 
+void sgd_step_multi(const Tensor& table, int64_t nchunks, double lr,
+                    double momentum, double wd) {
+  TORCH_CHECK(table.is_cuda() && table.scalar_type() == torch::kInt64 &&
+              table.is_contiguous());
+  TORCH_CHECK(table.numel() >= nchunks * 6, "chunk table too small");
+  al_sgd_step_multi(table.data_ptr<int64_t>(), (int)nchunks, (float)lr,
+                    (float)momentum, (float)wd, cur_stream());
+}
+
+void sgd_step_multi_dev(const Tensor& table, int64_t nchunks,
+                        const Tensor& hyper) {
+  TORCH_CHECK(table.is_cuda() && table.scalar_type() == torch::kInt64 &&
+              table.is_contiguous());
+  TORCH_CHECK(table.numel() >= nchunks * 6, "chunk table too small");
+  TORCH_CHECK(hyper.is_cuda() && hyper.scalar_type() == torch::kFloat32 &&
+              hyper.numel() >= 3);
+  al_sgd_step_multi_dev(table.data_ptr<int64_t>(), (int)nchunks,
+                        hyper.data_ptr<float>(), cur_stream());
+}
+
 void adam_step(Tensor& p, const Tensor& g, Tensor& m, Tensor& v, double lr, double b1,
                double b2, double eps, double wd, double bc1, double bc2) {
   TORCH_CHECK(p.is_cuda() && p.scalar_type() == torch::kFloat32);
@@ -451,5 +475,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ce_fwd", &ce_fwd);
   m.def("ce_bwd", &ce_bwd);
   m.def("sgd_step", &sgd_step);
+  m.def("sgd_step_multi", &sgd_step_multi);
+  m.def("sgd_step_multi_dev", &sgd_step_multi_dev);
   m.def("adam_step", &adam_step);
 }

@@ -34,6 +34,81 @@ extern "C" void al_sgd_step(float* p, const float* g, float* buf, float lr,
                        buf, lr, momentum, wd, n, nullptr);
 }
 
+// ---------------------------------------------------------------------------
+// Multi-tensor SGD: ONE launch updates every parameter (the per-tensor
+// variant cost 161 launches/step on ResNet-50, profiles/
+// r50_b256_step_breakdown_final.md). The host packs a chunk table
+// (6 int64 per chunk: p, g, momentum-buf, bf16-shadow pointers + element
+// offset + count); each workgroup owns one chunk. Scalar fp32 accesses:
+// grads may be element-aligned views into a DDP bucket, so 16-byte vector
+// loads are not safe, and the pass is HBM-bound anyway.
+struct SgdChunk {
+  float* p;
+  const float* g;
+  float* buf;
+  bf16* shadow;
+  long off;
+  long n;
+};
+
+__global__ void sgd_mt_kernel(const SgdChunk* __restrict__ table, float lr,
+                              float momentum, float wd) {
+  SgdChunk e = table[blockIdx.x];
+  float* p = e.p + e.off;
+  const float* g = e.g + e.off;
+  float* buf = e.buf ? e.buf + e.off : nullptr;
+  bf16* sh = e.shadow ? e.shadow + e.off : nullptr;
+  const int n = (int)e.n;
+  for (int i = threadIdx.x; i < n; i += blockDim.x) {
+    float grad = g[i] + wd * p[i];
+    if (momentum != 0.f) {
+      const float b = buf[i] * momentum + grad;
+      buf[i] = b;
+      grad = b;
+    }
+    const float np = p[i] - lr * grad;
+    p[i] = np;
+    if (sh) sh[i] = f2bf(np);
+  }
+}
+
+extern "C" void al_sgd_step_multi(const void* table, int nchunks, float lr,
+                                  float momentum, float wd, hipStream_t stream) {
+  hipLaunchKernelGGL(sgd_mt_kernel, dim3(nchunks), dim3(256), 0, stream,
+                     (const SgdChunk*)table, lr, momentum, wd);
+}
+
+// Variant reading (lr, momentum, wd) from device memory: lets a hipGraph-
+// captured training step track the LR schedule — the host updates the
+// 3-float hyper buffer between replays instead of re-capturing.
+__global__ void sgd_mt_kernel_dev(const SgdChunk* __restrict__ table,
+                                  const float* __restrict__ hyper) {
+  const float lr = hyper[0], momentum = hyper[1], wd = hyper[2];
+  SgdChunk e = table[blockIdx.x];
+  float* p = e.p + e.off;
+  const float* g = e.g + e.off;
+  float* buf = e.buf ? e.buf + e.off : nullptr;
+  bf16* sh = e.shadow ? e.shadow + e.off : nullptr;
+  const int n = (int)e.n;
+  for (int i = threadIdx.x; i < n; i += blockDim.x) {
+    float grad = g[i] + wd * p[i];
+    if (momentum != 0.f) {
+      const float b = buf[i] * momentum + grad;
+      buf[i] = b;
+      grad = b;
+    }
+    const float np = p[i] - lr * grad;
+    p[i] = np;
+    if (sh) sh[i] = f2bf(np);
+  }
+}
+
+extern "C" void al_sgd_step_multi_dev(const void* table, int nchunks,
+                                      const float* hyper, hipStream_t stream) {
+  hipLaunchKernelGGL(sgd_mt_kernel_dev, dim3(nchunks), dim3(256), 0, stream,
+                     (const SgdChunk*)table, hyper);
+}
+
 __global__ void adam_kernel(float* __restrict__ p, const float* __restrict__ g,
                             float* __restrict__ m, float* __restrict__ v, float lr,
                             float b1, float b2, float eps, float wd, float bc1,

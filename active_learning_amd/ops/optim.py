@@ -27,6 +27,8 @@ def _invalidate(p):
 
 
 class FusedSGD(torch.optim.Optimizer):
+    _CHUNK = 32768  # elements per workgroup in the multi-tensor kernel
+
     def __init__(self, params, lr, momentum=0.0, weight_decay=0.0, dampening=0.0,
                  nesterov=False):
         if nesterov:
@@ -36,17 +38,45 @@ class FusedSGD(torch.optim.Optimizer):
         defaults = dict(lr=lr, momentum=momentum, weight_decay=weight_decay,
                         dampening=dampening)
         super().__init__(params, defaults)
+        self._mt_cache = {}  # group idx -> (fingerprint, device table, nchunks)
+        self._mt_keepalive = []  # pinned/device tables referenced by live graphs
+        self._hyper_dev = None   # device (lr, momentum, wd); graph-capture mode
+        self._hyper_host = None
+
+    def enable_device_hyper(self, device):
+        """Switch the fused update to read (lr, momentum, wd) from device
+        memory so a hipGraph-captured step tracks LR-schedule changes without
+        re-capture; call sync_hyper() OUTSIDE capture whenever the schedule
+        steps."""
+        assert len(self.param_groups) == 1, "device-hyper mode: single group"
+        self._hyper_host = torch.zeros(3, dtype=torch.float32).pin_memory()
+        self._hyper_dev = torch.zeros(3, dtype=torch.float32, device=device)
+        self.sync_hyper()
+
+    def sync_hyper(self):
+        g = self.param_groups[0]
+        vals = (float(g["lr"]), float(g["momentum"]), float(g["weight_decay"]))
+        if tuple(self._hyper_host.tolist()) != vals:
+            self._hyper_host.copy_(torch.tensor(vals))
+            self._hyper_dev.copy_(self._hyper_host, non_blocking=True)
 
     @torch.no_grad()
     def step(self, closure=None):
         loss = closure() if closure is not None else None
         ext = load_extension() if extension_available() else None
-        for group in self.param_groups:
+        for gi, group in enumerate(self.param_groups):
             lr = group["lr"]
             momentum = group["momentum"]
             wd = group["weight_decay"]
+            fused = []
             for p in group["params"]:
                 if p.grad is None:
+                    continue
+                if (ext is not None and p.is_cuda
+                        and p.dtype == torch.float32 and p.is_contiguous()
+                        and p.grad.dtype == torch.float32
+                        and p.grad.is_contiguous()):
+                    fused.append(p)
                     continue
                 g = p.grad
                 state = self.state[p]
@@ -79,7 +109,64 @@ class FusedSGD(torch.optim.Optimizer):
                         buf.mul_(momentum).add_(gf)
                         gf = buf
                     p.add_(gf, alpha=-lr)
+            if fused:
+                self._multi_tensor_step(ext, gi, fused, lr, momentum, wd)
         return loss
+
+    def _multi_tensor_step(self, ext, gi, params, lr, momentum, wd):
+        """ONE kernel launch for the whole parameter set (the per-tensor path
+        cost 161 launches/step on ResNet-50). Packs a chunk table of
+        (p, grad, momentum-buf, bf16-shadow, offset, count) int64 rows; the
+        device copy is cached and only refreshed when any pointer moves, so
+        steady-state steps (and hipGraph captures) launch with no H2D
+        traffic."""
+        fp = []
+        post = []
+        for p in params:
+            state = self.state[p]
+            if momentum != 0 and "momentum_buffer" not in state:
+                state["momentum_buffer"] = torch.zeros_like(p, dtype=torch.float32)
+            buf = state.get("momentum_buffer")
+            cache = getattr(p, "_al_cast", None)
+            shadow = (cache[1] if cache is not None
+                      and cache[1].numel() == p.numel()
+                      and cache[1].dtype == torch.bfloat16 else None)
+            fp.append((p.data_ptr(), p.grad.data_ptr(),
+                       buf.data_ptr() if buf is not None else 0,
+                       shadow.data_ptr() if shadow is not None else 0,
+                       p.numel()))
+            post.append((p, shadow))
+        fp = tuple(fp)
+        entry = self._mt_cache.get(gi)
+        if entry is None or entry[0] != fp:
+            rows = []
+            for pp, gp, bp, sp, n in fp:
+                off = 0
+                while off < n:
+                    rows.append((pp, gp, bp, sp, off, min(self._CHUNK, n - off)))
+                    off += self._CHUNK
+            # pinned staging + async copy: legal inside hipGraph capture
+            # (grad pointers can move when backward allocates from a capture
+            # pool); the pinned buffer must stay alive as long as any graph
+            # that captured the copy, hence the keepalive list.
+            host = torch.tensor(rows, dtype=torch.int64).pin_memory()
+            dev = torch.empty_like(host, device=params[0].device)
+            dev.copy_(host, non_blocking=True)
+            self._mt_keepalive.append((host, dev))
+            entry = (fp, dev, len(rows))
+            self._mt_cache[gi] = entry
+        if self._hyper_dev is not None:
+            ext.sgd_step_multi_dev(entry[1], entry[2], self._hyper_dev)
+        else:
+            ext.sgd_step_multi(entry[1], entry[2], lr, momentum, wd)
+        for p, shadow in post:
+            if shadow is not None:
+                if getattr(shadow, "_al_wt", None) is not None:
+                    shadow._al_wt = None
+                p._al_cast = (p._version, shadow)
+            else:
+                _invalidate(p)
+            bump_tick(p)
 
 
 class FusedAdam(torch.optim.Optimizer):

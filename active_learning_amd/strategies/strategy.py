@@ -314,6 +314,15 @@ class Strategy:
         print(f"Rank {rank} training starts.")
         self.logger.info(f"Starting training on round {self.round}")
 
+        # hipGraph-capture the train step (fwd+CE+bwd+fused SGD in one
+        # replay, zero launch gaps). Single-process only: the DDP all-reduce
+        # schedule is host-hook-driven. AL_TRAIN_GRAPH=0 disables.
+        graphed = None
+        if (self.world_size == 1 and device.type == "cuda"
+                and os.environ.get("AL_TRAIN_GRAPH", "1") == "1"):
+            from ..ops.graph import GraphedTrainStep
+            graphed = GraphedTrainStep(self.net, optimizer, criterion, device)
+
         for epoch in range(1, self.n_epoch + 1):
             if train_sampler is not None:
                 train_sampler.set_epoch(epoch)
@@ -323,7 +332,8 @@ class Strategy:
             if self.freeze_feature or ("init_pretrained_ckpt_path" in self.train_args):
                 self.net.eval()
             step = self._train(rank=rank, epoch=epoch, loader_tr=loader_tr,
-                               optimizer=optimizer, criterion=criterion, step=step)
+                               optimizer=optimizer, criterion=criterion, step=step,
+                               graphed=graphed)
             scheduler.step()
             if self.validation_and_early_stopping(rank, epoch, weight_paths):
                 break
@@ -336,11 +346,17 @@ class Strategy:
         if self.world_size > 1 and isinstance(self.net, BucketedDDP):
             self.net = self.net.module
 
-    def _train(self, rank, epoch, loader_tr, optimizer, criterion, step):
+    def _train(self, rank, epoch, loader_tr, optimizer, criterion, step,
+               graphed=None):
         """One training epoch — the hot loop (strategy.py:249-284). The loss
         stays on-device between log points: a per-batch .cpu() would force a
         full pipeline sync every iteration."""
         for batch_idx, (x, y, _idxs) in enumerate(loader_tr):
+            if graphed is not None:
+                loss = graphed.step(x, y)
+                self._log_train_batch(rank, epoch, batch_idx, loader_tr, loss)
+                step += 1
+                continue
             x = x.to(self.device, non_blocking=True)
             y = y.to(self.device, non_blocking=True)
             optimizer.zero_grad(set_to_none=True)
@@ -350,16 +366,19 @@ class Strategy:
             if isinstance(self.net, BucketedDDP):
                 self.net.finalize_grads()
             optimizer.step()
-            if batch_idx % 25 == 0:  # reference cadence (strategy.py:276-279)
-                cur_loss = loss.detach().float().cpu()
-                msg = (f"\tRound {self.round}, Epoch {epoch}, batch "
-                       f"{batch_idx}/{len(loader_tr)}, loss is {cur_loss} on worker "
-                       f"rank {rank}")
-                self.logger.info(msg)
-                if self.world_size == 1 or rank == 1:
-                    print(msg)
+            self._log_train_batch(rank, epoch, batch_idx, loader_tr, loss)
             step += 1
         return step
+
+    def _log_train_batch(self, rank, epoch, batch_idx, loader_tr, loss):
+        if batch_idx % 25 == 0:  # reference cadence (strategy.py:276-279)
+            cur_loss = loss.detach().float().cpu()
+            msg = (f"\tRound {self.round}, Epoch {epoch}, batch "
+                   f"{batch_idx}/{len(loader_tr)}, loss is {cur_loss} on worker "
+                   f"rank {rank}")
+            self.logger.info(msg)
+            if self.world_size == 1 or rank == 1:
+                print(msg)
 
     # -- validation / early stop (strategy.py:383-442) --------------------- #
 
