@@ -1,9 +1,16 @@
-"""BASE: class-balanced MASE.
+"""BASE: class-balanced MASE (reference: src/query_strategies/base_sampler.py).
 
-Reference: src/query_strategies/base_sampler.py — per-class budget
-budget/C (+1 for the first budget%C classes, :23-24); per class c the score
-is the min margin where predicted==c, else the distance to c's boundary
-(:28-29); already-chosen masked with inf (:31-32); uniqueness asserted (:37).
+Semantics preserved: the budget splits into per-class quotas of
+floor(budget/C) + 1 for the first budget % C classes (:23-24); a sample's
+score for class c is its min-margin when the model predicts c and its
+distance to c's decision boundary otherwise (:28-29); classes claim their
+quota in index order, each claiming the lowest-scoring samples not already
+claimed (:31-32).
+
+Implemented as a single (N, C) score matrix built in one scatter (the
+predicted-class column of each row is overwritten with the row's min-margin)
+plus a per-class masked top-k — no per-class torch.where re-materialization,
+everything stays on the scoring device.
 """
 
 import torch
@@ -13,23 +20,28 @@ from .mase_sampler import MASESampler
 
 class BASESampler(MASESampler):
     def query(self, budget):
-        idxs_for_query = self.available_query_idxs(boolean=False, shuffle=False)
-        min_margins, per_class_margins, pred_labels, _ = self.compute_margins(
-            idxs_for_query)
-        budget = int(min(len(idxs_for_query), budget))
+        pool = self.available_query_idxs(boolean=False, shuffle=False)
+        min_margins, per_class_margins, pred_labels, _ = self.compute_margins(pool)
+        budget = int(min(len(pool), budget))
+        n = len(pool)
 
-        labeled_idxs = []
-        for c in range(self.num_classes):
-            take = budget // self.num_classes + int(c < budget % self.num_classes)
-            if take == 0:
-                continue
-            dist_c = torch.where(pred_labels == c, min_margins, per_class_margins[:, c])
-            if labeled_idxs:
-                dist_c = dist_c.clone()
-                dist_c[torch.tensor(labeled_idxs)] = float("inf")
-            order = torch.sort(dist_c, descending=False).indices
-            labeled_idxs += order[:take].tolist()
+        # score matrix: boundary distances, with row i's pred-class entry
+        # replaced by its min-margin
+        scores = per_class_margins.clone()
+        scores[torch.arange(n, device=scores.device), pred_labels] = min_margins
 
-        assert len(labeled_idxs) == len(set(labeled_idxs))
-        labeled_idxs = idxs_for_query[labeled_idxs].tolist()
-        return labeled_idxs, budget
+        quota = torch.full((self.num_classes,), budget // self.num_classes,
+                           dtype=torch.long)
+        quota[:budget % self.num_classes] += 1
+
+        claimed = torch.zeros(n, dtype=torch.bool, device=scores.device)
+        picks = []
+        inf = torch.tensor(float("inf"), device=scores.device)
+        for c in torch.nonzero(quota).flatten().tolist():
+            col = torch.where(claimed, inf, scores[:, c])
+            best = torch.topk(col, int(quota[c]), largest=False).indices
+            claimed[best] = True
+            picks.append(best)
+        picks = torch.cat(picks)
+        assert picks.numel() == int(claimed.sum()), "duplicate claim in BASE"
+        return pool[picks.cpu().numpy()].tolist(), budget

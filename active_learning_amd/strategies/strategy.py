@@ -323,6 +323,9 @@ class Strategy:
             from ..ops.graph import GraphedTrainStep
             graphed = GraphedTrainStep(self.net, optimizer, criterion, device)
 
+        # subclasses that co-train auxiliary models (VAAL) hook in here
+        self._setup_aux(device, rank)
+
         for epoch in range(1, self.n_epoch + 1):
             if train_sampler is not None:
                 train_sampler.set_epoch(epoch)
@@ -331,10 +334,12 @@ class Strategy:
             # pretrained backbone (strategy.py:363-367)
             if self.freeze_feature or ("init_pretrained_ckpt_path" in self.train_args):
                 self.net.eval()
+            self._epoch_start(epoch)
             step = self._train(rank=rank, epoch=epoch, loader_tr=loader_tr,
                                optimizer=optimizer, criterion=criterion, step=step,
                                graphed=graphed)
             scheduler.step()
+            self._epoch_end()
             if self.validation_and_early_stopping(rank, epoch, weight_paths):
                 break
 
@@ -343,8 +348,26 @@ class Strategy:
         print(msg)
         self.logger.info(msg)
         self.logger.info(f"Finished training on round {self.round}")
+        self._teardown_aux()
         if self.world_size > 1 and isinstance(self.net, BucketedDDP):
             self.net = self.net.module
+
+    # -- trainer hooks for strategies that co-train auxiliary models -------- #
+
+    def _setup_aux(self, device, rank):
+        """Called once per round after the classifier is on-device/DDP-wrapped
+        and its optimizer exists; VAAL builds its VAE/discriminator stack
+        here."""
+
+    def _epoch_start(self, epoch):
+        """Called at each epoch start after the classifier's train/eval-mode
+        bookkeeping."""
+
+    def _epoch_end(self):
+        """Called after the classifier's scheduler steps."""
+
+    def _teardown_aux(self):
+        """Called when the round's training finishes (before DDP unwrap)."""
 
     def _train(self, rank, epoch, loader_tr, optimizer, criterion, step,
                graphed=None):
