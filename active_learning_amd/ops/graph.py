@@ -53,26 +53,51 @@ class GraphedTrainStep:
                     self._bn_tensors.append(t)
         can_hyper = hasattr(optimizer, "enable_device_hyper") and \
             len(optimizer.param_groups) == 1
-        if can_hyper:
+        self._persistent = False
+        if can_hyper and device.type == "cuda":
             optimizer.enable_device_hyper(device)
+            # persistent grads: storage stays put across steps (backward
+            # accumulates in place, the fused SGD zeroes after use) so the
+            # captured step is allocation-free and the chunk table is stable
+            optimizer.enable_persistent_grads()
+            self._persistent = True
 
     # ------------------------------------------------------------------ #
+    def _dealias_grads(self):
+        """Persistent grads must not ALIAS the conv-wgrad arena: backward
+        writes dw into the arena view and then accumulates p.grad += dw —
+        if p.grad IS the view, the grad doubles. Clone any aliased grad into
+        its own stable storage (runs between eager steps, outside capture)."""
+        from .functional import _grad_arena
+        flat = _grad_arena.flat
+        if flat is None:
+            return
+        lo = flat.data_ptr()
+        hi = lo + flat.numel() * flat.element_size()
+        for group in self.opt.param_groups:
+            for p in group["params"]:
+                g = p.grad
+                if g is not None and g.is_cuda and lo <= g.data_ptr() < hi:
+                    p.grad = g.detach().clone()
+
     def _eager(self, x, y):
         if getattr(self.opt, "_hyper_dev", None) is not None:
             self.opt.sync_hyper()
         x = x.to(self.device, non_blocking=True)
         y = y.to(self.device, non_blocking=True)
-        self.opt.zero_grad(set_to_none=True)
+        if not self._persistent:
+            self.opt.zero_grad(set_to_none=True)
         out = self.net(x)
         loss = self.crit(out, y)
         loss.backward()
+        if self._persistent:
+            self._dealias_grads()
         self.opt.step()
         return loss
 
     def _capture(self):
         g = torch.cuda.CUDAGraph()
         with torch.cuda.graph(g):
-            self.opt.zero_grad(set_to_none=True)
             out = self.net(self.x_static)
             loss = self.crit(out, self.y_static)
             loss.backward()

@@ -61,9 +61,9 @@ void al_adam_step(float* p, const float* g, float* m, float* v, float lr, float 
                   float b2, float eps, float wd, float bc1, float bc2, long n,
                   void* stream);
 void al_sgd_step_multi(const void* table, int nchunks, float lr, float momentum,
-                       float wd, void* stream);
+                       float wd, int zero_grad, void* stream);
 void al_sgd_step_multi_dev(const void* table, int nchunks, const float* hyper,
-                           void* stream);
+                           int zero_grad, void* stream);
 }
 
 namespace {
@@ -424,23 +424,23 @@ void sgd_step(Tensor& p, const Tensor& g, Tensor& buf, double lr, double momentu
 }
 
 void sgd_step_multi(const Tensor& table, int64_t nchunks, double lr,
-                    double momentum, double wd) {
+                    double momentum, double wd, bool zero_grad) {
   TORCH_CHECK(table.is_cuda() && table.scalar_type() == torch::kInt64 &&
               table.is_contiguous());
   TORCH_CHECK(table.numel() >= nchunks * 6, "chunk table too small");
   al_sgd_step_multi(table.data_ptr<int64_t>(), (int)nchunks, (float)lr,
-                    (float)momentum, (float)wd, cur_stream());
+                    (float)momentum, (float)wd, zero_grad ? 1 : 0, cur_stream());
 }
 
 void sgd_step_multi_dev(const Tensor& table, int64_t nchunks,
-                        const Tensor& hyper) {
+                        const Tensor& hyper, bool zero_grad) {
   TORCH_CHECK(table.is_cuda() && table.scalar_type() == torch::kInt64 &&
               table.is_contiguous());
   TORCH_CHECK(table.numel() >= nchunks * 6, "chunk table too small");
   TORCH_CHECK(hyper.is_cuda() && hyper.scalar_type() == torch::kFloat32 &&
               hyper.numel() >= 3);
   al_sgd_step_multi_dev(table.data_ptr<int64_t>(), (int)nchunks,
-                        hyper.data_ptr<float>(), cur_stream());
+                        hyper.data_ptr<float>(), zero_grad ? 1 : 0, cur_stream());
 }
 
 void adam_step(Tensor& p, const Tensor& g, Tensor& m, Tensor& v, double lr, double b1,

@@ -51,16 +51,22 @@ struct SgdChunk {
   long n;
 };
 
+// ZG (persistent-grad mode): zero the gradient in the same pass after
+// consuming it — the training step then needs NO zero_grad / grad realloc,
+// which keeps gradient storage stable so a hipGraph-captured step contains
+// no allocations and the chunk table never rebuilds inside capture.
+template <bool ZG>
 __global__ void sgd_mt_kernel(const SgdChunk* __restrict__ table, float lr,
                               float momentum, float wd) {
   SgdChunk e = table[blockIdx.x];
   float* p = e.p + e.off;
-  const float* g = e.g + e.off;
+  float* g = const_cast<float*>(e.g) + e.off;
   float* buf = e.buf ? e.buf + e.off : nullptr;
   bf16* sh = e.shadow ? e.shadow + e.off : nullptr;
   const int n = (int)e.n;
   for (int i = threadIdx.x; i < n; i += blockDim.x) {
     float grad = g[i] + wd * p[i];
+    if (ZG) g[i] = 0.f;
     if (momentum != 0.f) {
       const float b = buf[i] * momentum + grad;
       buf[i] = b;
@@ -73,25 +79,32 @@ __global__ void sgd_mt_kernel(const SgdChunk* __restrict__ table, float lr,
 }
 
 extern "C" void al_sgd_step_multi(const void* table, int nchunks, float lr,
-                                  float momentum, float wd, hipStream_t stream) {
-  hipLaunchKernelGGL(sgd_mt_kernel, dim3(nchunks), dim3(256), 0, stream,
-                     (const SgdChunk*)table, lr, momentum, wd);
+                                  float momentum, float wd, int zero_grad,
+                                  hipStream_t stream) {
+  if (zero_grad)
+    hipLaunchKernelGGL((sgd_mt_kernel<true>), dim3(nchunks), dim3(256), 0, stream,
+                       (const SgdChunk*)table, lr, momentum, wd);
+  else
+    hipLaunchKernelGGL((sgd_mt_kernel<false>), dim3(nchunks), dim3(256), 0, stream,
+                       (const SgdChunk*)table, lr, momentum, wd);
 }
 
 // Variant reading (lr, momentum, wd) from device memory: lets a hipGraph-
 // captured training step track the LR schedule — the host updates the
 // 3-float hyper buffer between replays instead of re-capturing.
+template <bool ZG>
 __global__ void sgd_mt_kernel_dev(const SgdChunk* __restrict__ table,
                                   const float* __restrict__ hyper) {
   const float lr = hyper[0], momentum = hyper[1], wd = hyper[2];
   SgdChunk e = table[blockIdx.x];
   float* p = e.p + e.off;
-  const float* g = e.g + e.off;
+  float* g = const_cast<float*>(e.g) + e.off;
   float* buf = e.buf ? e.buf + e.off : nullptr;
   bf16* sh = e.shadow ? e.shadow + e.off : nullptr;
   const int n = (int)e.n;
   for (int i = threadIdx.x; i < n; i += blockDim.x) {
     float grad = g[i] + wd * p[i];
+    if (ZG) g[i] = 0.f;
     if (momentum != 0.f) {
       const float b = buf[i] * momentum + grad;
       buf[i] = b;
@@ -104,9 +117,14 @@ __global__ void sgd_mt_kernel_dev(const SgdChunk* __restrict__ table,
 }
 
 extern "C" void al_sgd_step_multi_dev(const void* table, int nchunks,
-                                      const float* hyper, hipStream_t stream) {
-  hipLaunchKernelGGL(sgd_mt_kernel_dev, dim3(nchunks), dim3(256), 0, stream,
-                     (const SgdChunk*)table, hyper);
+                                      const float* hyper, int zero_grad,
+                                      hipStream_t stream) {
+  if (zero_grad)
+    hipLaunchKernelGGL((sgd_mt_kernel_dev<true>), dim3(nchunks), dim3(256), 0,
+                       stream, (const SgdChunk*)table, hyper);
+  else
+    hipLaunchKernelGGL((sgd_mt_kernel_dev<false>), dim3(nchunks), dim3(256), 0,
+                       stream, (const SgdChunk*)table, hyper);
 }
 
 __global__ void adam_kernel(float* __restrict__ p, const float* __restrict__ g,

@@ -42,6 +42,14 @@ class FusedSGD(torch.optim.Optimizer):
         self._mt_keepalive = []  # pinned/device tables referenced by live graphs
         self._hyper_dev = None   # device (lr, momentum, wd); graph-capture mode
         self._hyper_host = None
+        self._persistent = False  # grads kept in stable storage, zeroed in-kernel
+
+    def enable_persistent_grads(self):
+        """Persistent-grad mode for hipGraph capture: the caller must NOT call
+        zero_grad — gradient tensors keep their storage across steps (backward
+        accumulates in place) and the fused update zeroes them after
+        consumption. This keeps the captured step allocation-free."""
+        self._persistent = True
 
     def enable_device_hyper(self, device):
         """Switch the fused update to read (lr, momentum, wd) from device
@@ -109,6 +117,8 @@ class FusedSGD(torch.optim.Optimizer):
                         buf.mul_(momentum).add_(gf)
                         gf = buf
                     p.add_(gf, alpha=-lr)
+                    if self._persistent:
+                        g.zero_()
             if fused:
                 self._multi_tensor_step(ext, gi, fused, lr, momentum, wd)
         return loss
@@ -139,16 +149,19 @@ class FusedSGD(torch.optim.Optimizer):
         fp = tuple(fp)
         entry = self._mt_cache.get(gi)
         if entry is None or entry[0] != fp:
+            if torch.cuda.is_current_stream_capturing():
+                # a table rebuild needs host allocations, which invalidate
+                # hipGraph capture — fall back to per-tensor launches for
+                # this (captured) step. With persistent grads the
+                # fingerprint is stable and this path never triggers.
+                self._per_tensor_fallback(ext, params, lr, momentum, wd)
+                return
             rows = []
             for pp, gp, bp, sp, n in fp:
                 off = 0
                 while off < n:
                     rows.append((pp, gp, bp, sp, off, min(self._CHUNK, n - off)))
                     off += self._CHUNK
-            # pinned staging + async copy: legal inside hipGraph capture
-            # (grad pointers can move when backward allocates from a capture
-            # pool); the pinned buffer must stay alive as long as any graph
-            # that captured the copy, hence the keepalive list.
             host = torch.tensor(rows, dtype=torch.int64).pin_memory()
             dev = torch.empty_like(host, device=params[0].device)
             dev.copy_(host, non_blocking=True)
@@ -156,11 +169,33 @@ class FusedSGD(torch.optim.Optimizer):
             entry = (fp, dev, len(rows))
             self._mt_cache[gi] = entry
         if self._hyper_dev is not None:
-            ext.sgd_step_multi_dev(entry[1], entry[2], self._hyper_dev)
+            ext.sgd_step_multi_dev(entry[1], entry[2], self._hyper_dev,
+                                   self._persistent)
         else:
-            ext.sgd_step_multi(entry[1], entry[2], lr, momentum, wd)
+            ext.sgd_step_multi(entry[1], entry[2], lr, momentum, wd,
+                               self._persistent)
         for p, shadow in post:
             if shadow is not None:
+                if getattr(shadow, "_al_wt", None) is not None:
+                    shadow._al_wt = None
+                p._al_cast = (p._version, shadow)
+            else:
+                _invalidate(p)
+            bump_tick(p)
+
+    def _per_tensor_fallback(self, ext, params, lr, momentum, wd):
+        for p in params:
+            buf = self.state[p].get("momentum_buffer")
+            cache = getattr(p, "_al_cast", None)
+            shadow = (cache[1] if cache is not None
+                      and cache[1].numel() == p.numel()
+                      and cache[1].dtype == torch.bfloat16
+                      else p.new_empty(0, dtype=torch.bfloat16))
+            ext.sgd_step(p, p.grad, buf if buf is not None else p.new_empty(0),
+                         lr, momentum, wd, shadow)
+            if self._persistent:
+                p.grad.zero_()
+            if shadow.numel():
                 if getattr(shadow, "_al_wt", None) is not None:
                     shadow._al_wt = None
                 p._al_cast = (p._version, shadow)

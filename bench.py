@@ -104,25 +104,18 @@ def main():
     net.train()
     use_graph = (args.graph == "on" or (args.graph == "auto" and world == 1
                                         and dev_kind == "cuda"))
-    graphs = []
     if use_graph:
-        # warm up eagerly, then capture one hipGraph per data buffer; replay
-        # covers fwd + CE + bwd + fused SGD with zero per-kernel launch gaps
-        try:
-            for i in range(max(3, args.warmup)):
-                step(i)
-            sync()
-            for i in range(n_buf):
-                g = torch.cuda.CUDAGraph()
-                with torch.cuda.graph(g):
-                    step(i)
-                graphs.append(g)
+        # shared capture path with the real trainer (ops/graph.py):
+        # persistent grads + device-hyper SGD keep the captured step
+        # allocation-free; fwd + CE + bwd + fused SGD replay as one graph
+        from active_learning_amd.ops.graph import GraphedTrainStep
+        # capture triggers on the LAST untimed warmup call so its one-time
+        # cost never lands in the timed region
+        gs = GraphedTrainStep(net, opt, cross_entropy, device,
+                              warmup=max(2, args.warmup - 1))
 
-            def step(i):  # noqa: F811 — replay path
-                graphs[i % n_buf].replay()
-        except Exception as e:  # never lose the bench to a capture failure
-            print(f"# hipGraph capture unavailable ({e!r}); running eager", flush=True)
-            graphs = []
+        def step(i):  # noqa: F811 — graphed path
+            return gs.step(xs[i % n_buf], ys[i % n_buf])
     for i in range(args.warmup):
         step(i)
     sync()
