@@ -103,6 +103,32 @@ def test_conv_bwd_weight(case):
     assert relerr(dw, ref) < tol, f"conv bwd_weight {case}: relerr {relerr(dw, ref)}"
 
 
+WGRAD_V4_CASES = [
+    # big-L shapes that route to the 8-phase 256² wgrad (KT >= 12)
+    (48, 28, 28, 256, 256, 3, 1, 1),    # Nw=2304: 9 rsc tiles, KT~21
+    (32, 14, 14, 512, 512, 3, 1, 1),    # K=512 x Nw=4608: tails both dims
+    (256, 14, 14, 1024, 256, 1, 1, 0),  # 1x1 l3.conv1 at bench scale (KT=13)
+]
+
+
+@pytest.mark.parametrize("case", WGRAD_V4_CASES)
+def test_conv_bwd_weight_v4(case):
+    """8-phase 256² wgrad (wgrad4_kernel) vs torch fp32 on-device."""
+    n, h, w, c, k, r, stride, pad = case
+    p = (h + 2 * pad - r) // stride + 1
+    torch.manual_seed(5)
+    x = torch.randn(n, h, w, c, device="cuda")
+    dy = torch.randn(n, p, p, k, device="cuda")
+    ref = torch.nn.grad.conv2d_weight(
+        x.permute(0, 3, 1, 2).contiguous(), [k, c, r, r],
+        dy.permute(0, 3, 1, 2).contiguous(), stride=stride,
+        padding=pad).permute(0, 2, 3, 1)
+    dw = EXT.conv2d_bwd_weight(dy.to(torch.bfloat16), x.to(torch.bfloat16),
+                               r, r, stride, pad)
+    err = relerr(dw, ref)
+    assert err < 0.03, f"wgrad v4 {case}: relerr {err}"
+
+
 # --------------------------------------------------------------------------- #
 # bn / pool through the autograd Functions (GPU path vs CPU path)
 # --------------------------------------------------------------------------- #
