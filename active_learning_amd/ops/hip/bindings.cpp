@@ -8,6 +8,7 @@
 #include <torch/extension.h>
 #include <c10/hip/HIPStream.h>
 
+#include <algorithm>
 #include <vector>
 
 using torch::Tensor;
@@ -64,6 +65,10 @@ void al_sgd_step_multi(const void* table, int nchunks, float lr, float momentum,
                        float wd, int zero_grad, void* stream);
 void al_sgd_step_multi_dev(const void* table, int nchunks, const float* hyper,
                            int zero_grad, void* stream);
+int al_kcenter_greedy(const float* dist, float* min_dist, unsigned char* labeled,
+                      long* sel, const float* randu, float* partial, void* st,
+                      long n, int iters, int j_init, int randomize, int nblocks,
+                      void* stream);
 }
 
 namespace {
@@ -452,6 +457,31 @@ void adam_step(Tensor& p, const Tensor& g, Tensor& m, Tensor& v, double lr, doub
                cur_stream());
 }
 
+int64_t kcenter_greedy_dev(const Tensor& dist, Tensor& min_dist, Tensor& labeled,
+                           Tensor& sel, const Tensor& randu, int64_t j_init,
+                           bool randomize) {
+  TORCH_CHECK(dist.is_cuda() && dist.scalar_type() == torch::kFloat32 &&
+              dist.is_contiguous());
+  const long n = dist.size(0);
+  TORCH_CHECK(dist.size(1) == n && min_dist.numel() == n);
+  TORCH_CHECK(labeled.scalar_type() == torch::kUInt8 && labeled.numel() == n);
+  TORCH_CHECK(sel.scalar_type() == torch::kInt64);
+  const int iters = (int)sel.numel();
+  if (iters == 0) return 0;
+  if (randomize) TORCH_CHECK(randu.numel() >= iters);
+  int nblocks = (int)std::min<long>(512, (n + 1023) / 1024);
+  if (nblocks < 1) nblocks = 1;
+  auto fopts = dist.options();
+  Tensor partial = torch::zeros({2 * nblocks}, fopts);
+  Tensor st = torch::zeros({8}, fopts.dtype(torch::kInt32));
+  return al_kcenter_greedy(
+      dist.data_ptr<float>(), min_dist.data_ptr<float>(),
+      labeled.data_ptr<unsigned char>(), sel.data_ptr<int64_t>(),
+      randomize ? randu.data_ptr<float>() : nullptr, partial.data_ptr<float>(),
+      st.data_ptr(), n, iters, (int)j_init, randomize ? 1 : 0, nblocks,
+      cur_stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -477,5 +507,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sgd_step", &sgd_step);
   m.def("sgd_step_multi", &sgd_step_multi);
   m.def("sgd_step_multi_dev", &sgd_step_multi_dev);
+  m.def("kcenter_greedy_dev", &kcenter_greedy_dev);
   m.def("adam_step", &adam_step);
 }

@@ -815,6 +815,237 @@ __global__ void wgrad4_kernel(const bf16* __restrict__ dy,
         }
 }
 
+// ---------------------------------------------------------------------------
+// wgrad v5: 2-phase variant of the 256² ring. v4's 4-phase schedule leaves
+// only 8 MFMAs per wave per barrier interval (vs v2's 32 per drain) and
+// measured BELOW v2 on most shapes. v5 processes a full A-half row of
+// quadrants per phase — (qm,0) then (qm,1) with the B h1 fragments read
+// mid-phase — giving 32 MFMAs per phase at the same register budget.
+// B half-slots are consumed every phase, so their refills trail the phase's
+// closing barrier; group-top wait is vmcnt(2) (only the trailing B h0
+// prefetch may stay in flight).
+// ---------------------------------------------------------------------------
+
+__launch_bounds__(512, 1)
+__global__ void wgrad5_kernel(const bf16* __restrict__ dy,
+                              const bf16* __restrict__ x,
+                              float* __restrict__ dw,
+                              const bf16* __restrict__ zero,
+                              WgradShape sh, int grid_k, long l_per_z) {
+  constexpr int BL = 64;
+  constexpr int HALF = 128 * BL;
+  constexpr int NL = 2;
+  const int bk = blockIdx.x % grid_k;
+  const int bn = blockIdx.x / grid_k;
+  const int k0 = bk * 256;
+  const int n0 = bn * 256;
+  const long lz0 = (long)blockIdx.y * l_per_z;
+  const long lz1 = min(sh.L, lz0 + l_per_z);
+  if (lz0 >= lz1) return;
+  const int KT = (int)((lz1 - lz0 + BL - 1) / BL);
+
+  extern __shared__ __attribute__((aligned(16))) char smem5[];
+  bf16* As = (bf16*)smem5;
+  bf16* Bs = As + 4 * HALF;
+
+  const int tid = threadIdx.x;
+  const int wid = tid >> 6, lane = tid & 63;
+  const int wrq = wid >> 2, wcq = wid & 3;
+  const int l15 = lane & 15, l4 = lane >> 4;
+
+  long a_off[2][NL];
+  bool a_chok[2][NL];
+  int a_l[2][NL];
+#pragma unroll
+  for (int h = 0; h < 2; ++h)
+#pragma unroll
+    for (int li = 0; li < NL; ++li) {
+      const int ci = li * 512 + tid;
+      const int g = ci >> 7, l = (ci & 127) >> 1, h2 = ci & 1;
+      const int ch = k0 + h * 128 + g * 16 + h2 * 8;
+      a_l[h][li] = l;
+      a_chok[h][li] = ch + 8 <= sh.K;
+      a_off[h][li] = (lz0 + l) * (long)sh.K + ch;
+    }
+  int b_n[2][NL], b_p[2][NL], b_q[2][NL], b_ho[2][NL], b_wo[2][NL], b_l[2][NL];
+  long b_coff[2][NL];
+  bool b_chok[2][NL];
+  const int dn = (int)(BL / ((long)sh.P * sh.Q));
+  const int rem = (int)(BL % ((long)sh.P * sh.Q));
+  const int dp = rem / sh.Q, dq = rem % sh.Q;
+#pragma unroll
+  for (int h = 0; h < 2; ++h)
+#pragma unroll
+    for (int li = 0; li < NL; ++li) {
+      const int ci = li * 512 + tid;
+      const int g = ci >> 7, l = (ci & 127) >> 1, h2 = ci & 1;
+      const int nw = n0 + h * 128 + g * 16 + h2 * 8;
+      b_l[h][li] = l;
+      b_chok[h][li] = nw + 8 <= sh.Nw;
+      const int c = nw % sh.C;
+      const int rs = nw / sh.C;
+      b_ho[h][li] = (rs / sh.S) - sh.pad;
+      b_wo[h][li] = (rs % sh.S) - sh.pad;
+      b_coff[h][li] = c;
+      const long m = lz0 + l;
+      b_q[h][li] = (int)(m % sh.Q);
+      long t = m / sh.Q;
+      b_p[h][li] = (int)(t % sh.P);
+      b_n[h][li] = (int)(t / sh.P);
+    }
+  auto advance_a = [&](int h) {
+#pragma unroll
+    for (int li = 0; li < NL; ++li) a_off[h][li] += (long)BL * sh.K;
+  };
+  auto advance_b = [&](int h) {
+#pragma unroll
+    for (int li = 0; li < NL; ++li) {
+      b_q[h][li] += dq;
+      if (b_q[h][li] >= sh.Q) { b_q[h][li] -= sh.Q; ++b_p[h][li]; }
+      b_p[h][li] += dp;
+      if (b_p[h][li] >= sh.P) { b_p[h][li] -= sh.P; ++b_n[h][li]; }
+      b_n[h][li] += dn;
+    }
+  };
+  auto stage_a = [&](int h, int kt, bf16* slot) {
+    const long l0 = lz0 + (long)kt * BL;
+#pragma unroll
+    for (int li = 0; li < NL; ++li) {
+      const bf16* src = (kt < KT && a_chok[h][li] && l0 + a_l[h][li] < lz1)
+                            ? dy + a_off[h][li] : zero;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)src,
+          (__attribute__((address_space(3))) void*)(slot + (li * 512 + wid * 64) * 8),
+          16, 0, 0);
+    }
+  };
+  auto stage_b = [&](int h, int kt, bf16* slot) {
+    const long l0 = lz0 + (long)kt * BL;
+#pragma unroll
+    for (int li = 0; li < NL; ++li) {
+      const int hh = b_p[h][li] * sh.stride + b_ho[h][li];
+      const int ww = b_q[h][li] * sh.stride + b_wo[h][li];
+      const bool ok = kt < KT && b_chok[h][li] && (l0 + b_l[h][li] < lz1) &&
+                      (unsigned)hh < (unsigned)sh.H && (unsigned)ww < (unsigned)sh.W;
+      const bf16* src = ok
+          ? x + (((long)b_n[h][li] * sh.H + hh) * sh.W + ww) * sh.C + b_coff[h][li]
+          : zero;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)src,
+          (__attribute__((address_space(3))) void*)(slot + (li * 512 + wid * 64) * 8),
+          16, 0, 0);
+    }
+  };
+
+  typedef __attribute__((address_space(3))) s16x4* lds5_v4p;
+  auto trfrag = [&](const bf16* img, int group, int mc) -> bf16x8 {
+    const bf16* p = img + group * 1024 + mc * 32 * 16 + l4 * 8 * 16 + l15 * 4;
+    union { s16x4 h[2]; bf16x8 v; } u;
+    u.h[0] = __builtin_amdgcn_ds_read_tr16_b64_v4i16((lds5_v4p)p);
+    u.h[1] = __builtin_amdgcn_ds_read_tr16_b64_v4i16((lds5_v4p)(p + 64));
+    return u.v;
+  };
+
+  f32x4 acc[2][2][4][2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j)
+#pragma unroll
+      for (int f = 0; f < 4; ++f)
+#pragma unroll
+        for (int n = 0; n < 2; ++n) acc[i][j][f][n] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  // prologue (12 loads/thread): tile0 all four halves, then tile1's A h0 and
+  // B h0 — matches the steady-state order so vmcnt(2) at the group top
+  // drains everything except the trailing B h0 prefetch.
+  stage_a(0, 0, As);
+  stage_a(1, 0, As + HALF);
+  stage_b(0, 0, Bs);
+  stage_b(1, 0, Bs + HALF);
+  advance_a(0); advance_b(0);
+  stage_a(0, 1, As + 2 * HALF);
+  stage_b(0, 1, Bs + 2 * HALF);
+
+  for (int g = 0; g < KT; ++g) {
+    const int buf = g & 1;
+    asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+#pragma unroll
+    for (int qm = 0; qm < 2; ++qm) {
+      const bf16* ah = As + (buf * 2 + qm) * HALF;
+      const bf16* b0 = Bs + (buf * 2) * HALF;
+      const bf16* b1 = Bs + (buf * 2 + 1) * HALF;
+      bf16x8 af[4][2], bf[2][2];
+#pragma unroll
+      for (int mc = 0; mc < 2; ++mc) {
+#pragma unroll
+        for (int fg = 0; fg < 4; ++fg) af[fg][mc] = trfrag(ah, wrq * 4 + fg, mc);
+#pragma unroll
+        for (int ng = 0; ng < 2; ++ng) bf[ng][mc] = trfrag(b0, wcq * 2 + ng, mc);
+      }
+      if (qm == 0) { advance_a(1); stage_a(1, g + 1, As + ((buf ^ 1) * 2 + 1) * HALF); }
+      else         { advance_a(0); stage_a(0, g + 2, As + (buf * 2) * HALF); }
+      __builtin_amdgcn_s_barrier();
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int mc = 0; mc < 2; ++mc)
+#pragma unroll
+        for (int fg = 0; fg < 4; ++fg)
+#pragma unroll
+          for (int ng = 0; ng < 2; ++ng)
+            acc[qm][0][fg][ng] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                af[fg][mc], bf[ng][mc], acc[qm][0][fg][ng], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+      // second quadrant: B h1 fragments mid-phase (reads target the CURRENT
+      // buf's B h1 slot; the in-flight refill writes a different slot)
+#pragma unroll
+      for (int mc = 0; mc < 2; ++mc)
+#pragma unroll
+        for (int ng = 0; ng < 2; ++ng) bf[ng][mc] = trfrag(b1, wcq * 2 + ng, mc);
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int mc = 0; mc < 2; ++mc)
+#pragma unroll
+        for (int fg = 0; fg < 4; ++fg)
+#pragma unroll
+          for (int ng = 0; ng < 2; ++ng)
+            acc[qm][1][fg][ng] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                af[fg][mc], bf[ng][mc], acc[qm][1][fg][ng], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+      __builtin_amdgcn_s_barrier();
+    }
+    // trailing B refills: both B halves were last read inside this group,
+    // and the closing barrier above orders every wave's reads before these
+    // glds writes. B h1 <- g+1 first, then B h0 <- g+2 (the only loads
+    // vmcnt(2) leaves in flight at the next group top).
+    advance_b(1);
+    stage_b(1, g + 1, Bs + ((buf ^ 1) * 2 + 1) * HALF);
+    advance_b(0);
+    stage_b(0, g + 2, Bs + (buf * 2) * HALF);
+  }
+
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+#pragma unroll
+  for (int qm = 0; qm < 2; ++qm)
+#pragma unroll
+    for (int qn = 0; qn < 2; ++qn)
+#pragma unroll
+      for (int fg = 0; fg < 4; ++fg)
+#pragma unroll
+        for (int ng = 0; ng < 2; ++ng) {
+          const int col = n0 + qn * 128 + wcq * 32 + ng * 16 + l15;
+          if (col >= sh.Nw) continue;
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            const int row = k0 + qm * 128 + wrq * 64 + fg * 16 + l4 * 4 + r;
+            if (row < sh.K)
+              atomicAdd(&dw[(long)row * sh.Nw + col], acc[qm][qn][fg][ng][r]);
+          }
+        }
+}
+
 // direct fallback: one dW element per thread, strided over L (safety net;
 // normal stems go through the packed-im2col MFMA path)
 __global__ void wgrad_direct_kernel(const bf16* __restrict__ dy,
@@ -849,13 +1080,16 @@ extern "C" void al_conv2d_wgrad(const void* dy, const void* x, float* dw,
   sh.L = (long)N * P * Q;
   sh.Nw = R * S * C;
   if (C % 8 == 0 && K % 8 == 0) {
-    // v4: 8-phase 256² schedule for the wide layers. Needs a K-chain long
-    // enough to amortize the 3-tile-deep prologue (KT >= threshold) and
-    // enough (k, rsc) coverage that the 256² tile isn't mostly zero-page.
-    static int v4 = -1, v4_minkt = 12;
-    if (v4 < 0) {
+    // 256² ring schedules for the wide layers (AL_WGRAD_V4: 0=off, 1=v4
+    // 4-phase, 2=v5 2-phase). Need a K-chain long enough to amortize the
+    // 3-tile-deep prologue (KT >= threshold) and enough (k, rsc) coverage
+    // that the 256² tile isn't mostly zero-page. Measured (tools/ab_wgrad):
+    // v4 beats v2 only on l3.conv2-class shapes; v5 doubles the MFMA per
+    // barrier interval.
+    static int v4 = -2, v4_minkt = 12;
+    if (v4 == -2) {
       const char* e = getenv("AL_WGRAD_V4");
-      v4 = (e && e[0] == '0') ? 0 : 1;
+      v4 = e ? atoi(e) : 2;
       const char* t = getenv("AL_WGRAD_V4_MINKT");
       if (t) v4_minkt = atoi(t);
     }
@@ -874,12 +1108,20 @@ extern "C" void al_conv2d_wgrad(const void* dy, const void* x, float* dw,
           (void)hipFuncSetAttribute((const void*)wgrad4_kernel,
                                     hipFuncAttributeMaxDynamicSharedMemorySize,
                                     163840);
+          (void)hipFuncSetAttribute((const void*)wgrad5_kernel,
+                                    hipFuncAttributeMaxDynamicSharedMemorySize,
+                                    163840);
           attr4 = true;
         }
         const size_t lds4 = 8 * (size_t)(128 * 64) * sizeof(bf16);  // 128 KiB
-        hipLaunchKernelGGL(wgrad4_kernel, dim3(tiles4, z4), dim3(512), lds4,
-                           stream, (const bf16*)dy, (const bf16*)x, dw,
-                           (const bf16*)zero_page, sh, grid_k4, lpz4);
+        if (v4 >= 2)
+          hipLaunchKernelGGL(wgrad5_kernel, dim3(tiles4, z4), dim3(512), lds4,
+                             stream, (const bf16*)dy, (const bf16*)x, dw,
+                             (const bf16*)zero_page, sh, grid_k4, lpz4);
+        else
+          hipLaunchKernelGGL(wgrad4_kernel, dim3(tiles4, z4), dim3(512), lds4,
+                             stream, (const bf16*)dy, (const bf16*)x, dw,
+                             (const bf16*)zero_page, sh, grid_k4, lpz4);
         return;
       }
     }

@@ -22,6 +22,8 @@ Reference counterparts:
     argmax/sampling all stay on device (HBM-resident per BASELINE.json).
 """
 
+import os
+
 import torch
 import torch.nn.functional as F
 
@@ -180,6 +182,29 @@ def kcenter_greedy(dist: torch.Tensor, labeled: torch.Tensor, budget: int,
         labeled[j] = True
         min_dist = dist[j].clone()  # symmetric: row == column
         start = 1
+
+    if (dist.is_cuda and budget > start
+            and os.environ.get("AL_KCENTER_DEV", "1") == "1"):
+        # persistent cooperative kernel: the ENTIRE greedy loop is one launch
+        # (fused min-update + argmax / inverse-CDF sample per iteration,
+        # grid.sync between) instead of 3+ launches per iteration
+        from .extension import require_extension
+        ext = require_extension()
+        iters = budget - start
+        lab_u8 = labeled.to(torch.uint8)
+        randu = (torch.rand(iters, device=dev) if randomize
+                 else torch.empty(0, device=dev))
+        md = min_dist.contiguous()
+        sel_tail = sel_buf[start:].contiguous()
+        # j_init < 0: min_dist is already up to date (labeled-init path);
+        # j_init >= 0 would re-apply a row — the caller already applied it,
+        # so always pass -1 and keep min_dist authoritative
+        rc = ext.kcenter_greedy_dev(dist, md, lab_u8, sel_tail, randu, -1,
+                                    bool(randomize))
+        if rc == 0:
+            sel_buf[start:] = sel_tail
+            return sel_buf.cpu().tolist()
+        # cooperative launch unsupported: fall through to the torch loop
 
     for it in range(start, budget):
         if randomize:

@@ -129,6 +129,32 @@ def test_conv_bwd_weight_v4(case):
     assert err < 0.03, f"wgrad v4 {case}: relerr {err}"
 
 
+def test_kcenter_persistent_kernel():
+    """The cooperative persistent k-center kernel must select exactly the
+    same points as the torch reference loop (deterministic mode), and produce
+    valid unique unlabeled selections in k-means++ mode."""
+    import os
+    from active_learning_amd.ops.scoring import kcenter_greedy, pairwise_sqdist
+    torch.manual_seed(9)
+    n, m, budget = 3000, 64, 200
+    feats = torch.randn(n, m, device="cuda")
+    dist = pairwise_sqdist(feats)
+    labeled = torch.zeros(n, dtype=torch.bool, device="cuda")
+    labeled[:37] = True
+
+    os.environ["AL_KCENTER_DEV"] = "0"
+    ref = kcenter_greedy(dist, labeled, budget, randomize=False)
+    os.environ["AL_KCENTER_DEV"] = "1"
+    got = kcenter_greedy(dist, labeled, budget, randomize=False)
+    os.environ.pop("AL_KCENTER_DEV")
+    assert got == ref, "persistent kernel diverged from torch greedy loop"
+
+    sel = kcenter_greedy(dist, labeled, budget, randomize=True)
+    assert len(sel) == budget
+    assert len(set(sel)) == budget, "duplicate selection in k-means++ mode"
+    assert not labeled[torch.tensor(sel)].any().item(), "selected labeled point"
+
+
 # --------------------------------------------------------------------------- #
 # bn / pool through the autograd Functions (GPU path vs CPU path)
 # --------------------------------------------------------------------------- #
