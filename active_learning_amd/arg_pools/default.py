@@ -32,6 +32,21 @@ args_pool = {
         "lr_scheduler_args": {"step_size": 60, "gamma": 0.1},
         "rd0_pretrained_ckpt_path": None,
     },
+    # Not in the reference's default pool: the reference ships the
+    # ImageNet-LT loader + split files but no arg-pool entry, so the dataset
+    # cannot actually run there (PARITY.md). Entry mirrors "imagenet" with
+    # inverse-frequency weighted CE like "imbalanced_cifar10".
+    "imbalanced_imagenet": {
+        "eval_split": 0.01,
+        "loader_tr_args": {"batch_size": 128, "num_workers": 12, "prefetch_factor": 2},
+        "loader_te_args": {"batch_size": 128, "num_workers": 12, "prefetch_factor": 2},
+        "optimizer": "SGD",
+        "optimizer_args": {"lr": 0.1, "weight_decay": 1e-4, "momentum": 0.9},
+        "lr_scheduler": "StepLR",
+        "lr_scheduler_args": {"step_size": 60, "gamma": 0.1},
+        "rd0_pretrained_ckpt_path": None,
+        "imbalanced_training": True,
+    },
     # Synthetic datasets (not in the reference): used by bench.py and tests where
     # no dataset files exist on disk (this environment has no network).
     "synthetic_cifar10": {

@@ -91,6 +91,39 @@ def test_imagenet_lt_lists(tmp_path):
     assert y == 0 and idx == 2
 
 
+def test_imagenet_lt_fixture_e2e(tmp_path):
+    """Committed ImageNet-LT fixture (tests/fixtures/imagenet_lt — the
+    reference's exact `path label` line format,
+    src/data_utils/ImageNet_LT/ImageNet_LT_test.txt:1) through a full
+    imbalanced-imagenet debug round."""
+    import os
+    import shutil
+    from PIL import Image
+    from active_learning_amd.cli import get_args
+    from active_learning_amd.main_al import main
+
+    fix = os.path.join(os.path.dirname(__file__), "fixtures", "imagenet_lt")
+    for name in ("ImageNet_LT_train.txt", "ImageNet_LT_test.txt"):
+        shutil.copy(os.path.join(fix, name), tmp_path / name)
+        with open(os.path.join(fix, name)) as fh:
+            for line in fh:
+                rel, lbl = line.split()
+                assert rel.split("/")[0] in ("train", "val")
+                p = tmp_path / rel
+                p.parent.mkdir(parents=True, exist_ok=True)
+                Image.new("RGB", (24, 24), (int(lbl) * 50, 10, 10)).save(p)
+
+    args = get_args([
+        "--dataset", "imbalanced_imagenet", "--dataset_dir", str(tmp_path),
+        "--strategy", "RandomSampler", "--rounds", "1", "--round_budget", "5",
+        "--n_epoch", "1", "--early_stop_patience", "1", "--debug_mode",
+        "--ckpt_path", str(tmp_path / "ck"), "--log_dir", str(tmp_path / "lg"),
+        "--model", "SSLResNet18"])
+    s = main(args)
+    assert s.idxs_lb.sum() == 5
+    assert len(s.train_set) == 50  # fixture train list length (debug cap == len)
+
+
 def test_imbalanced_e2e_round(tmp_path, cifar_dir):
     """Imbalanced dataset + inverse-frequency weighted CE through a debug
     round (reference: imbalanced_training flag, strategy.py:352-356)."""
