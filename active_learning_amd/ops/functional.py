@@ -28,6 +28,14 @@ from torch.autograd import Function
 from .extension import require_extension
 
 
+def _native_ok(x):
+    """The hand-written gfx950 kernels take bf16 NHWC; fp32-on-GPU
+    (--compute_dtype fp32) runs the same ops through the torch math below —
+    executed on-device via MIOpen/rocBLAS — as a numerics escape hatch
+    (reference trains fp32 throughout)."""
+    return x.is_cuda and x.dtype == torch.bfloat16
+
+
 def _cpu_conv_args(x_nhwc, w_krsc):
     return x_nhwc.permute(0, 3, 1, 2), w_krsc.permute(0, 3, 1, 2)
 
@@ -199,7 +207,7 @@ class Conv2dNHWC(Function):
         ctx.w_c = w_c
         ctx.stride, ctx.padding = stride, padding
         ctx.weight_dtype = weight.dtype
-        if x.is_cuda:
+        if _native_ok(x):
             ext = require_extension()
             _grad_arena.mark_step()
             K, R, S, C = w_c.shape
@@ -219,7 +227,7 @@ class Conv2dNHWC(Function):
         w_c = ctx.w_c
         dy = dy.contiguous()
         dx = dw = None
-        if x.is_cuda:
+        if _native_ok(x):
             ext = require_extension()
             K, R, S, C = w_c.shape
             if ctx.needs_input_grad[0]:
@@ -278,7 +286,7 @@ class Conv2dNHWCStats(Function):
         ctx.w_c = w_c
         ctx.stride, ctx.padding = stride, padding
         ctx.weight_dtype = weight.dtype
-        if x.is_cuda:
+        if _native_ok(x):
             ext = require_extension()
             _grad_arena.mark_step()
             K, R, S, C = w_c.shape
@@ -348,7 +356,7 @@ class BatchNormAct(Function):
                             running_mean.mul_(1 - momentum).add_(mean, alpha=momentum)
                             running_var.mul_(1 - momentum).add_(unbiased,
                                                                 alpha=momentum)
-                elif x.is_cuda and running_mean is not None:
+                elif _native_ok(x) and running_mean is not None:
                     from .extension import require_extension as _re
                     with torch.no_grad():
                         mean, invstd = _re().bn_finalize(
@@ -368,7 +376,7 @@ class BatchNormAct(Function):
                             running_mean.mul_(1 - momentum).add_(mean, alpha=momentum)
                             running_var.mul_(1 - momentum).add_(unbiased,
                                                                 alpha=momentum)
-            elif x.is_cuda and not sync:
+            elif _native_ok(x) and not sync:
                 # fused path: partial sums -> mean/invstd + running update in
                 # two kernels, no small ATen ops, no host sync
                 ext = require_extension()
@@ -380,7 +388,7 @@ class BatchNormAct(Function):
                     bump_tick(running_var)
                 n = float(n_local)
             else:
-                if x.is_cuda:
+                if _native_ok(x):
                     ext = require_extension()
                     s, ss = ext.bn_stats(x)  # fp32 per-channel sums
                 else:
@@ -408,7 +416,7 @@ class BatchNormAct(Function):
             invstd = (running_var.to(torch.float32) + eps).rsqrt()
             n = float(n_local)
 
-        if x.is_cuda:
+        if _native_ok(x):
             ext = require_extension()
             # want_mask: the backward re-streams a 1-bit relu mask instead of
             # y. (grad mode is always off inside Function.forward, so gate on
@@ -429,7 +437,7 @@ class BatchNormAct(Function):
             relu_mask = None
         # GPU path saves the bit mask (backward never re-reads y); CPU path
         # keeps y for its fp32 reference masks
-        ctx.save_for_backward(x, relu_mask if x.is_cuda else y, weight, mean,
+        ctx.save_for_backward(x, relu_mask if _native_ok(x) else y, weight, mean,
                               invstd)
         ctx.use_batch_stats = use_batch_stats
         ctx.relu = relu
@@ -441,9 +449,9 @@ class BatchNormAct(Function):
     @staticmethod
     def backward(ctx, dy):
         x, mask_or_y, weight, mean, invstd = ctx.saved_tensors
-        y = None if x.is_cuda else mask_or_y
+        y = None if _native_ok(x) else mask_or_y
         dy = dy.contiguous()
-        if x.is_cuda:
+        if _native_ok(x):
             ext = require_extension()
             # reduce pass: per-channel sums of dy~ and dy~*xhat (dy~ = mask * dy)
             sum_dy, sum_dy_xhat = ext.bn_bwd_reduce(dy, x, mask_or_y, mean, invstd,
@@ -472,7 +480,7 @@ class BatchNormAct(Function):
         n_global = ctx.n
 
         dres = None
-        if x.is_cuda:
+        if _native_ok(x):
             ext = require_extension()
             dx, dres_t = ext.bn_bwd(dy, x, mask_or_y, mean, invstd, weight, sum_dy,
                                     sum_dy_xhat, n_global, ctx.use_batch_stats,
@@ -506,7 +514,8 @@ def batch_norm_act(x, weight, bias, running_mean, running_var, use_batch_stats,
 class MaxPool2dNHWC(Function):
     @staticmethod
     def forward(ctx, x, kernel, stride, padding):
-        if x.is_cuda:
+        ctx.native = _native_ok(x)
+        if ctx.native:
             ext = require_extension()
             y, idx = ext.maxpool2d_fwd(x, kernel, stride, padding)
         else:
@@ -524,7 +533,7 @@ class MaxPool2dNHWC(Function):
         (idx,) = ctx.saved_tensors
         kernel, stride, padding = ctx.params
         dy = dy.contiguous()
-        if dy.is_cuda:
+        if ctx.native:
             ext = require_extension()
             dx = ext.maxpool2d_bwd(dy, idx, ctx.x_shape[1], ctx.x_shape[2],
                                    kernel, stride, padding)
@@ -551,7 +560,7 @@ class GlobalAvgPoolNHWC(Function):
     def forward(ctx, x):
         ctx.x_shape = x.shape
         ctx.x_dtype = x.dtype
-        if x.is_cuda:
+        if _native_ok(x):
             ext = require_extension()
             return ext.global_avg_pool(x)
         return x.float().mean(dim=(1, 2)).to(x.dtype)

@@ -93,16 +93,23 @@ class GraphedTrainStep:
         if self._persistent:
             self._dealias_grads()
         self.opt.step()
-        return loss
+        # return DETACHED: a caller-held loss would keep this step's autograd
+        # graph (incl. default-stream AccumulateGrad nodes) alive across the
+        # upcoming capture, which aborts hipGraph capture with a stream
+        # mismatch (observed: core dump in torch 2.10's input_buffer.cpp)
+        return loss.detach()
 
     def _capture(self):
+        import gc
+        gc.collect()  # drop any lingering warmup autograd graphs for certain
         g = torch.cuda.CUDAGraph()
         with torch.cuda.graph(g):
             out = self.net(self.x_static)
             loss = self.crit(out, self.y_static)
             loss.backward()
             self.opt.step()
-            self.loss_static = loss
+            # detached view shares storage: each replay refreshes the value
+            self.loss_static = loss.detach()
         self._graph = g
 
     def _post_replay(self):

@@ -1086,14 +1086,20 @@ extern "C" void al_conv2d_wgrad(const void* dy, const void* x, float* dw,
     // that the 256² tile isn't mostly zero-page. Measured (tools/ab_wgrad):
     // v4 beats v2 only on l3.conv2-class shapes; v5 doubles the MFMA per
     // barrier interval.
+    // Measured (tools/ab_wgrad, B=256 shapes): the 256² ring beats the
+    // 2-barrier v2 only on the 3x3 K=256 class (l3.conv2: v5 0.137 ms vs v2
+    // 0.154); everywhere else v2's 2-blocks/CU overlap wins (64 KiB LDS vs
+    // the ring's 128 KiB -> 1 block/CU), the same occupancy regime that
+    // killed the v3 ring. Default routing: v5 on that class only.
     static int v4 = -2, v4_minkt = 12;
     if (v4 == -2) {
       const char* e = getenv("AL_WGRAD_V4");
-      v4 = e ? atoi(e) : 2;
+      v4 = e ? atoi(e) : 3;   // 0 off, 1 v4 all, 2 v5 all, 3 v5 where it wins
       const char* t = getenv("AL_WGRAD_V4_MINKT");
       if (t) v4_minkt = atoi(t);
     }
-    if (v4 && K >= 192 && sh.Nw >= 192) {
+    const bool v5_wins = (R * S > 1) && K >= 192 && K < 384 && sh.Nw >= 2048;
+    if ((v4 == 1 || v4 == 2 || (v4 == 3 && v5_wins)) && K >= 192 && sh.Nw >= 192) {
       const int grid_k4 = (K + 255) / 256;
       const int grid_n4 = (sh.Nw + 255) / 256;
       const int tiles4 = grid_k4 * grid_n4;
@@ -1114,7 +1120,7 @@ extern "C" void al_conv2d_wgrad(const void* dy, const void* x, float* dw,
           attr4 = true;
         }
         const size_t lds4 = 8 * (size_t)(128 * 64) * sizeof(bf16);  // 128 KiB
-        if (v4 >= 2)
+        if (v4 >= 2 || v4 == 3)
           hipLaunchKernelGGL(wgrad5_kernel, dim3(tiles4, z4), dim3(512), lds4,
                              stream, (const bf16*)dy, (const bf16*)x, dw,
                              (const bf16*)zero_page, sh, grid_k4, lpz4);
