@@ -43,9 +43,9 @@ def main(args):
     net = get_networks(args.dataset, args.model)
     net.freeze_feature = args.freeze_feature
     if getattr(args, "compute_dtype", "bf16") == "fp32":
-        # fp32 device compute is CPU-only in round 1 (GPU kernels are bf16
-        # with fp32 accumulate/master weights; see PARITY.md) — disable the
-        # bf16 cast so CPU runs full fp32.
+        # fp32 numerics-parity path: disables the bf16 cast; on GPU the ops
+        # dispatch to on-device torch math (MIOpen/rocBLAS) instead of the
+        # bf16 HIP kernels (ops/functional.py::_native_ok; PARITY.md).
         net.encoder.compute_dtype = None
 
     eval_idxs = generate_eval_idxs(train_set, train_args["eval_split"], random_seed=99)

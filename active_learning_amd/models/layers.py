@@ -119,7 +119,7 @@ class _TransposedConv2d(torch.autograd.Function):
         k = w_c.shape[1]
         h_out = (x.shape[1] - 1) * stride - 2 * padding + k
         w_out = (x.shape[2] - 1) * stride - 2 * padding + k
-        if x.is_cuda:
+        if x.is_cuda and x.dtype == torch.bfloat16:
             from ..ops.extension import require_extension
             # convT fwd == conv bwd-data; its "wt" is (Cout,R,S,Cin)
             return require_extension().conv2d_bwd_data(x, wt_cached(w_c), stride,
@@ -135,7 +135,7 @@ class _TransposedConv2d(torch.autograd.Function):
         w_c = ctx.w_c
         dy = dy.contiguous()
         dx = dw = None
-        if x.is_cuda:
+        if x.is_cuda and x.dtype == torch.bfloat16:
             from ..ops.extension import require_extension
             ext = require_extension()
             if ctx.needs_input_grad[0]:

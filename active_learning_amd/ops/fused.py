@@ -53,7 +53,8 @@ def conv_bn_act(conv, bn, x, residual=None):
         return batch_norm_act(y, bn.weight, bn.bias, bn.running_mean,
                               bn.running_var, True, bn.momentum, bn.eps, bn.relu,
                               residual, bn._pg(), pre_sums=(s, ss))
-    if (not _DISABLED and x.is_cuda and not bn.training
+    if (not _DISABLED and x.is_cuda and x.dtype == torch.bfloat16
+            and not bn.training
             and not torch.is_grad_enabled() and extension_available()):
         ext = load_extension()
         w_c = cast_cached(conv.weight, x.dtype)

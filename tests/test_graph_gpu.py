@@ -91,6 +91,48 @@ def test_graphed_step_tail_batch():
     assert torch.isfinite(loss2).item()
 
 
+def test_fp32_gpu_path():
+    """--compute_dtype fp32 on GPU: ops run on-device torch math; forward and
+    backward match the CPU fp32 reference to tight tolerance (1e-4)."""
+    from active_learning_amd.models import get_networks
+    torch.manual_seed(0)
+    net_c = get_networks("synthetic_cifar10", "SSLResNet18")
+    net_c.encoder.compute_dtype = None
+    torch.manual_seed(0)
+    net_g = get_networks("synthetic_cifar10", "SSLResNet18").cuda()
+    net_g.encoder.compute_dtype = None  # fp32 stays fp32 on GPU
+    x = torch.randn(4, 3, 16, 16)
+    y = torch.randint(0, 10, (4,))
+    net_c.train()
+    net_g.train()
+    out_c = net_c(x)
+    out_g = net_g(x.cuda())
+    rel = (out_g.cpu() - out_c).norm() / out_c.norm().clamp_min(1e-6)
+    assert rel.item() < 1e-4, f"fp32 GPU forward relerr {rel.item()}"
+    torch.nn.functional.cross_entropy(out_c, y).backward()
+    torch.nn.functional.cross_entropy(out_g, y.cuda()).backward()
+    for (n_, pc), (_, pg) in zip(net_c.named_parameters(), net_g.named_parameters()):
+        if pc.grad is None:
+            continue
+        denom = pc.grad.norm().clamp_min(1e-5)
+        err = ((pg.grad.cpu() - pc.grad).norm() / denom).item()
+        assert err < 5e-3, f"fp32 GPU grad mismatch at {n_}: {err}"
+
+
+def test_fp32_gpu_e2e_round(tmp_path):
+    """A full --compute_dtype fp32 debug round on the GPU."""
+    from active_learning_amd.cli import get_args
+    from active_learning_amd.main_al import main
+    args = get_args([
+        "--dataset", "synthetic_cifar10", "--rounds", "1", "--round_budget", "10",
+        "--n_epoch", "1", "--early_stop_patience", "1", "--debug_mode",
+        "--ckpt_path", str(tmp_path / "ck"), "--log_dir", str(tmp_path / "lg"),
+        "--model", "SSLResNet18", "--compute_dtype", "fp32"])
+    s = main(args)
+    assert s.net.encoder.compute_dtype is None
+    assert s.idxs_lb.sum() == 15
+
+
 def test_graphed_trainer_e2e():
     """Strategy.parallel_train_fn with AL_TRAIN_GRAPH on (default): a short
     round trains through graph replays and saves a best ckpt."""
