@@ -93,52 +93,115 @@ __global__ void kcenter_kernel(const float* __restrict__ dist,
         partial[nb + blockIdx.x] = b;
       }
       grid.sync();
-      // block 0 thread 0: pick mode, locate the target block, leave the
-      // in-block residual mass in g_target and the block id in g_found<0 form
-      if (blockIdx.x == 0 && tid == 0) {
-        float total = 0.f;
-        for (int b = 0; b < nb; ++b) total += partial[b];
-        int mode = (!__builtin_isfinite(total) || total <= 0.f) ? 1 : 0;
+      // block 0, wave 0: pick mode and locate the target block with a
+      // wave-parallel prefix over the <=512 per-block partials
+      if (blockIdx.x == 0 && tid < kWave) {
+        const int lane = tid;
+        float wsum = 0.f, csum = 0.f;
+        for (int b = lane; b < nb; b += kWave) {
+          wsum += partial[b];
+          csum += partial[nb + b];
+        }
+        float total = wave_bcast(wave_reduce_sum(wsum), 0);
+        const float ctotal = wave_bcast(wave_reduce_sum(csum), 0);
+        const int mode = (!__builtin_isfinite(total) || total <= 0.f) ? 1 : 0;
         const float* p = partial + (mode ? nb : 0);
-        float tot = total;
-        if (mode) {
-          tot = 0.f;
-          for (int b = 0; b < nb; ++b) tot += p[b];
-        }
+        const float tot = mode ? ctotal : total;
         float target = randu[t] * tot;
-        int tb = nb - 1;
+        // sequential over ceil(nb/64) strips, wave-scan within each
+        int tb = -1;
         float acc = 0.f;
-        for (int b = 0; b < nb; ++b) {
-          if (target < acc + p[b]) { tb = b; target -= acc; break; }
-          acc += p[b];
+        for (int s0 = 0; s0 < nb && tb < 0; s0 += kWave) {
+          const int b = s0 + lane;
+          float v = (b < nb) ? p[b] : 0.f;
+          // inclusive wave prefix sum
+          float pref = v;
+#pragma unroll
+          for (int off = 1; off < kWave; off <<= 1) {
+            const float o = __shfl_up(pref, off, 64);
+            if (lane >= off) pref += o;
+          }
+          const float strip = wave_bcast(pref, kWave - 1);
+          // first lane whose [acc+pref-v, acc+pref) contains target
+          const bool hit = (b < nb) && (target < acc + pref) &&
+                           (target >= acc + pref - v);
+          const unsigned long long ball = __ballot(hit);
+          if (ball) {
+            const int l = __ffsll((unsigned long long)ball) - 1;
+            const float pref_l = wave_bcast(pref, l);
+            const float v_l = wave_bcast(v, l);
+            tb = s0 + l;
+            target -= acc + pref_l - v_l;
+          }
+          acc += strip;
         }
-        // target may exceed the tail block's mass by rounding: clamp later
-        st->g_mode = mode;
-        st->g_target = target;
-        st->g_found = -1 - tb;   // negative encoding: block tb scans
+        if (tb < 0) tb = nb - 1;  // rounding tail
+        if (lane == 0) {
+          st->g_mode = mode;
+          st->g_target = target;
+          st->g_found = -1 - tb;   // negative encoding: block tb scans
+        }
       }
       grid.sync();
-      // pass 2: the selected block scans its range serially (chunk is tiny)
-      if (-1 - st->g_found == blockIdx.x && tid == 0) {
+      // pass 2: the selected block finds its in-range index with a
+      // block-parallel prefix (per-thread contiguous sub-ranges, LDS scan)
+      if (-1 - st->g_found == blockIdx.x) {
         const int mode = st->g_mode;
-        float target = st->g_target;
-        long pick = -1, last_ok = -1;
-        for (long i = i0; i < i1; ++i) {
+        const long span = i1 - i0;
+        const long per = (span + blockDim.x - 1) / blockDim.x;
+        const long j0 = i0 + tid * per;
+        const long j1 = min(i1, j0 + per);
+        float mysum = 0.f;
+        for (long i = j0; i < j1; ++i) {
           if (labeled[i]) continue;
-          const float w = mode ? 1.f : fmaxf(min_dist[i], 0.f);
-          if (mode == 0 && w <= 0.f) continue;
-          last_ok = i;
-          if (target < w) { pick = i; break; }
-          target -= w;
+          mysum += mode ? 1.f : fmaxf(min_dist[i], 0.f);
         }
-        if (pick < 0) pick = last_ok;   // rounding tail: take the last valid
-        if (pick < 0) {                 // fully labeled/zero block (degenerate)
+        __shared__ float s_pref[256];
+        s_pref[tid] = mysum;
+        __syncthreads();
+        // simple Hillis-Steele inclusive scan over 256 entries
+        for (int off = 1; off < (int)blockDim.x; off <<= 1) {
+          float add = (tid >= off) ? s_pref[tid - off] : 0.f;
+          __syncthreads();
+          s_pref[tid] += add;
+          __syncthreads();
+        }
+        const float target = st->g_target;
+        const float excl = (tid == 0) ? 0.f : s_pref[tid - 1];
+        const bool mine = (mysum > 0.f) && (target >= excl) &&
+                          (target < excl + mysum);
+        const bool tail = (tid == (int)blockDim.x - 1) &&
+                          (target >= s_pref[blockDim.x - 1]);
+        if (mine || (tail && mysum > 0.f)) {
+          float rem = mine ? (target - excl) : mysum * 0.999f;
+          long pick = -1, last_ok = -1;
+          for (long i = j0; i < j1; ++i) {
+            if (labeled[i]) continue;
+            const float w = mode ? 1.f : fmaxf(min_dist[i], 0.f);
+            if (mode == 0 && w <= 0.f) continue;
+            last_ok = i;
+            if (rem < w) { pick = i; break; }
+            rem -= w;
+          }
+          if (pick < 0) pick = last_ok;
+          if (pick >= 0) {
+            sel[t] = pick;
+            labeled[pick] = 1;
+            st->g_j = (int)pick;
+            st->g_found = (int)1;  // claimed
+          }
+        }
+        __syncthreads();
+        if (tid == 0 && st->g_found < 0) {
+          // degenerate: every sub-range empty — fall back to first unlabeled
           for (long i = 0; i < n; ++i)
-            if (!labeled[i]) { pick = i; break; }
+            if (!labeled[i]) {
+              sel[t] = i;
+              labeled[i] = 1;
+              st->g_j = (int)i;
+              break;
+            }
         }
-        sel[t] = pick;
-        labeled[pick] = 1;
-        st->g_j = (int)pick;
       }
       grid.sync();
     } else {
