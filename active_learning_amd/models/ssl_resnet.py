@@ -10,6 +10,7 @@ MASE boundary sanity check, mase_sampler.py:85-90).
 import torch
 import torch.nn as nn
 
+from ..ops.linear import NativeLinear
 from .resnet import resnet18_encoder, resnet50_encoder
 
 _ENCODERS = {"resnet18": resnet18_encoder, "resnet50": resnet50_encoder}
@@ -24,7 +25,9 @@ class ResNetSimCLR(nn.Module):
         self.encoder = _ENCODERS[base_model](cifar_stem=cifar_stem)
         self.dim_mlp = self.encoder.embed_dim
         self.num_classes = num_classes
-        self.linear = nn.Linear(self.dim_mlp, num_classes)
+        # state_dict-compatible nn.Linear running first-party MFMA GEMMs on
+        # CUDA fp32 (ops/linear.py; reference: resnet_simclr.py:22)
+        self.linear = NativeLinear(self.dim_mlp, num_classes)
         self.freeze_feature = False
 
     def forward(self, x, return_features=False, specify_input_layer=None):

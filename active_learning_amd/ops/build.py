@@ -14,7 +14,7 @@ import sys
 import sysconfig
 
 HIP_SOURCES = ["bn.hip", "pool.hip", "scoring.hip", "optim.hip", "igemm.hip",
-               "wgrad.hip", "kcenter.hip"]
+               "wgrad.hip", "kcenter.hip", "linear.hip"]
 BINDING = "bindings.cpp"
 
 PKG_DIR = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))  # .../active_learning_amd

@@ -69,6 +69,10 @@ int al_kcenter_greedy(const float* dist, float* min_dist, unsigned char* labeled
                       long* sel, const float* randu, float* partial, void* st,
                       long n, int iters, int j_init, int randomize, int nblocks,
                       void* stream);
+void al_linear_fwd(const float* x, const float* w, const float* bias, float* out,
+                   int B, int M, int C, void* stream);
+void al_linear_bwd(const float* dy, const float* x, const float* w, float* dx,
+                   float* dw, float* db, int B, int M, int C, void* stream);
 }
 
 namespace {
@@ -482,6 +486,34 @@ int64_t kcenter_greedy_dev(const Tensor& dist, Tensor& min_dist, Tensor& labeled
       cur_stream());
 }
 
+Tensor linear_fwd(const Tensor& x, const Tensor& w, const Tensor& bias) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kFloat32 && x.is_contiguous());
+  TORCH_CHECK(w.scalar_type() == torch::kFloat32 && w.is_contiguous());
+  const int B = (int)x.size(0), M = (int)x.size(1), C = (int)w.size(0);
+  TORCH_CHECK(w.size(1) == M);
+  Tensor out = torch::empty({B, C}, x.options());
+  al_linear_fwd(x.data_ptr<float>(), w.data_ptr<float>(),
+                bias.numel() ? bias.data_ptr<float>() : nullptr,
+                out.data_ptr<float>(), B, M, C, cur_stream());
+  return out;
+}
+
+std::vector<Tensor> linear_bwd(const Tensor& dy, const Tensor& x, const Tensor& w,
+                               bool want_dx, bool want_dw, bool want_db) {
+  TORCH_CHECK(dy.is_cuda() && dy.scalar_type() == torch::kFloat32 &&
+              dy.is_contiguous());
+  const int B = (int)x.size(0), M = (int)x.size(1), C = (int)w.size(0);
+  Tensor dx = want_dx ? torch::empty({B, M}, x.options()) : Tensor();
+  Tensor dw = want_dw ? torch::empty({C, M}, x.options()) : Tensor();
+  Tensor db = want_db ? torch::empty({C}, x.options()) : Tensor();
+  al_linear_bwd(dy.data_ptr<float>(), x.contiguous().data_ptr<float>(),
+                w.contiguous().data_ptr<float>(),
+                want_dx ? dx.data_ptr<float>() : nullptr,
+                want_dw ? dw.data_ptr<float>() : nullptr,
+                want_db ? db.data_ptr<float>() : nullptr, B, M, C, cur_stream());
+  return {dx, dw, db};
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -508,5 +540,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sgd_step_multi", &sgd_step_multi);
   m.def("sgd_step_multi_dev", &sgd_step_multi_dev);
   m.def("kcenter_greedy_dev", &kcenter_greedy_dev);
+  m.def("linear_fwd", &linear_fwd);
+  m.def("linear_bwd", &linear_bwd);
   m.def("adam_step", &adam_step);
 }

@@ -1,0 +1,141 @@
+// First-party linear-head GEMM (fp32) on the exact-f32 MFMA.
+//
+// The classifier head (resnet_simclr.py:22,33,38) is (B,M)x(M,C) with
+// B<=256, M<=2048, C<=1000 — tiny next to the convs, but it is the last
+// library GEMM on the hot path (COVERAGE row 68). CDNA4 has an EXACT
+// fp32-input MFMA (mfma_f32_16x16x4f32, guide §3: 157 TF rate, identical
+// numerics to VALU fp32), so the head and both its backward GEMMs run
+// first-party without any precision change vs the reference's fp32 Linear.
+//
+// One templated kernel serves the three contractions:
+//   fwd: out(B,C) = X(B,M) @ W(C,M)^T  -> TA=0 TB=1 (+bias)
+//   dx:  out(B,M) = dY(B,C) @ W(C,M)   -> TA=0 TB=0
+//   dw:  out(C,M) = dY^T(C,B) @ X(B,M) -> TA=1 TB=0
+// 64x64 tiles, 4 waves (2x2, 32x32 each), 64-deep fp32 LDS staging with a
+// +1-float row pad (stride 65 => conflict-free strided access).
+
+#include "al_common.h"
+
+constexpr int LDP = 65;  // padded LDS row stride (floats)
+
+template <bool TA, bool TB, bool BIAS>
+__launch_bounds__(256)
+__global__ void linear_gemm_kernel(const float* __restrict__ A,
+                                   const float* __restrict__ Bm,
+                                   const float* __restrict__ bias,
+                                   float* __restrict__ out,
+                                   int M_, int N_, int K_, int lda, int ldb,
+                                   int grid_m) {
+  __shared__ float As[64 * LDP];
+  __shared__ float Bs[64 * LDP];
+  const int bm = blockIdx.x % grid_m, bn = blockIdx.x / grid_m;
+  const int r0 = bm * 64, n0 = bn * 64;
+  const int tid = threadIdx.x;
+  const int wid = tid >> 6, lane = tid & 63;
+  const int wr = wid >> 1, wc = wid & 1;
+  const int l15 = lane & 15, l4 = lane >> 4;
+
+  f32x4 acc[2][2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  for (int k0 = 0; k0 < K_; k0 += 64) {
+    // stage A tile [row][k]
+    for (int idx = tid; idx < 64 * 64; idx += 256) {
+      int r, k;
+      if (TA) { k = idx >> 6; r = idx & 63; }          // k-major global reads
+      else    { r = idx >> 6; k = idx & 63; }
+      const int gr = r0 + r, gk = k0 + k;
+      float v = 0.f;
+      if (gr < M_ && gk < K_)
+        v = TA ? A[(long)gk * lda + gr] : A[(long)gr * lda + gk];
+      As[r * LDP + k] = v;
+    }
+    // stage B tile [k][n]
+    for (int idx = tid; idx < 64 * 64; idx += 256) {
+      int k, n;
+      if (TB) { n = idx >> 6; k = idx & 63; }          // n-major global reads
+      else    { k = idx >> 6; n = idx & 63; }
+      const int gn = n0 + n, gk = k0 + k;
+      float v = 0.f;
+      if (gn < N_ && gk < K_)
+        v = TB ? Bm[(long)gn * ldb + gk] : Bm[(long)gk * ldb + gn];
+      Bs[k * LDP + n] = v;
+    }
+    __syncthreads();
+#pragma unroll 4
+    for (int kk = 0; kk < 16; ++kk) {
+      float a[2], b[2];
+#pragma unroll
+      for (int fi = 0; fi < 2; ++fi)
+        a[fi] = As[(wr * 32 + fi * 16 + l15) * LDP + kk * 4 + l4];
+#pragma unroll
+      for (int fj = 0; fj < 2; ++fj)
+        b[fj] = Bs[(kk * 4 + l4) * LDP + wc * 32 + fj * 16 + l15];
+#pragma unroll
+      for (int fi = 0; fi < 2; ++fi)
+#pragma unroll
+        for (int fj = 0; fj < 2; ++fj)
+          acc[fi][fj] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+              a[fi], b[fj], acc[fi][fj], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int fi = 0; fi < 2; ++fi)
+#pragma unroll
+    for (int fj = 0; fj < 2; ++fj) {
+      const int col = n0 + wc * 32 + fj * 16 + l15;
+      if (col >= N_) continue;
+      const float bv = BIAS ? bias[col] : 0.f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = r0 + wr * 32 + fi * 16 + l4 * 4 + r;
+        if (row < M_) out[(long)row * N_ + col] = acc[fi][fj][r] + bv;
+      }
+    }
+}
+
+// column sums of dY (B,C) -> db (C)
+__global__ void colsum_kernel(const float* __restrict__ dy, float* __restrict__ db,
+                              int B, int C) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float s = 0.f;
+  for (int b = 0; b < B; ++b) s += dy[(long)b * C + c];
+  db[c] = s;
+}
+
+extern "C" void al_linear_fwd(const float* x, const float* w, const float* bias,
+                              float* out, int B, int M, int C,
+                              hipStream_t stream) {
+  const int gm = (B + 63) / 64, gn = (C + 63) / 64;
+  if (bias)
+    hipLaunchKernelGGL((linear_gemm_kernel<false, true, true>), dim3(gm * gn),
+                       dim3(256), 0, stream, x, w, bias, out, B, C, M, M, M, gm);
+  else
+    hipLaunchKernelGGL((linear_gemm_kernel<false, true, false>), dim3(gm * gn),
+                       dim3(256), 0, stream, x, w, nullptr, out, B, C, M, M, M, gm);
+}
+
+extern "C" void al_linear_bwd(const float* dy, const float* x, const float* w,
+                              float* dx, float* dw, float* db, int B, int M,
+                              int C, hipStream_t stream) {
+  if (dx) {
+    const int gm = (B + 63) / 64, gn = (M + 63) / 64;
+    hipLaunchKernelGGL((linear_gemm_kernel<false, false, false>), dim3(gm * gn),
+                       dim3(256), 0, stream, dy, w, nullptr, dx, B, M, C, C, M, gm);
+  }
+  if (dw) {
+    const int gm = (C + 63) / 64, gn = (M + 63) / 64;
+    hipLaunchKernelGGL((linear_gemm_kernel<true, false, false>), dim3(gm * gn),
+                       dim3(256), 0, stream, dy, x, nullptr, dw, C, M, B, C, M, gm);
+  }
+  if (db) {
+    hipLaunchKernelGGL(colsum_kernel, dim3((C + 255) / 256), dim3(256), 0,
+                       stream, dy, db, B, C);
+  }
+}

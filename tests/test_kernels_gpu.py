@@ -129,6 +129,32 @@ def test_conv_bwd_weight_v4(case):
     assert err < 0.03, f"wgrad v4 {case}: relerr {err}"
 
 
+@pytest.mark.parametrize("shape", [(16, 512, 10), (128, 2048, 1000),
+                                   (100, 100, 77)])
+def test_linear_head_gpu(shape):
+    """First-party fp32 MFMA linear head fwd/bwd vs torch (exact-f32 MFMA:
+    tight tolerance)."""
+    from active_learning_amd.ops.linear import linear
+    b, m, c = shape
+    torch.manual_seed(1)
+    x = torch.randn(b, m, device="cuda", requires_grad=True)
+    w = torch.randn(c, m, device="cuda", requires_grad=True) * 0.05
+    w = w.detach().requires_grad_(True)
+    bias = torch.randn(c, device="cuda", requires_grad=True)
+    y = linear(x, w, bias)
+    ref = torch.nn.functional.linear(x.detach(), w.detach(), bias.detach())
+    assert relerr(y, ref) < 1e-5, f"linear fwd {shape}"
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    xr = x.detach().requires_grad_(True)
+    wr = w.detach().requires_grad_(True)
+    br = bias.detach().requires_grad_(True)
+    torch.nn.functional.linear(xr, wr, br).backward(dy)
+    assert relerr(x.grad, xr.grad) < 1e-5, f"linear dx {shape}"
+    assert relerr(w.grad, wr.grad) < 1e-5, f"linear dw {shape}"
+    assert relerr(bias.grad, br.grad) < 1e-5, f"linear db {shape}"
+
+
 def test_kcenter_persistent_kernel():
     """The cooperative persistent k-center kernel must select exactly the
     same points as the torch reference loop (deterministic mode), and produce
