@@ -233,12 +233,11 @@ class Conv2dNHWC(Function):
             if ctx.needs_input_grad[0]:
                 if R == 1 and S == 1 and ctx.stride > 1 and ctx.padding == 0:
                     # 1x1 strided conv (ResNet downsample): dx is nonzero only
-                    # at stride-aligned pixels — dense GEMM on dY + strided
-                    # scatter instead of a 4x-redundant gather igemm
+                    # at stride-aligned pixels — dense GEMM on dY + a one-pass
+                    # scatter kernel (zeros + copy fused) instead of a
+                    # 4x-redundant gather igemm
                     tmp = ext.conv2d_fwd(dy, wt_cached(w_c), 1, 0)  # (N,P,Q,C)
-                    n_, h_, w_ = x.shape[0], x.shape[1], x.shape[2]
-                    dx = torch.zeros(n_, h_, w_, C, dtype=dy.dtype, device=dy.device)
-                    dx[:, ::ctx.stride, ::ctx.stride, :] = tmp
+                    dx = ext.scatter_s2(tmp, x.shape[1], x.shape[2], ctx.stride)
                 else:
                     dx = ext.conv2d_bwd_data(dy, wt_cached(w_c), ctx.stride,
                                              ctx.padding, x.shape[1], x.shape[2])

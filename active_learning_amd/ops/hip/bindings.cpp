@@ -69,6 +69,8 @@ int al_kcenter_greedy(const float* dist, float* min_dist, unsigned char* labeled
                       long* sel, const float* randu, float* partial, void* st,
                       long n, int iters, int j_init, int randomize, int nblocks,
                       void* stream);
+void al_scatter_s2(const void* tmp, void* dx, int N, int H, int W, int C, int P,
+                   int Q, int stride, void* stream);
 void al_linear_fwd(const float* x, const float* w, const float* bias, float* out,
                    int B, int M, int C, void* stream);
 void al_linear_bwd(const float* dy, const float* x, const float* w, float* dx,
@@ -486,6 +488,17 @@ int64_t kcenter_greedy_dev(const Tensor& dist, Tensor& min_dist, Tensor& labeled
       cur_stream());
 }
 
+Tensor scatter_s2(const Tensor& tmp, int64_t H, int64_t W, int64_t stride) {
+  check_bf16_contig(tmp, "tmp");
+  const int N = (int)tmp.size(0), P = (int)tmp.size(1), Q = (int)tmp.size(2),
+            C = (int)tmp.size(3);
+  TORCH_CHECK(C % 8 == 0);
+  Tensor dx = torch::empty({(long)N, H, W, (long)C}, tmp.options());
+  al_scatter_s2(tmp.data_ptr(), dx.data_ptr(), N, (int)H, (int)W, C, P, Q,
+                (int)stride, cur_stream());
+  return dx;
+}
+
 Tensor linear_fwd(const Tensor& x, const Tensor& w, const Tensor& bias) {
   TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kFloat32 && x.is_contiguous());
   TORCH_CHECK(w.scalar_type() == torch::kFloat32 && w.is_contiguous());
@@ -540,6 +553,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sgd_step_multi", &sgd_step_multi);
   m.def("sgd_step_multi_dev", &sgd_step_multi_dev);
   m.def("kcenter_greedy_dev", &kcenter_greedy_dev);
+  m.def("scatter_s2", &scatter_s2);
   m.def("linear_fwd", &linear_fwd);
   m.def("linear_bwd", &linear_bwd);
   m.def("adam_step", &adam_step);

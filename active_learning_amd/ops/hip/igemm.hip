@@ -1080,6 +1080,44 @@ extern "C" void al_conv2d_mm(int mode, const void* A, const void* B, void* out,
 }
 
 // ---------------------------------------------------------------------------
+// strided scatter for the 1x1-stride-2 bwd-data route (ResNet downsample):
+// dx[n, h, w, c] = (h % s == 0 && w % s == 0) ? tmp[n, h/s, w/s, c] : 0
+// — one write pass replacing ATen's zeros-fill + strided-copy pair
+// (functional.py previously: torch.zeros + dx[:, ::2, ::2] = tmp).
+// ---------------------------------------------------------------------------
+
+__global__ void scatter_s2_kernel(const bf16* __restrict__ tmp,
+                                  bf16* __restrict__ dx, int N, int H, int W,
+                                  int C, int P, int Q, int stride) {
+  // one 16B chunk (8 bf16, C % 8 == 0) per iteration, fully coalesced
+  const long chunks = (long)N * H * W * (C / 8);
+  for (long i = grid_stride_begin(); i < chunks; i += grid_stride_step()) {
+    const int cc = (int)(i % (C / 8));
+    long t = i / (C / 8);
+    const int w = (int)(t % W);
+    t /= W;
+    const int h = (int)(t % H);
+    const long n = t / H;
+    s16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
+    if (h % stride == 0 && w % stride == 0) {
+      const int p = h / stride, q = w / stride;
+      if (p < P && q < Q)
+        v = *(const s16x8*)(tmp + (((n * P + p) * Q + q) * C) + cc * 8);
+    }
+    *(s16x8*)(dx + i * 8) = v;
+  }
+}
+
+extern "C" void al_scatter_s2(const void* tmp, void* dx, int N, int H, int W,
+                              int C, int P, int Q, int stride,
+                              hipStream_t stream) {
+  const long chunks = (long)N * H * W * (C / 8);
+  int blocks = (int)min((chunks + 255) / 256, (long)8192);
+  hipLaunchKernelGGL(scatter_s2_kernel, dim3(blocks), dim3(256), 0, stream,
+                     (const bf16*)tmp, (bf16*)dx, N, H, W, C, P, Q, stride);
+}
+
+// ---------------------------------------------------------------------------
 // im2col packing for shapes the MFMA kernel cannot gather directly (the C=3
 // stems): materialize A = im2col(x) zero-padded to KDpad columns (KDpad % 64
 // == 0); the conv then runs as a 1x1 MFMA igemm over the packed buffer.

@@ -76,7 +76,7 @@ def config4_badge(n_labeled=50_000, n_unlabeled=80_000, budget=10_000):
     res = {}
     (out, res["inference_s"]) = timed(
         f"imagenet pool inference ({n//1000}k)",
-        lambda: pool_inference(net, n, 512, 224))
+        lambda: pool_inference(net, n, 256, 224))
     logits, emb = out
     (ae, res["gram_s"]) = timed(
         "BADGE factorized Gram (130k x 130k)",
