@@ -99,6 +99,7 @@ def main(args):
     strategy.world_size = args.world_size
 
     os.makedirs(args.log_dir, exist_ok=True)
+    os.environ["AL_TRACK_LOG_DIR"] = args.log_dir  # inherited by spawned ranks
     today = date.today()
     log_filename = f"{args.exp_hash}_{today.month:02d}{today.day:02d}.log"
     logger = setup_logging(args.log_dir, log_filename)

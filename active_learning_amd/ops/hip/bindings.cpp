@@ -71,6 +71,7 @@ int al_kcenter_greedy(const float* dist, float* min_dist, unsigned char* labeled
                       void* stream);
 void al_scatter_s2(const void* tmp, void* dx, int N, int H, int W, int C, int P,
                    int Q, int stride, void* stream);
+int al_linear_needs_zero(int M_, int N_, int K_);
 void al_linear_fwd(const float* x, const float* w, const float* bias, float* out,
                    int B, int M, int C, void* stream);
 void al_linear_bwd(const float* dy, const float* x, const float* w, float* dx,
@@ -504,7 +505,9 @@ Tensor linear_fwd(const Tensor& x, const Tensor& w, const Tensor& bias) {
   TORCH_CHECK(w.scalar_type() == torch::kFloat32 && w.is_contiguous());
   const int B = (int)x.size(0), M = (int)x.size(1), C = (int)w.size(0);
   TORCH_CHECK(w.size(1) == M);
-  Tensor out = torch::empty({B, C}, x.options());
+  Tensor out = al_linear_needs_zero(B, C, M)
+                   ? torch::zeros({B, C}, x.options())
+                   : torch::empty({B, C}, x.options());
   al_linear_fwd(x.data_ptr<float>(), w.data_ptr<float>(),
                 bias.numel() ? bias.data_ptr<float>() : nullptr,
                 out.data_ptr<float>(), B, M, C, cur_stream());
@@ -516,8 +519,13 @@ std::vector<Tensor> linear_bwd(const Tensor& dy, const Tensor& x, const Tensor& 
   TORCH_CHECK(dy.is_cuda() && dy.scalar_type() == torch::kFloat32 &&
               dy.is_contiguous());
   const int B = (int)x.size(0), M = (int)x.size(1), C = (int)w.size(0);
-  Tensor dx = want_dx ? torch::empty({B, M}, x.options()) : Tensor();
-  Tensor dw = want_dw ? torch::empty({C, M}, x.options()) : Tensor();
+  auto alloc = [&](long r, long c, int K_) {
+    return al_linear_needs_zero((int)r, (int)c, K_)
+               ? torch::zeros({r, c}, x.options())
+               : torch::empty({r, c}, x.options());
+  };
+  Tensor dx = want_dx ? alloc(B, M, C) : Tensor();
+  Tensor dw = want_dw ? alloc(C, M, B) : Tensor();
   Tensor db = want_db ? torch::empty({C}, x.options()) : Tensor();
   al_linear_bwd(dy.data_ptr<float>(), x.contiguous().data_ptr<float>(),
                 w.contiguous().data_ptr<float>(),

@@ -18,7 +18,10 @@
 
 constexpr int LDP = 65;  // padded LDS row stride (floats)
 
-template <bool TA, bool TB, bool BIAS>
+// ZSPLIT: blockIdx.y strides the contraction so small (B,C) outputs still
+// fill the 256 CUs; partial tiles accumulate with fp32 atomics into a
+// zeroed output (bias applied by the first K-slice only).
+template <bool TA, bool TB, bool BIAS, bool ZSPLIT = false>
 __launch_bounds__(256)
 __global__ void linear_gemm_kernel(const float* __restrict__ A,
                                    const float* __restrict__ Bm,
@@ -41,7 +44,9 @@ __global__ void linear_gemm_kernel(const float* __restrict__ A,
 #pragma unroll
     for (int j = 0; j < 2; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  for (int k0 = 0; k0 < K_; k0 += 64) {
+  const int kz0 = ZSPLIT ? (int)blockIdx.y * 64 : 0;
+  const int kstep = ZSPLIT ? (int)gridDim.y * 64 : 64;
+  for (int k0 = kz0; k0 < K_; k0 += kstep) {
     // stage A tile [row][k]
     for (int idx = tid; idx < 64 * 64; idx += 256) {
       int r, k;
@@ -90,11 +95,13 @@ __global__ void linear_gemm_kernel(const float* __restrict__ A,
     for (int fj = 0; fj < 2; ++fj) {
       const int col = n0 + wc * 32 + fj * 16 + l15;
       if (col >= N_) continue;
-      const float bv = BIAS ? bias[col] : 0.f;
+      const float bv = (BIAS && (!ZSPLIT || blockIdx.y == 0)) ? bias[col] : 0.f;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int row = r0 + wr * 32 + fi * 16 + l4 * 4 + r;
-        if (row < M_) out[(long)row * N_ + col] = acc[fi][fj][r] + bv;
+        if (row >= M_) continue;
+        if (ZSPLIT) atomicAdd(&out[(long)row * N_ + col], acc[fi][fj][r] + bv);
+        else out[(long)row * N_ + col] = acc[fi][fj][r] + bv;
       }
     }
 }
@@ -109,16 +116,38 @@ __global__ void colsum_kernel(const float* __restrict__ dy, float* __restrict__ 
   db[c] = s;
 }
 
+static inline int lin_zsplit(int tiles, int K_) {
+  int z = 256 / (tiles < 1 ? 1 : tiles);
+  const int kt = (K_ + 63) / 64;
+  if (z > kt) z = kt;
+  return z < 1 ? 1 : z;
+}
+
+// whether the launcher will K-split (caller must then pass a ZEROED out)
+extern "C" int al_linear_needs_zero(int M_, int N_, int K_) {
+  return lin_zsplit(((M_ + 63) / 64) * ((N_ + 63) / 64), K_) > 1;
+}
+
 extern "C" void al_linear_fwd(const float* x, const float* w, const float* bias,
                               float* out, int B, int M, int C,
                               hipStream_t stream) {
   const int gm = (B + 63) / 64, gn = (C + 63) / 64;
-  if (bias)
-    hipLaunchKernelGGL((linear_gemm_kernel<false, true, true>), dim3(gm * gn),
+  const int z = lin_zsplit(gm * gn, M);
+  dim3 grid(gm * gn, z);
+  if (z > 1) {
+    if (bias)
+      hipLaunchKernelGGL((linear_gemm_kernel<false, true, true, true>), grid,
+                         dim3(256), 0, stream, x, w, bias, out, B, C, M, M, M, gm);
+    else
+      hipLaunchKernelGGL((linear_gemm_kernel<false, true, false, true>), grid,
+                         dim3(256), 0, stream, x, w, nullptr, out, B, C, M, M, M, gm);
+  } else if (bias) {
+    hipLaunchKernelGGL((linear_gemm_kernel<false, true, true>), grid,
                        dim3(256), 0, stream, x, w, bias, out, B, C, M, M, M, gm);
-  else
-    hipLaunchKernelGGL((linear_gemm_kernel<false, true, false>), dim3(gm * gn),
+  } else {
+    hipLaunchKernelGGL((linear_gemm_kernel<false, true, false>), grid,
                        dim3(256), 0, stream, x, w, nullptr, out, B, C, M, M, M, gm);
+  }
 }
 
 extern "C" void al_linear_bwd(const float* dy, const float* x, const float* w,
@@ -126,13 +155,25 @@ extern "C" void al_linear_bwd(const float* dy, const float* x, const float* w,
                               int C, hipStream_t stream) {
   if (dx) {
     const int gm = (B + 63) / 64, gn = (M + 63) / 64;
-    hipLaunchKernelGGL((linear_gemm_kernel<false, false, false>), dim3(gm * gn),
-                       dim3(256), 0, stream, dy, w, nullptr, dx, B, M, C, C, M, gm);
+    const int z = lin_zsplit(gm * gn, C);
+    dim3 grid(gm * gn, z);
+    if (z > 1)
+      hipLaunchKernelGGL((linear_gemm_kernel<false, false, false, true>), grid,
+                         dim3(256), 0, stream, dy, w, nullptr, dx, B, M, C, C, M, gm);
+    else
+      hipLaunchKernelGGL((linear_gemm_kernel<false, false, false>), grid,
+                         dim3(256), 0, stream, dy, w, nullptr, dx, B, M, C, C, M, gm);
   }
   if (dw) {
     const int gm = (C + 63) / 64, gn = (M + 63) / 64;
-    hipLaunchKernelGGL((linear_gemm_kernel<true, false, false>), dim3(gm * gn),
-                       dim3(256), 0, stream, dy, x, nullptr, dw, C, M, B, C, M, gm);
+    const int z = lin_zsplit(gm * gn, B);
+    dim3 grid(gm * gn, z);
+    if (z > 1)
+      hipLaunchKernelGGL((linear_gemm_kernel<true, false, false, true>), grid,
+                         dim3(256), 0, stream, dy, x, nullptr, dw, C, M, B, C, M, gm);
+    else
+      hipLaunchKernelGGL((linear_gemm_kernel<true, false, false>), grid,
+                         dim3(256), 0, stream, dy, x, nullptr, dw, C, M, B, C, M, gm);
   }
   if (db) {
     hipLaunchKernelGGL(colsum_kernel, dim3((C + 255) / 256), dim3(256), 0,

@@ -33,7 +33,10 @@ class Experiment:
                  experiment_key=None, mirror_comet=True, **_ignored):
         self.project_name = project_name
         self.disabled = disabled
-        self.log_dir = log_dir
+        # spawned trainer ranks re-attach ExistingExperiment without the CLI
+        # args in scope (strategy.py:290-293); the parent exports the real
+        # log dir so the rank-0 JSONL lands in the same place
+        self.log_dir = os.environ.get("AL_TRACK_LOG_DIR", log_dir)
         self.key = experiment_key or uuid.uuid4().hex[:16]
         self.name = None
         self.tags = []

@@ -2,6 +2,7 @@
 BucketedDDP gradient averaging, unused-param handling, SyncBN statistics,
 and packed eval gathering. The same code paths run over RCCL on MI355X."""
 
+import json
 import os
 
 import numpy as np
@@ -207,6 +208,24 @@ def _strategy_train_spawned(rank, world):
     s.backend = "gloo"
     s.update(np.arange(20), 20)
     s._init_distributed = lambda r: None  # group already up via _entry
+    # simulate the spawn environment (strategy.py:290-293): the parent
+    # detached the tracker and rank 0 re-attaches an ExistingExperiment
+    log_dir = os.path.join(s.base_ckpt_path, "track_logs")
+    s.comet_exp_key = "gloo2key"
+    s.comet_experiment = None
+    os.environ["AL_TRACK_LOG_DIR"] = log_dir
+    s.es_params["use_es"] = True
+    s.es_params["patience"] = 1000  # never early-stop before the log cadence
+    s.n_epoch = 25  # epoch 25 hits the rank-0 validation metric log cadence
+    s.train_args["loader_tr_args"]["batch_size"] = 20  # keep epochs fast
     s.parallel_train_fn(rank)
     if rank == 0:
         assert os.path.exists(s.generate_weight_paths()["best_ckpt"])
+        assert s.comet_experiment is not None
+        assert s.comet_experiment.get_key() == "gloo2key"
+        # the re-attached tracker wrote rank-0 validation metrics to the
+        # SAME key's JSONL
+        path = os.path.join(log_dir, "metrics_gloo2key.jsonl")
+        assert os.path.exists(path), "re-attached tracker JSONL missing"
+        names = [json.loads(line).get("name", "") for line in open(path)]
+        assert any("validation_accuracy" in n for n in names), names
