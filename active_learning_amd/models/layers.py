@@ -70,7 +70,8 @@ class BatchNormAct2d(nn.Module):
     def forward(self, x, residual=None):
         return AF.batch_norm_act(x, self.weight, self.bias, self.running_mean,
                                  self.running_var, self.training, self.momentum,
-                                 self.eps, self.relu, residual, self._pg())
+                                 self.eps, self.relu, residual, self._pg(),
+                                 fuse_backward=True)
 
     def extra_repr(self):
         return f"{self.num_features}, relu={self.relu}"

@@ -195,6 +195,25 @@ class _GradArena:
 
 _grad_arena = _GradArena()
 
+_BNBACK_ON = os.environ.get("AL_BNBACK_FUSE", "1") == "1"
+
+
+def _claim_bnback(x):
+    """Claim the upstream BatchNorm's backward side-channel (set by
+    BatchNormAct.forward on its output when it has a single consumer): this
+    conv's bwd-data will then mask dx in its epilogue and pre-reduce the
+    BN-backward channel sums, so bn_bwd_reduce never runs for that BN."""
+    if not _BNBACK_ON:
+        return None
+    bnb = getattr(x, "_al_bnback", None)
+    if bnb is None or getattr(x, "_al_bnback_claimed", False):
+        return None
+    try:
+        x._al_bnback_claimed = True
+    except Exception:
+        return None
+    return bnb
+
 
 class Conv2dNHWC(Function):
     """y[N,P,Q,K] = conv(x[N,H,W,C], w[K,R,S,C]; stride, pad), no bias
@@ -207,6 +226,7 @@ class Conv2dNHWC(Function):
         ctx.w_c = w_c
         ctx.stride, ctx.padding = stride, padding
         ctx.weight_dtype = weight.dtype
+        ctx.bnb = _claim_bnback(x)
         if _native_ok(x):
             ext = require_extension()
             _grad_arena.mark_step()
@@ -238,6 +258,19 @@ class Conv2dNHWC(Function):
                     # 4x-redundant gather igemm
                     tmp = ext.conv2d_fwd(dy, wt_cached(w_c), 1, 0)  # (N,P,Q,C)
                     dx = ext.scatter_s2(tmp, x.shape[1], x.shape[2], ctx.stride)
+                elif (getattr(ctx, "bnb", None) is not None
+                      and (R * S * K) % 64 == 0 and K % 8 == 0):
+                    # fuse the upstream BN's backward reduction into this
+                    # bwd-data epilogue: dx comes back PRE-MASKED with the
+                    # channel sums attached for BatchNormAct.backward
+                    mask, xbn, mean, invstd = ctx.bnb
+                    dx, s, sx = ext.conv2d_bwd_data_bn(
+                        dy, wt_cached(w_c), ctx.stride, ctx.padding,
+                        x.shape[1], x.shape[2], mask, xbn, mean, invstd)
+                    try:
+                        dx._al_bnsums = (mask, s, sx)
+                    except Exception:
+                        pass
                 else:
                     dx = ext.conv2d_bwd_data(dy, wt_cached(w_c), ctx.stride,
                                              ctx.padding, x.shape[1], x.shape[2])
@@ -285,6 +318,7 @@ class Conv2dNHWCStats(Function):
         ctx.w_c = w_c
         ctx.stride, ctx.padding = stride, padding
         ctx.weight_dtype = weight.dtype
+        ctx.bnb = _claim_bnback(x)
         if _native_ok(x):
             ext = require_extension()
             _grad_arena.mark_step()
@@ -332,7 +366,8 @@ class BatchNormAct(Function):
 
     @staticmethod
     def forward(ctx, x, weight, bias, running_mean, running_var, use_batch_stats,
-                momentum, eps, relu, residual, pg, pre_sums=None):
+                momentum, eps, relu, residual, pg, pre_sums=None,
+                fuse_backward=False):
         n_local = x.numel() // x.shape[-1]
         sync = (pg is not None and dist.is_initialized()
                 and dist.get_world_size(pg) > 1)
@@ -443,6 +478,15 @@ class BatchNormAct(Function):
         ctx.has_residual = residual is not None
         ctx.pg = pg
         ctx.n = n
+        if (fuse_backward and relu and residual is None and relu_mask is not None
+                and relu_mask.numel() > 0 and _native_ok(x)):
+            # single-consumer side-channel: the downstream conv's bwd-data
+            # epilogue masks dy and pre-reduces (sum dy~, sum dy~*xhat), so
+            # the backward below can skip bn_bwd_reduce entirely
+            try:
+                y._al_bnback = (relu_mask, x, mean, invstd)
+            except Exception:
+                pass
         return y
 
     @staticmethod
@@ -450,11 +494,20 @@ class BatchNormAct(Function):
         x, mask_or_y, weight, mean, invstd = ctx.saved_tensors
         y = None if _native_ok(x) else mask_or_y
         dy = dy.contiguous()
+        premasked = False
         if _native_ok(x):
             ext = require_extension()
-            # reduce pass: per-channel sums of dy~ and dy~*xhat (dy~ = mask * dy)
-            sum_dy, sum_dy_xhat = ext.bn_bwd_reduce(dy, x, mask_or_y, mean, invstd,
-                                                    ctx.relu)
+            pre = getattr(dy, "_al_bnsums", None)
+            if (pre is not None and not ctx.has_residual and ctx.relu
+                    and pre[0].data_ptr() == mask_or_y.data_ptr()):
+                # sums were computed in the producing conv's bwd-data
+                # epilogue, and dy arrived PRE-MASKED — skip the reduce pass
+                sum_dy, sum_dy_xhat = pre[1], pre[2]
+                premasked = True
+            else:
+                # reduce pass: per-channel sums of dy~, dy~*xhat (dy~ = mask*dy)
+                sum_dy, sum_dy_xhat = ext.bn_bwd_reduce(dy, x, mask_or_y, mean,
+                                                        invstd, ctx.relu)
         else:
             dyf = dy.float()
             if ctx.relu:
@@ -483,7 +536,7 @@ class BatchNormAct(Function):
             ext = require_extension()
             dx, dres_t = ext.bn_bwd(dy, x, mask_or_y, mean, invstd, weight, sum_dy,
                                     sum_dy_xhat, n_global, ctx.use_batch_stats,
-                                    ctx.relu, ctx.has_residual)
+                                    ctx.relu and not premasked, ctx.has_residual)
             if ctx.has_residual:
                 dres = dres_t
         else:
@@ -499,15 +552,16 @@ class BatchNormAct(Function):
             else:
                 dx = g * dyf
             dx = dx.to(x.dtype)
-        return dx, dgamma, dbeta, None, None, None, None, None, None, dres, None, None
+        return (dx, dgamma, dbeta, None, None, None, None, None, None, dres,
+                None, None, None)
 
 
 def batch_norm_act(x, weight, bias, running_mean, running_var, use_batch_stats,
                    momentum=0.1, eps=1e-5, relu=True, residual=None, pg=None,
-                   pre_sums=None):
+                   pre_sums=None, fuse_backward=False):
     return BatchNormAct.apply(x, weight, bias, running_mean, running_var,
                               use_batch_stats, momentum, eps, relu, residual, pg,
-                              pre_sums)
+                              pre_sums, fuse_backward)
 
 
 class MaxPool2dNHWC(Function):

@@ -52,7 +52,8 @@ def conv_bn_act(conv, bn, x, residual=None):
         y, s, ss = conv2d_with_stats(x, conv.weight, conv.stride, conv.padding)
         return batch_norm_act(y, bn.weight, bn.bias, bn.running_mean,
                               bn.running_var, True, bn.momentum, bn.eps, bn.relu,
-                              residual, bn._pg(), pre_sums=(s, ss))
+                              residual, bn._pg(), pre_sums=(s, ss),
+                              fuse_backward=True)
     if (not _DISABLED and x.is_cuda and x.dtype == torch.bfloat16
             and not bn.training
             and not torch.is_grad_enabled() and extension_available()):

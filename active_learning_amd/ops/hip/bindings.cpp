@@ -18,7 +18,9 @@ void al_conv2d_mm(int mode, const void* A, const void* B, void* out,
                   const void* zero_page, int N, int H, int W, int C, int K, int R,
                   int S, int P, int Q, int stride, int pad, const float* epi_scale,
                   const float* epi_shift, const void* epi_res, int epi_relu,
-                  float* stat_sum, float* stat_sumsq, void* stream);
+                  float* stat_sum, float* stat_sumsq, const void* bnb_mask,
+                  const void* bnb_x, const float* bnb_mean,
+                  const float* bnb_invstd, void* stream);
 void al_conv2d_wgrad(const void* dy, const void* x, float* dw,
                      const void* zero_page, int N, int H, int W,
                      int C, int K, int R, int S, int P, int Q, int stride, int pad,
@@ -110,7 +112,7 @@ Tensor conv2d_fwd(const Tensor& x, const Tensor& w, long stride, long pad) {
   auto y = torch::empty({N, P, Q, K}, x.options());
   al_conv2d_mm(0, x.data_ptr(), w.data_ptr(), y.data_ptr(),
                zero_page(x).data_ptr(), N, H, W, C, K, R, S, P, Q, (int)stride,
-               (int)pad, nullptr, nullptr, nullptr, 0, nullptr, nullptr,
+               (int)pad, nullptr, nullptr, nullptr, 0, nullptr, nullptr, nullptr, nullptr, nullptr, nullptr,
                cur_stream());
   return y;
 }
@@ -135,7 +137,7 @@ std::vector<Tensor> conv2d_fwd_stats(const Tensor& x, const Tensor& w, long stri
   al_conv2d_mm(0, x.data_ptr(), w.data_ptr(), y.data_ptr(),
                zero_page(x).data_ptr(), N, H, W, C, K, R, S, P, Q, (int)stride,
                (int)pad, nullptr, nullptr, nullptr, 0, sum.data_ptr<float>(),
-               sumsq.data_ptr<float>(), cur_stream());
+               sumsq.data_ptr<float>(), nullptr, nullptr, nullptr, nullptr, cur_stream());
   return {y, sum, sumsq};
 }
 
@@ -159,7 +161,7 @@ Tensor conv2d_fwd_fused(const Tensor& x, const Tensor& w, long stride, long pad,
                (int)pad, scale.contiguous().data_ptr<float>(),
                shift.contiguous().data_ptr<float>(),
                has_res ? residual.data_ptr() : nullptr, relu ? 1 : 0, nullptr,
-               nullptr, cur_stream());
+               nullptr, nullptr, nullptr, nullptr, nullptr, cur_stream());
   return y;
 }
 
@@ -175,9 +177,40 @@ Tensor conv2d_bwd_data(const Tensor& dy, const Tensor& wt, long stride, long pad
   auto dx = torch::empty({N, (long)H, (long)W, C}, dy.options());
   al_conv2d_mm(1, dy.data_ptr(), wt.data_ptr(), dx.data_ptr(),
                zero_page(dy).data_ptr(), N, (int)H, (int)W, C, K, R, S, P, Q,
-               (int)stride, (int)pad, nullptr, nullptr, nullptr, 0, nullptr, nullptr,
+               (int)stride, (int)pad, nullptr, nullptr, nullptr, 0, nullptr, nullptr, nullptr, nullptr, nullptr, nullptr,
                cur_stream());
   return dx;
+}
+
+std::vector<Tensor> conv2d_bwd_data_bn(const Tensor& dy, const Tensor& wt,
+                                       long stride, long pad, long H, long W,
+                                       const Tensor& mask, const Tensor& xbn,
+                                       const Tensor& mean, const Tensor& invstd) {
+  // bwd-data with the upstream BatchNorm's backward reduction fused into the
+  // epilogue: dx is returned PRE-MASKED (dy~ = relu_mask * dx) along with
+  // per-channel (sum dy~, sum dy~*xhat) — bn_bwd_reduce never runs.
+  check_bf16_contig(dy, "dy");
+  check_bf16_contig(wt, "wt");
+  check_bf16_contig(xbn, "xbn");
+  const int N = dy.size(0), P = dy.size(1), Q = dy.size(2), K = dy.size(3);
+  const int C = wt.size(0), R = wt.size(1), S = wt.size(2);
+  TORCH_CHECK(wt.size(3) == K, "channel mismatch (wt must be CRSK)");
+  TORCH_CHECK(mask.scalar_type() == torch::kUInt8 && mask.is_contiguous());
+  TORCH_CHECK(mask.numel() * 8 == (long)N * H * W * C, "mask size");
+  TORCH_CHECK(xbn.numel() == (long)N * H * W * C, "xbn size");
+  TORCH_CHECK(mean.scalar_type() == torch::kFloat32 && mean.numel() == C);
+  TORCH_CHECK(invstd.scalar_type() == torch::kFloat32 && invstd.numel() == C);
+  auto dx = torch::empty({N, (long)H, (long)W, C}, dy.options());
+  auto both = torch::zeros({2L * C}, dy.options().dtype(torch::kFloat32));
+  auto sum = both.narrow(0, 0, C);
+  auto sumx = both.narrow(0, C, C);
+  al_conv2d_mm(1, dy.data_ptr(), wt.data_ptr(), dx.data_ptr(),
+               zero_page(dy).data_ptr(), N, (int)H, (int)W, C, K, R, S, P, Q,
+               (int)stride, (int)pad, nullptr, nullptr, nullptr, 0,
+               sum.data_ptr<float>(), sumx.data_ptr<float>(), mask.data_ptr(),
+               xbn.data_ptr(), mean.data_ptr<float>(),
+               invstd.data_ptr<float>(), cur_stream());
+  return {dx, sum, sumx};
 }
 
 Tensor conv2d_bwd_weight(const Tensor& dy, const Tensor& x, long R, long S,
@@ -542,6 +575,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv2d_fwd_fused", &conv2d_fwd_fused);
   m.def("conv2d_fwd_stats", &conv2d_fwd_stats);
   m.def("conv2d_bwd_data", &conv2d_bwd_data);
+  m.def("conv2d_bwd_data_bn", &conv2d_bwd_data_bn);
   m.def("conv2d_bwd_weight", &conv2d_bwd_weight);
   m.def("conv2d_bwd_weight_into", &conv2d_bwd_weight_into);
   m.def("bn_stats", &bn_stats);

@@ -126,7 +126,11 @@ __global__ void igemm_kernel(const bf16* __restrict__ A, const bf16* __restrict_
                              const float* __restrict__ epi_shift,
                              const bf16* __restrict__ epi_res, int epi_relu,
                              float* __restrict__ stat_sum,
-                             float* __restrict__ stat_sumsq) {
+                             float* __restrict__ stat_sumsq,
+                             const unsigned char* __restrict__ bnb_mask,
+                             const bf16* __restrict__ bnb_x,
+                             const float* __restrict__ bnb_mean,
+                             const float* __restrict__ bnb_invstd) {
   constexpr int BM = GWR * 64, BN = GWC * 64, BK = 64;
   constexpr int NA = 2 * GWR, NB = 2 * GWC;  // 16B chunks per thread/operand
   // XCD-aware block remap (T1): contiguous output tiles on one XCD share B
@@ -398,11 +402,28 @@ __global__ void igemm_kernel(const bf16* __restrict__ A, const bf16* __restrict_
     if (stat_sum) {
       // every chunk of this thread shares the same column group
       // (cc = tid % CPR since 256 % CPR == 0): accumulate in registers
+      if (bnb_mask) {
+        // BN-backward pre-reduce: this dx IS the upstream BatchNorm's dy.
+        // Mask it in place and accumulate (sum dy~, sum dy~*xhat) so the
+        // separate bn_bwd_reduce pass never runs (ops/functional.py).
+        const unsigned char mb = bnb_mask[off >> 3];
+        const s16x8 xv = *(const s16x8*)(bnb_x + off);
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const float f = (col0 + j < sh.Nout) ? bits2f(v[j]) : 0.f;
-        st_s[j] += f;
-        st_q[j] += f * f;
+        for (int j = 0; j < 8; ++j) {
+          float g = ((mb >> j) & 1) && (col0 + j < sh.Nout) ? bits2f(v[j]) : 0.f;
+          v[j] = f2bits(g);
+          st_s[j] += g;
+          if (col0 + j < sh.Nout)
+            st_q[j] += g * (bits2f(xv[j]) - bnb_mean[col0 + j]) *
+                       bnb_invstd[col0 + j];
+        }
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const float f = (col0 + j < sh.Nout) ? bits2f(v[j]) : 0.f;
+          st_s[j] += f;
+          st_q[j] += f * f;
+        }
       }
     }
     if (col0 + 8 <= sh.Nout) {
@@ -496,7 +517,11 @@ __global__ void gemm256_nt_kernel(const bf16* __restrict__ A,
                                   const float* __restrict__ epi_shift,
                                   const bf16* __restrict__ epi_res, int epi_relu,
                                   float* __restrict__ stat_sum,
-                                  float* __restrict__ stat_sumsq) {
+                                  float* __restrict__ stat_sumsq,
+                                  const unsigned char* __restrict__ bnb_mask,
+                                  const bf16* __restrict__ bnb_x,
+                                  const float* __restrict__ bnb_mean,
+                                  const float* __restrict__ bnb_invstd) {
   const long M = sh.M;
   const int Nout = sh.Nout, KD = sh.KD;
   constexpr int BM = BMT, BN = BNT, BK = 64;
@@ -801,11 +826,24 @@ __global__ void gemm256_nt_kernel(const bf16* __restrict__ A,
     const long off = m * Nout + col0;
     s16x8 v = *(const s16x8*)(etile + row * BN + cc * 8);
     if (stat_sum) {
+      if (bnb_mask) {  // BN-backward pre-reduce (see igemm_kernel epilogue)
+        const unsigned char mb = bnb_mask[off >> 3];
+        const s16x8 xv = *(const s16x8*)(bnb_x + off);
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const float f = bits2f(v[j]);
-        st_s[j] += f;
-        st_q[j] += f * f;
+        for (int j = 0; j < 8; ++j) {
+          float g = (mb >> j) & 1 ? bits2f(v[j]) : 0.f;
+          v[j] = f2bits(g);
+          st_s[j] += g;
+          st_q[j] += g * (bits2f(xv[j]) - bnb_mean[col0 + j]) *
+                     bnb_invstd[col0 + j];
+        }
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const float f = bits2f(v[j]);
+          st_s[j] += f;
+          st_q[j] += f * f;
+        }
       }
     }
     if (epi_scale) {
@@ -869,7 +907,11 @@ __global__ void conv_direct_kernel(const bf16* __restrict__ A,
                                    const float* __restrict__ epi_shift,
                                    const bf16* __restrict__ epi_res, int epi_relu,
                                    float* __restrict__ stat_sum,
-                                   float* __restrict__ stat_sumsq) {
+                                   float* __restrict__ stat_sumsq,
+                                   const unsigned char* __restrict__ bnb_mask,
+                                   const bf16* __restrict__ bnb_x,
+                                   const float* __restrict__ bnb_mean,
+                                   const float* __restrict__ bnb_invstd) {
   const long total = sh.M * sh.Nout;
   for (long i = grid_stride_begin(); i < total; i += grid_stride_step()) {
     const long m = i / sh.Nout;
@@ -909,7 +951,9 @@ extern "C" void al_conv2d_mm(int mode, const void* A, const void* B, void* out,
                              int K, int R, int S, int P, int Q, int stride, int pad,
                              const float* epi_scale, const float* epi_shift,
                              const void* epi_res, int epi_relu, float* stat_sum,
-                             float* stat_sumsq, hipStream_t stream) {
+                             float* stat_sumsq, const void* bnb_mask,
+                             const void* bnb_x, const float* bnb_mean,
+                             const float* bnb_invstd, hipStream_t stream) {
   ConvShape sh;
   sh.N = N; sh.H = H; sh.W = W; sh.C = C; sh.K = K; sh.R = R; sh.S = S;
   sh.P = P; sh.Q = Q; sh.stride = stride; sh.pad = pad;
@@ -960,12 +1004,12 @@ extern "C" void al_conv2d_mm(int mode, const void* A, const void* B, void* out,
             hipLaunchKernelGGL((igemm_kernel<MODE_BWD_S2, 4, 1>), grid, block, lds,
                                stream, (const bf16*)A, (const bf16*)B, (bf16*)out,
                                (const bf16*)zero_page, c, gm, epi_scale, epi_shift,
-                               (const bf16*)epi_res, epi_relu, stat_sum, stat_sumsq);
+                               (const bf16*)epi_res, epi_relu, stat_sum, stat_sumsq, (const unsigned char*)bnb_mask, (const bf16*)bnb_x, bnb_mean, bnb_invstd);
           else
             hipLaunchKernelGGL((igemm_kernel<MODE_BWD_S2, 2, 2>), grid, block, lds,
                                stream, (const bf16*)A, (const bf16*)B, (bf16*)out,
                                (const bf16*)zero_page, c, gm, epi_scale, epi_shift,
-                               (const bf16*)epi_res, epi_relu, stat_sum, stat_sumsq);
+                               (const bf16*)epi_res, epi_relu, stat_sum, stat_sumsq, (const unsigned char*)bnb_mask, (const bf16*)bnb_x, bnb_mean, bnb_invstd);
         }
       return;
     }
@@ -1033,7 +1077,8 @@ extern "C" void al_conv2d_mm(int mode, const void* A, const void* B, void* out,
       (gemm256_nt_kernel<MODE_, BNT_, BMT_>), \
       grid, block, lds, stream, (const bf16*)A, (const bf16*)B, (bf16*)out, \
       (const bf16*)zero_page, sh, grid_m, epi_scale, epi_shift, \
-      (const bf16*)epi_res, epi_relu, stat_sum, stat_sumsq)
+      (const bf16*)epi_res, epi_relu, stat_sum, stat_sumsq, \
+      (const unsigned char*)bnb_mask, (const bf16*)bnb_x, bnb_mean, bnb_invstd)
       if (bnt == 256) {
         if (pure) LAUNCH256(MODE_PURE, 256, 256);
         else if (mode == MODE_FWD) LAUNCH256(MODE_FWD, 256, 256);
@@ -1058,7 +1103,8 @@ extern "C" void al_conv2d_mm(int mode, const void* A, const void* B, void* out,
     const int grid_n = (sh.Nout + BN - 1) / BN;
     const size_t lds = 2 * (size_t)(BM + BN) * 64 * sizeof(bf16);
     dim3 grid(grid_m * grid_n), block(256);
-#define LAUNCH(MODE_, GWR_, GWC_) hipLaunchKernelGGL((igemm_kernel<MODE_, GWR_, GWC_>), grid, block, lds, stream, (const bf16*)A, (const bf16*)B, (bf16*)out, (const bf16*)zero_page, sh, grid_m, epi_scale, epi_shift, (const bf16*)epi_res, epi_relu, stat_sum, stat_sumsq)
+#define LAUNCH(MODE_, GWR_, GWC_) hipLaunchKernelGGL((igemm_kernel<MODE_, GWR_, GWC_>), grid, block, lds, stream, (const bf16*)A, (const bf16*)B, (bf16*)out, (const bf16*)zero_page, sh, grid_m, epi_scale, epi_shift, (const bf16*)epi_res, epi_relu, stat_sum, stat_sumsq, \
+      (const unsigned char*)bnb_mask, (const bf16*)bnb_x, bnb_mean, bnb_invstd)
     if (mode == MODE_FWD) {
       if (narrow) LAUNCH(MODE_FWD, 4, 1); else LAUNCH(MODE_FWD, 2, 2);
     } else {
@@ -1071,11 +1117,11 @@ extern "C" void al_conv2d_mm(int mode, const void* A, const void* B, void* out,
     if (mode == MODE_FWD)
       hipLaunchKernelGGL((conv_direct_kernel<MODE_FWD>), dim3(blocks), dim3(256), 0,
                          stream, (const bf16*)A, (const bf16*)B, (bf16*)out, sh,
-                         epi_scale, epi_shift, (const bf16*)epi_res, epi_relu, stat_sum, stat_sumsq);
+                         epi_scale, epi_shift, (const bf16*)epi_res, epi_relu, stat_sum, stat_sumsq, (const unsigned char*)bnb_mask, (const bf16*)bnb_x, bnb_mean, bnb_invstd);
     else
       hipLaunchKernelGGL((conv_direct_kernel<MODE_BWD_DATA>), dim3(blocks), dim3(256),
                          0, stream, (const bf16*)A, (const bf16*)B, (bf16*)out, sh,
-                         epi_scale, epi_shift, (const bf16*)epi_res, epi_relu, stat_sum, stat_sumsq);
+                         epi_scale, epi_shift, (const bf16*)epi_res, epi_relu, stat_sum, stat_sumsq, (const unsigned char*)bnb_mask, (const bf16*)bnb_x, bnb_mean, bnb_invstd);
   }
 }
 

@@ -129,6 +129,74 @@ def test_conv_bwd_weight_v4(case):
     assert err < 0.03, f"wgrad v4 {case}: relerr {err}"
 
 
+@pytest.mark.parametrize("case", [
+    (4, 16, 16, 64, 64, 3, 1, 1),      # 128^2 igemm route
+    (8, 56, 56, 64, 512, 3, 1, 1),     # gemm256 gather bwd route
+    (2, 14, 14, 256, 512, 1, 1, 0),    # gemm256 pure bwd route
+])
+def test_conv_bwd_data_bn_fused(case):
+    """conv2d_bwd_data_bn (BN-backward reduction fused into the bwd-data
+    epilogue) must equal conv2d_bwd_data + bn_bwd_reduce: pre-masked dx and
+    the per-channel (sum dy~, sum dy~*xhat)."""
+    n, h, w, c, k, r, stride, pad = case
+    p = (h + 2 * pad - r) // stride + 1
+    torch.manual_seed(2)
+    dy = torch.randn(n, p, p, k).cuda().to(torch.bfloat16)
+    wt = (torch.randn(k, r, r, c) * 0.05).cuda().to(torch.bfloat16)
+    wt_perm = wt.permute(3, 1, 2, 0).contiguous()
+    xbn = torch.randn(n, h, w, c).cuda().to(torch.bfloat16)
+    mean = torch.randn(c, device="cuda")
+    invstd = torch.rand(c, device="cuda") + 0.5
+    rows = n * h * w
+    mask = torch.randint(0, 256, (rows, c // 8), dtype=torch.uint8,
+                         device="cuda")
+
+    dx_ref = EXT.conv2d_bwd_data(dy, wt_perm, stride, pad, h, w)
+    s_ref, sx_ref = EXT.bn_bwd_reduce(dx_ref, xbn, mask, mean, invstd, True)
+    bits = torch.stack([(mask >> j) & 1 for j in range(8)], dim=2).reshape(
+        rows, c).to(torch.bfloat16).view(n, h, w, c)
+    dx_masked_ref = (dx_ref * bits)
+
+    dx, s, sx = EXT.conv2d_bwd_data_bn(dy, wt_perm, stride, pad, h, w, mask,
+                                       xbn, mean, invstd)
+    assert relerr(dx, dx_masked_ref) < 1e-3, "fused dx(pre-masked) mismatch"
+    assert relerr(s, s_ref) < 1e-3, "fused sum_dy mismatch"
+    assert relerr(sx, sx_ref) < 1e-3, "fused sum_dy_xhat mismatch"
+
+
+def test_resnet_bnback_fusion_matches_disabled():
+    """Model-level: gradients with the BN-backward fusion active must match
+    the unfused path (same kernels otherwise)."""
+    import os
+    from active_learning_amd.models import get_networks
+    from active_learning_amd.ops import functional as AFn
+    torch.manual_seed(4)
+    net = get_networks("synthetic_cifar10", "SSLResNet18").cuda()
+    x = torch.randn(8, 3, 16, 16, device="cuda")
+    y = torch.randint(0, 10, (8,), device="cuda")
+
+    def run(fuse):
+        old = AFn._BNBACK_ON
+        AFn._BNBACK_ON = fuse
+        try:
+            for p_ in net.parameters():
+                p_.grad = None
+            net.train()
+            out = net(x)
+            torch.nn.functional.cross_entropy(out, y).backward()
+            return {n_: p_.grad.detach().clone() for n_, p_ in
+                    net.named_parameters() if p_.grad is not None}
+        finally:
+            AFn._BNBACK_ON = old
+
+    g_off = run(False)
+    g_on = run(True)
+    assert g_on.keys() == g_off.keys()
+    for n_ in g_off:
+        err = relerr(g_on[n_], g_off[n_])
+        assert err < 5e-3, f"bnback fusion changed grad of {n_}: {err}"
+
+
 @pytest.mark.parametrize("shape", [(16, 512, 10), (128, 2048, 1000),
                                    (100, 100, 77)])
 def test_linear_head_gpu(shape):
