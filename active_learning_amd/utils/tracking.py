@@ -35,8 +35,11 @@ class Experiment:
         self.disabled = disabled
         # spawned trainer ranks re-attach ExistingExperiment without the CLI
         # args in scope (strategy.py:290-293); the parent exports the real
-        # log dir so the rank-0 JSONL lands in the same place
-        self.log_dir = os.environ.get("AL_TRACK_LOG_DIR", log_dir)
+        # log dir so the rank-0 JSONL lands in the same place. Only the
+        # DEFAULT is overridden — explicit log_dir arguments win.
+        if log_dir == "./logs":
+            log_dir = os.environ.get("AL_TRACK_LOG_DIR", log_dir)
+        self.log_dir = log_dir
         self.key = experiment_key or uuid.uuid4().hex[:16]
         self.name = None
         self.tags = []
