@@ -73,6 +73,8 @@ int al_kcenter_greedy(const float* dist, float* min_dist, unsigned char* labeled
                       void* stream);
 void al_scatter_s2(const void* tmp, void* dx, int N, int H, int W, int C, int P,
                    int Q, int stride, void* stream);
+void al_badge_gram(const float* a, const float* e, const float* d, float* out,
+                   long N, int Ka, int Ke, void* stream);
 int al_linear_needs_zero(int M_, int N_, int K_);
 void al_linear_fwd(const float* x, const float* w, const float* bias, float* out,
                    int B, int M, int C, void* stream);
@@ -533,6 +535,19 @@ Tensor scatter_s2(const Tensor& tmp, int64_t H, int64_t W, int64_t stride) {
   return dx;
 }
 
+Tensor badge_gram(const Tensor& a, const Tensor& e, const Tensor& d) {
+  TORCH_CHECK(a.is_cuda() && a.scalar_type() == torch::kFloat32 && a.is_contiguous());
+  TORCH_CHECK(e.scalar_type() == torch::kFloat32 && e.is_contiguous());
+  TORCH_CHECK(d.scalar_type() == torch::kFloat32 && d.is_contiguous());
+  const long N = a.size(0);
+  TORCH_CHECK(e.size(0) == N && d.numel() == N);
+  Tensor out = torch::empty({N, N}, a.options());
+  al_badge_gram(a.data_ptr<float>(), e.data_ptr<float>(), d.data_ptr<float>(),
+                out.data_ptr<float>(), N, (int)a.size(1), (int)e.size(1),
+                cur_stream());
+  return out;
+}
+
 Tensor linear_fwd(const Tensor& x, const Tensor& w, const Tensor& bias) {
   TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kFloat32 && x.is_contiguous());
   TORCH_CHECK(w.scalar_type() == torch::kFloat32 && w.is_contiguous());
@@ -596,6 +611,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sgd_step_multi_dev", &sgd_step_multi_dev);
   m.def("kcenter_greedy_dev", &kcenter_greedy_dev);
   m.def("scatter_s2", &scatter_s2);
+  m.def("badge_gram", &badge_gram);
   m.def("linear_fwd", &linear_fwd);
   m.def("linear_bwd", &linear_bwd);
   m.def("adam_step", &adam_step);

@@ -106,6 +106,98 @@ __global__ void linear_gemm_kernel(const float* __restrict__ A,
     }
 }
 
+// ---------------------------------------------------------------------------
+// Fused BADGE gradient-embedding pairwise distances:
+//   out[i][j] = d[i] + d[j] - 2 * (a_i . a_j) * (e_i . e_j)
+// (<g_i,g_j> = (a_i.a_j)(e_i.e_j) for g = a (x) e — the (B, C*M) embedding
+// never exists; reference materializes it, badge_sampler.py:36-48.)
+// The round-1 torch composition (two rocBLAS skinny GEMMs + 4 elementwise
+// passes per row chunk) re-streamed the N x chunk fp32 intermediates ~8x;
+// this kernel computes both rank-K grams on f32 MFMA from LDS tiles and
+// writes the N x N output ONCE. 64x64 tile per block, 4 waves (2x2).
+// ---------------------------------------------------------------------------
+
+__launch_bounds__(256)
+__global__ void badge_gram_kernel(const float* __restrict__ Av,
+                                  const float* __restrict__ Ev,
+                                  const float* __restrict__ d,
+                                  float* __restrict__ out,
+                                  long N, int Ka, int Ke) {
+  __shared__ float As[64 * LDP];
+  __shared__ float Bs[64 * LDP];
+  const long r0 = (long)blockIdx.x * 64, n0 = (long)blockIdx.y * 64;
+  const int tid = threadIdx.x;
+  const int wid = tid >> 6, lane = tid & 63;
+  const int wr = wid >> 1, wc = wid & 1;
+  const int l15 = lane & 15, l4 = lane >> 4;
+
+  f32x4 accA[2][2], accE[2][2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      accA[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+      accE[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+    }
+
+  // one gram accumulation: rows from src[r0..], cols from src[n0..]
+  auto gram = [&](const float* src, int K_, f32x4 (*acc)[2]) {
+    for (int k0 = 0; k0 < K_; k0 += 64) {
+      for (int idx = tid; idx < 64 * 64; idx += 256) {
+        const int r = idx >> 6, k = idx & 63;
+        const long gr = r0 + r, gn = n0 + r;
+        const int gk = k0 + k;
+        As[r * LDP + k] = (gr < N && gk < K_) ? src[gr * K_ + gk] : 0.f;
+        Bs[r * LDP + k] = (gn < N && gk < K_) ? src[gn * K_ + gk] : 0.f;
+      }
+      __syncthreads();
+#pragma unroll 4
+      for (int kk = 0; kk < 16; ++kk) {
+        float a[2], b[2];
+#pragma unroll
+        for (int fi = 0; fi < 2; ++fi)
+          a[fi] = As[(wr * 32 + fi * 16 + l15) * LDP + kk * 4 + l4];
+#pragma unroll
+        for (int fj = 0; fj < 2; ++fj)
+          b[fj] = Bs[(wc * 32 + fj * 16 + l15) * LDP + kk * 4 + l4];
+#pragma unroll
+        for (int fi = 0; fi < 2; ++fi)
+#pragma unroll
+          for (int fj = 0; fj < 2; ++fj)
+            acc[fi][fj] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                a[fi], b[fj], acc[fi][fj], 0, 0, 0);
+      }
+      __syncthreads();
+    }
+  };
+  gram(Av, Ka, (f32x4(*)[2])accA);
+  gram(Ev, Ke, (f32x4(*)[2])accE);
+
+#pragma unroll
+  for (int fi = 0; fi < 2; ++fi)
+#pragma unroll
+    for (int fj = 0; fj < 2; ++fj) {
+      const long col = n0 + wc * 32 + fj * 16 + l15;
+      if (col >= N) continue;
+      const float dj = d[col];
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const long row = r0 + wr * 32 + fi * 16 + l4 * 4 + r;
+        if (row < N)
+          out[row * N + col] =
+              d[row] + dj - 2.f * accA[fi][fj][r] * accE[fi][fj][r];
+      }
+    }
+}
+
+extern "C" void al_badge_gram(const float* a, const float* e, const float* d,
+                              float* out, long N, int Ka, int Ke,
+                              hipStream_t stream) {
+  const long g = (N + 63) / 64;
+  hipLaunchKernelGGL(badge_gram_kernel, dim3((unsigned)g, (unsigned)g),
+                     dim3(256), 0, stream, a, e, d, out, N, Ka, Ke);
+}
+
 // column sums of dY (B,C) -> db (C)
 __global__ void colsum_kernel(const float* __restrict__ dy, float* __restrict__ db,
                               int B, int C) {

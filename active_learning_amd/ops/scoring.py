@@ -116,10 +116,16 @@ def badge_pairwise_sqdist(a_vec: torch.Tensor, e_vec: torch.Tensor,
 
     Row-chunked so only ONE N x N fp32 matrix is resident (68 GB at N=130k);
     the two Gram factors exist chunk-at-a-time."""
-    a = a_vec.float()
-    e = e_vec.float()
+    a = a_vec.float().contiguous()
+    e = e_vec.float().contiguous()
     n = a.shape[0]
-    d = (a * a).sum(dim=1) * (e * e).sum(dim=1)  # <g_i, g_i>
+    d = ((a * a).sum(dim=1) * (e * e).sum(dim=1)).contiguous()  # <g_i, g_i>
+    if a.is_cuda and os.environ.get("AL_BADGE_GRAM_DEV", "1") == "1":
+        # fused kernel: both rank-K grams on f32 MFMA + the distance combine,
+        # ONE write of the N x N output (the torch composition below
+        # re-streams ~8 N x chunk fp32 passes per row chunk)
+        from .extension import require_extension
+        return require_extension().badge_gram(a, e, d)
     out = torch.empty((n, n), dtype=torch.float32, device=a.device)
     for i0 in range(0, n, chunk):
         i1 = min(i0 + chunk, n)

@@ -223,6 +223,23 @@ def test_linear_head_gpu(shape):
     assert relerr(bias.grad, br.grad) < 1e-5, f"linear db {shape}"
 
 
+@pytest.mark.parametrize("dims", [(500, 16, 32), (300, 10, 512), (257, 7, 33)])
+def test_badge_gram_kernel(dims):
+    """Fused BADGE Gram-distance kernel vs the torch composition."""
+    import os
+    from active_learning_amd.ops.scoring import badge_pairwise_sqdist
+    n, ka, ke = dims
+    torch.manual_seed(3)
+    a = torch.randn(n, ka, device="cuda")
+    e = torch.randn(n, ke, device="cuda")
+    os.environ["AL_BADGE_GRAM_DEV"] = "0"
+    ref = badge_pairwise_sqdist(a, e)
+    os.environ["AL_BADGE_GRAM_DEV"] = "1"
+    got = badge_pairwise_sqdist(a, e)
+    os.environ.pop("AL_BADGE_GRAM_DEV")
+    assert relerr(got, ref) < 1e-5, f"badge_gram {dims}"
+
+
 def test_kcenter_persistent_kernel():
     """The cooperative persistent k-center kernel must select exactly the
     same points as the torch reference loop (deterministic mode), and produce
