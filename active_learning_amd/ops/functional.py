@@ -195,7 +195,11 @@ class _GradArena:
 
 _grad_arena = _GradArena()
 
-_BNBACK_ON = os.environ.get("AL_BNBACK_FUSE", "1") == "1"
+# Measured NEGATIVE at B=256 (profiles/fused_breakdown.md): the epilogue's
+# dependent global mask/x reads stall the bwd-data kernels (igemm<1,4,1>
+# +57%, stride-2 parity +160%) by more than the removed bn_bwd_reduce pass
+# saves. Kept opt-in for shapes/hardware where the tradeoff differs.
+_BNBACK_ON = os.environ.get("AL_BNBACK_FUSE", "0") == "1"
 
 
 def _claim_bnback(x):
@@ -268,7 +272,11 @@ class Conv2dNHWC(Function):
                         dy, wt_cached(w_c), ctx.stride, ctx.padding,
                         x.shape[1], x.shape[2], mask, xbn, mean, invstd)
                     try:
-                        dx._al_bnsums = (mask, s, sx)
+                        # stamp the tensor version: if autograd later
+                        # accumulates another grad INTO this tensor in place
+                        # (fan-in reuses the buffer), the version bumps and
+                        # the BN backward falls back to its own reduce
+                        dx._al_bnsums = (mask, s, sx, dx._version)
                     except Exception:
                         pass
                 else:
@@ -499,7 +507,8 @@ class BatchNormAct(Function):
             ext = require_extension()
             pre = getattr(dy, "_al_bnsums", None)
             if (pre is not None and not ctx.has_residual and ctx.relu
-                    and pre[0].data_ptr() == mask_or_y.data_ptr()):
+                    and pre[0].data_ptr() == mask_or_y.data_ptr()
+                    and dy._version == pre[3]):
                 # sums were computed in the producing conv's bwd-data
                 # epilogue, and dy arrived PRE-MASKED — skip the reduce pass
                 sum_dy, sum_dy_xhat = pre[1], pre[2]

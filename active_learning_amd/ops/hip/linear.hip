@@ -143,8 +143,12 @@ __global__ void badge_gram_kernel(const float* __restrict__ Av,
   // one gram accumulation: rows from src[r0..], cols from src[n0..]
   auto gram = [&](const float* src, int K_, f32x4 (*acc)[2]) {
     for (int k0 = 0; k0 < K_; k0 += 64) {
+      // only the live K columns are staged and contracted (the pooled BADGE
+      // factors are K=16/32: a fixed 64-deep loop would 4x the work)
+      const int kpad = min(64, ((K_ - k0) + 3) & ~3);
       for (int idx = tid; idx < 64 * 64; idx += 256) {
         const int r = idx >> 6, k = idx & 63;
+        if (k >= kpad) continue;
         const long gr = r0 + r, gn = n0 + r;
         const int gk = k0 + k;
         As[r * LDP + k] = (gr < N && gk < K_) ? src[gr * K_ + gk] : 0.f;
@@ -152,7 +156,7 @@ __global__ void badge_gram_kernel(const float* __restrict__ Av,
       }
       __syncthreads();
 #pragma unroll 4
-      for (int kk = 0; kk < 16; ++kk) {
+      for (int kk = 0; kk < kpad / 4; ++kk) {
         float a[2], b[2];
 #pragma unroll
         for (int fi = 0; fi < 2; ++fi)
