@@ -513,6 +513,13 @@ class BatchNormAct(Function):
                 # epilogue, and dy arrived PRE-MASKED — skip the reduce pass
                 sum_dy, sum_dy_xhat = pre[1], pre[2]
                 premasked = True
+                if os.environ.get("AL_BNBACK_DEBUG") == "1":
+                    rs, rsx = ext.bn_bwd_reduce(dy, x, mask_or_y, mean,
+                                                invstd, False)
+                    e1 = ((pre[1] - rs).norm() / rs.norm().clamp_min(1e-6)).item()
+                    e2 = ((pre[2] - rsx).norm() / rsx.norm().clamp_min(1e-6)).item()
+                    print(f"[bnback] C={x.shape[-1]} rows={x.numel()//x.shape[-1]}"
+                          f" sum_err={e1:.2e} sumx_err={e2:.2e}", flush=True)
             else:
                 # reduce pass: per-channel sums of dy~, dy~*xhat (dy~ = mask*dy)
                 sum_dy, sum_dy_xhat = ext.bn_bwd_reduce(dy, x, mask_or_y, mean,

@@ -29,6 +29,56 @@ from .functional import bump_tick
 logger = logging.getLogger("ActiveLearning")
 
 
+class GraphedInference:
+    """hipGraph-captured no-grad forward for the query/eval loops: the pool
+    pass replays one graph per batch instead of ~200 eager launches
+    (reference: full-pool inference per query, e.g. confidence_sampler.py:
+    27-36). Create one per pass — the folded-BN eval constants are baked at
+    capture, so weights must not change across calls (true within a query
+    or one validation sweep). Odd-sized tail batches fall back to eager."""
+
+    def __init__(self, fn, device, warmup=1):
+        self.fn = fn
+        self.device = device
+        self.warmup = warmup
+        self._calls = 0
+        self._graph = None
+        self._failed = device.type != "cuda"
+        self.x_static = None
+        self.out_static = None
+
+    @torch.no_grad()
+    def __call__(self, x):
+        if self._failed or (self.x_static is not None
+                            and x.shape != self.x_static.shape):
+            return self.fn(x.to(self.device, non_blocking=True))
+        if self._graph is None:
+            if self.x_static is None:
+                self.x_static = torch.empty(x.shape, dtype=x.dtype,
+                                            device=self.device)
+            self.x_static.copy_(x)
+            if self._calls < self.warmup:
+                self._calls += 1
+                return self.fn(self.x_static)
+            try:
+                torch.cuda.synchronize()
+                g = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g):
+                    self.out_static = self.fn(self.x_static)
+                self._graph = g
+            except Exception as e:
+                logger.warning(f"inference capture unavailable ({e!r}); eager")
+                self._failed = True
+                return self.fn(self.x_static)
+        else:
+            self.x_static.copy_(x, non_blocking=True)
+        self._graph.replay()
+        out = self.out_static
+        if isinstance(out, (tuple, list)):
+            return type(out)(o.clone() for o in out)
+        return out.clone()
+
+
 class GraphedTrainStep:
     """Wraps (net, optimizer, criterion) into a capturable step:
     loss = step(x, y). Capture happens lazily after `warmup` eager calls."""

@@ -125,7 +125,12 @@ __global__ void badge_gram_kernel(const float* __restrict__ Av,
                                   long N, int Ka, int Ke) {
   __shared__ float As[64 * LDP];
   __shared__ float Bs[64 * LDP];
-  const long r0 = (long)blockIdx.x * 64, n0 = (long)blockIdx.y * 64;
+  // grid-stride over output tiles: at N=130k there are ~4.1M 64x64 tiles,
+  // and one-workgroup-per-tile is DISPATCH-bound (~2 s of launch overhead
+  // measured); ~16k persistent workgroups loop instead
+  const long gm = (N + 63) / 64;
+  for (long t = blockIdx.x; t < gm * gm; t += gridDim.x) {
+  const long r0 = (t % gm) * 64, n0 = (t / gm) * 64;
   const int tid = threadIdx.x;
   const int wid = tid >> 6, lane = tid & 63;
   const int wr = wid >> 1, wc = wid & 1;
@@ -139,6 +144,7 @@ __global__ void badge_gram_kernel(const float* __restrict__ Av,
       accA[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
       accE[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
     }
+  __syncthreads();  // previous tile's epilogue reads done before restaging
 
   // one gram accumulation: rows from src[r0..], cols from src[n0..]
   auto gram = [&](const float* src, int K_, f32x4 (*acc)[2]) {
@@ -192,14 +198,16 @@ __global__ void badge_gram_kernel(const float* __restrict__ Av,
               d[row] + dj - 2.f * accA[fi][fj][r] * accE[fi][fj][r];
       }
     }
+  }  // tile loop
 }
 
 extern "C" void al_badge_gram(const float* a, const float* e, const float* d,
                               float* out, long N, int Ka, int Ke,
                               hipStream_t stream) {
   const long g = (N + 63) / 64;
-  hipLaunchKernelGGL(badge_gram_kernel, dim3((unsigned)g, (unsigned)g),
-                     dim3(256), 0, stream, a, e, d, out, N, Ka, Ke);
+  const long blocks = min(g * g, (long)16384);
+  hipLaunchKernelGGL(badge_gram_kernel, dim3((unsigned)blocks), dim3(256), 0,
+                     stream, a, e, d, out, N, Ka, Ke);
 }
 
 // column sums of dY (B,C) -> db (C)

@@ -40,14 +40,20 @@ def forward_pool(strategy, idxs, want_embedding=False, use_al_set=True, keep_dev
     net = strategy.net
     net.eval()
     net.to(strategy.device)
+    if want_embedding:
+        fwd = lambda t: net(t, return_features="finalembed")  # noqa: E731
+    else:
+        fwd = net
+    if strategy.device.type == "cuda" and os.environ.get("AL_EVAL_GRAPH", "1") == "1":
+        from ..ops.graph import GraphedInference
+        fwd = GraphedInference(fwd, strategy.device)
     logits_l, emb_l, y_l = [], [], []
     for x, y, _ in loader:
-        x = x.to(strategy.device, non_blocking=True)
         if want_embedding:
-            out, emb = net(x, return_features="finalembed")
+            out, emb = fwd(x)
             emb_l.append(emb.float() if keep_device else emb.float().cpu())
         else:
-            out = net(x)
+            out = fwd(x)
         logits_l.append(out.float() if keep_device else out.float().cpu())
         y_l.append(y)
     logits = torch.cat(logits_l, dim=0)

@@ -47,11 +47,17 @@ def accuracy(dataloader, net, top_k=(1, 5), **kwargs):
     count_byclass = torch.zeros(num_classes)
     total = 0
 
+    import os as _os
+    fwd = net
+    if device.type == "cuda" and _os.environ.get("AL_EVAL_GRAPH", "1") == "1":
+        # validation/test sweeps replay one captured graph per batch (the
+        # weights are fixed for the duration of one evaluate() call)
+        from ..ops.graph import GraphedInference
+        fwd = GraphedInference(net, device)
     with torch.no_grad():
         for batch_idx, (inputs, targets, _idxs) in enumerate(dataloader):
-            inputs = inputs.to(device, non_blocking=True)
             targets = targets.to(device, non_blocking=True)
-            output = net(inputs)
+            output = fwd(inputs)
             _, pred = torch.topk(output.float(), max_k, dim=1, largest=True, sorted=True)
             hit = pred == targets[:, None]
             for k in top_k:

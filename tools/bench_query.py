@@ -40,14 +40,18 @@ def timed(label, fn):
 
 @torch.no_grad()
 def pool_inference(net, n_images, batch, img, want_embedding=True):
-    """Forward the pool through the net, keeping logits/embeddings in HBM."""
+    """Forward the pool through the net, keeping logits/embeddings in HBM.
+    The fused-eval forward is hipGraph-captured and replayed per batch
+    (ops/graph.py::GraphedInference)."""
+    from active_learning_amd.ops.graph import GraphedInference
+    fwd = GraphedInference(lambda t: net(t, return_features="finalembed"),
+                           torch.device("cuda"))
     logits_l, emb_l = [], []
     done = 0
     x = torch.randn(batch, 3, img, img, device="cuda")
     while done < n_images:
         b = min(batch, n_images - done)
-        xb = x[:b]
-        out, emb = net(xb, return_features="finalembed")
+        out, emb = fwd(x[:b])
         logits_l.append(out.float())
         emb_l.append(emb.float())
         done += b
