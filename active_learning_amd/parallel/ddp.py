@@ -182,13 +182,25 @@ class BucketedDDP(nn.Module):
     def _rebuild_from_order(self):
         """Re-chunk buckets to the order backward actually produced grads, so
         from the second iteration on each bucket fills (and its all-reduce
-        launches) as early as possible."""
+        launches) as early as possible.
+
+        The observed order can differ ACROSS RANKS (autograd engine thread
+        timing), and mismatched bucket boundaries would make corresponding
+        all-reduces carry different parameters — so rank 0's order is
+        broadcast and every rank rebuilds identically (torch DDP does the
+        same for its bucket rebuild)."""
         self._order_final = True
         if not self._ready_order:
             return
-        seen = {id(p) for p in self._ready_order}
-        ordered = list(self._ready_order) + [p for p in self._params
-                                             if id(p) not in seen]
+        idx_of = {id(p): i for i, p in enumerate(self._params)}
+        order = [idx_of[id(p)] for p in self._ready_order]
+        seen = set(order)
+        order += [i for i in range(len(self._params)) if i not in seen]
+        device = self._params[0].device if self._params else torch.device("cpu")
+        t = torch.tensor(order, dtype=torch.int64, device=device)
+        dist.broadcast(t, src=0, group=self.pg)
+        order = t.tolist()
+        ordered = [self._params[i] for i in order]
         # preserve this step's reduced grads: they live in the OLD buckets'
         # storage via p.grad views, which survive the rebuild untouched.
         self._build_buckets(ordered, self.bucket_cap_mb)
