@@ -120,10 +120,13 @@ def badge_pairwise_sqdist(a_vec: torch.Tensor, e_vec: torch.Tensor,
     e = e_vec.float().contiguous()
     n = a.shape[0]
     d = ((a * a).sum(dim=1) * (e * e).sum(dim=1)).contiguous()  # <g_i, g_i>
-    if a.is_cuda and os.environ.get("AL_BADGE_GRAM_DEV", "1") == "1":
-        # fused kernel: both rank-K grams on f32 MFMA + the distance combine,
-        # ONE write of the N x N output (the torch composition below
-        # re-streams ~8 N x chunk fp32 passes per row chunk)
+    if (a.is_cuda and a.shape[1] + e.shape[1] <= 128
+            and os.environ.get("AL_BADGE_GRAM_DEV", "1") == "1"):
+        # fused kernel for the POOLED factors (16 x 32, the reference's
+        # ImageNet-scale BADGE): both rank-K grams on f32 MFMA + the distance
+        # combine, ONE write of the N x N output. Wide unpooled factors
+        # (plain CIFAR BADGE: K = C + M) stay on rocBLAS, which wins at
+        # large contraction depth.
         from .extension import require_extension
         return require_extension().badge_gram(a, e, d)
     out = torch.empty((n, n), dtype=torch.float32, device=a.device)

@@ -190,11 +190,15 @@ def test_resnet_bnback_fusion_matches_disabled():
             AFn._BNBACK_ON = old
 
     g_off = run(False)
+    g_off2 = run(False)  # calibrate: split-K wgrad + stats atomics reorder
     g_on = run(True)
     assert g_on.keys() == g_off.keys()
     for n_ in g_off:
+        base = relerr(g_off2[n_], g_off[n_])  # nondeterminism floor
         err = relerr(g_on[n_], g_off[n_])
-        assert err < 5e-3, f"bnback fusion changed grad of {n_}: {err}"
+        tol = max(5e-3, 10 * base)
+        assert err < tol, (f"bnback fusion changed grad of {n_}: {err} "
+                           f"(off-vs-off floor {base})")
 
 
 @pytest.mark.parametrize("shape", [(16, 512, 10), (128, 2048, 1000),
