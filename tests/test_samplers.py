@@ -234,6 +234,33 @@ def test_sharded_forward_pool_matches_single(monkeypatch, tmp_path):
     assert torch.allclose(logits, ref_logits, atol=1e-6)
     assert torch.allclose(emb, ref_emb, atol=1e-6)
 
+    # the worker pool is PERSISTENT: a second query reuses the same worker
+    # processes (no per-query spawn) and still matches
+    from active_learning_amd.strategies.common import _query_pool
+    pids = [p.pid for p in _query_pool.procs]
+    assert pids, "worker pool not alive after first query"
+    logits2, emb2, y2 = forward_pool(s, idxs, want_embedding=True)
+    assert [p.pid for p in _query_pool.procs] == pids, "workers were respawned"
+    assert torch.allclose(logits2, ref_logits, atol=1e-6)
+
+
+def test_graphed_inference_cpu_fallback():
+    """GraphedInference on CPU (or odd shapes) must transparently run the
+    wrapped fn eagerly."""
+    import torch
+    from active_learning_amd.ops.graph import GraphedInference
+    calls = []
+
+    def fn(t):
+        calls.append(tuple(t.shape))
+        return t * 2
+
+    gi = GraphedInference(fn, torch.device("cpu"))
+    x = torch.ones(4, 3)
+    assert torch.allclose(gi(x), x * 2)
+    assert torch.allclose(gi(torch.ones(2, 3)), torch.ones(2, 3) * 2)
+    assert calls == [(4, 3), (2, 3)]
+
 
 def test_shard_gating(monkeypatch, tmp_path):
     import numpy as np
