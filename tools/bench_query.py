@@ -80,7 +80,7 @@ def config4_badge(n_labeled=50_000, n_unlabeled=80_000, budget=10_000):
     res = {}
     (out, res["inference_s"]) = timed(
         f"imagenet pool inference ({n//1000}k)",
-        lambda: pool_inference(net, n, 256, 224))
+        lambda: pool_inference(net, n, 512, 224))  # 512: measured ~17% faster than 256
     logits, emb = out
     # the reference's ImageNet-scale BADGE pools the gradient embedding to
     # 16 x 32 dims (PartitionedBADGESampler -> use_adaptive_pool=True,
