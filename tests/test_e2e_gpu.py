@@ -94,3 +94,38 @@ def test_freeze_feature_round_gpu(tmp_path):
         "--freeze_feature"])
     s = main(args)
     assert s.idxs_lb.sum() == 5
+
+
+def test_pretrained_linear_eval_round_gpu(tmp_path):
+    """The headline reference flow (ssp_linear_evaluation): SSL checkpoint
+    loaded with key surgery each round, frozen backbone, net.eval()-while-
+    training BN semantics — under the hipGraph-captured trainer (frozen-stats
+    BN inside capture)."""
+    import torch
+    import numpy as np
+    import helpers
+    from active_learning_amd.models import get_networks
+    from active_learning_amd.strategies import RandomSampler
+
+    # fake SSL checkpoint: a differently-seeded model's state dict
+    torch.manual_seed(77)
+    donor = get_networks("synthetic_cifar10", "SSLResNet18")
+    ck = str(tmp_path / "ssl.pth")
+    torch.save({"state_dict": donor.state_dict()}, ck)
+
+    s = helpers.make_strategy(RandomSampler, ckpt_path=str(tmp_path),
+                              freeze_feature=True)
+    s.net.freeze_feature = True  # detach embedding (resnet_simclr.py:36-37)
+    s.train_args["init_pretrained_ckpt_path"] = ck
+    s.train_args["skip_key"] = "linear"  # reinit head survives (surgery path)
+    s.n_epoch = 3
+    s.update(np.arange(40), 40)
+    s.init_network_weights()
+    # the backbone must carry the donor's weights after surgery
+    got = s.net.encoder.conv1.weight.detach()
+    want = donor.encoder.conv1.weight.detach()
+    assert torch.allclose(got, want), "SSL checkpoint not loaded"
+    s.parallel_train_fn(0)
+    torch.cuda.synchronize()
+    import os
+    assert os.path.exists(s.generate_weight_paths()["best_ckpt"])
