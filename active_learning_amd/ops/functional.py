@@ -195,6 +195,56 @@ class _GradArena:
 
 _grad_arena = _GradArena()
 
+
+class _StatsArena:
+    """Flat fp32 slab for the conv-epilogue BN-stat accumulators (2K floats
+    per conv): ONE zero fill per forward pass instead of one per conv (the
+    per-conv torch.zeros showed as ~53 FillFunctor launches/step). A pass
+    boundary is detected when a key repeats (the same conv runs again)."""
+
+    def __init__(self):
+        self.enabled = os.environ.get("AL_STATS_ARENA", "1") == "1"
+        self.views = {}
+        self.shapes = {}
+        self.pending = []
+        self.flat = None
+        self.seen = set()
+
+    def take(self, key, n, device):
+        if not self.enabled:
+            return None
+        if key in self.seen:
+            # new forward pass: previous stats were consumed by bn_finalize
+            if self.pending:
+                self._rebuild(device)
+            elif self.flat is not None:
+                self.flat.zero_()
+            self.seen = set()
+        self.seen.add(key)
+        v = self.views.get(key)
+        if v is not None:
+            return v
+        if key not in [k for k, _ in self.pending]:
+            self.pending.append((key, n))
+        return None
+
+    def _rebuild(self, device):
+        live = self.seen | {k for k, _ in self.pending}
+        self.shapes = {k: s for k, s in self.shapes.items() if k in live}
+        for key, n in self.pending:
+            self.shapes[key] = n
+        self.pending = []
+        total = sum(self.shapes.values())
+        self.flat = torch.zeros(total, dtype=torch.float32, device=device)
+        off = 0
+        self.views = {}
+        for key, n in self.shapes.items():
+            self.views[key] = self.flat.narrow(0, off, n)
+            off += n
+
+
+_stats_arena = _StatsArena()
+
 # Measured NEGATIVE at B=256 (profiles/fused_breakdown.md): the epilogue's
 # dependent global mask/x reads stall the bwd-data kernels (igemm<1,4,1>
 # +57%, stride-2 parity +160%) by more than the removed bn_bwd_reduce pass
@@ -331,15 +381,19 @@ class Conv2dNHWCStats(Function):
             ext = require_extension()
             _grad_arena.mark_step()
             K, R, S, C = w_c.shape
+            buf = _stats_arena.take((w_c.data_ptr(), K), 2 * K, x.device)
+            empty = x.new_empty(0, dtype=torch.float32)
             if _igemm_eligible(C, R * S * C):
                 ctx.apack = None
-                y, s, ss = ext.conv2d_fwd_stats(x, w_c, stride, padding)
+                y, s, ss = ext.conv2d_fwd_stats(x, w_c, stride, padding,
+                                                buf if buf is not None else empty)
             else:
                 K, R, S, C = w_c.shape
                 kdpad = ((R * S * C + 63) // 64) * 64
                 apack = ext.im2col_pack(x, R, S, stride, padding, kdpad)
                 ctx.apack = apack
-                y, s, ss = ext.conv2d_fwd_stats(apack, _wpad_cached(w_c, kdpad), 1, 0)
+                y, s, ss = ext.conv2d_fwd_stats(apack, _wpad_cached(w_c, kdpad), 1, 0,
+                                                buf if buf is not None else empty)
         else:
             xc, wc = _cpu_conv_args(x, w_c)
             yt = F.conv2d(xc.float(), wc.float(), stride=stride, padding=padding)

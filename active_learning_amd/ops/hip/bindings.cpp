@@ -120,9 +120,11 @@ Tensor conv2d_fwd(const Tensor& x, const Tensor& w, long stride, long pad) {
 }
 
 std::vector<Tensor> conv2d_fwd_stats(const Tensor& x, const Tensor& w, long stride,
-                                     long pad) {
+                                     long pad, const Tensor& both_in) {
   // training-path fusion: conv + per-column sum/sumsq of the output in the
-  // same kernel (feeds the following BatchNorm's batch statistics)
+  // same kernel (feeds the following BatchNorm's batch statistics).
+  // both_in: optional PRE-ZEROED 2K fp32 slot from the python stats arena
+  // (one flat zero fill per step instead of one per conv); empty -> allocate.
   check_bf16_contig(x, "x");
   check_bf16_contig(w, "w");
   const int N = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
@@ -131,9 +133,8 @@ std::vector<Tensor> conv2d_fwd_stats(const Tensor& x, const Tensor& w, long stri
   const int P = out_dim(H, R, stride, pad), Q = out_dim(W, S, stride, pad);
   auto y = torch::empty({N, P, Q, K}, x.options());
   auto opts = x.options().dtype(torch::kFloat32);
-  // one zeroed allocation for both accumulators: halves the per-conv fill
-  // launches (~300 small FillFunctor launches/step showed up in the profile)
-  auto both = torch::zeros({2L * K}, opts);
+  Tensor both = both_in.numel() ? both_in : torch::zeros({2L * K}, opts);
+  TORCH_CHECK(both.numel() >= 2L * K && both.scalar_type() == torch::kFloat32);
   auto sum = both.narrow(0, 0, K);
   auto sumsq = both.narrow(0, K, K);
   al_conv2d_mm(0, x.data_ptr(), w.data_ptr(), y.data_ptr(),

@@ -508,7 +508,8 @@ def test_conv_stats_fusion_matches(shape):
     n, h, w_, c, k, r, stride, pad = shape
     x = torch.randn(n, h, w_, c, device="cuda").to(torch.bfloat16)
     w = (torch.randn(k, r, r, c, device="cuda") * 0.05).to(torch.bfloat16)
-    y, s, ss = EXT.conv2d_fwd_stats(x, w, stride, pad)
+    y, s, ss = EXT.conv2d_fwd_stats(x, w, stride, pad,
+                                    torch.empty(0, device="cuda"))
     y2 = EXT.conv2d_fwd(x, w, stride, pad)
     assert torch.equal(y, y2)
     yf = y.float()
