@@ -107,11 +107,14 @@ def test_pretrained_linear_eval_round_gpu(tmp_path):
     from active_learning_amd.models import get_networks
     from active_learning_amd.strategies import RandomSampler
 
-    # fake SSL checkpoint: a differently-seeded model's state dict
+    # fake SSL checkpoint: a differently-seeded model's state dict (saved
+    # WITH the native-layout marker — an unmarked dict is treated as an
+    # external OIHW checkpoint and gets the KRSC permute)
+    from active_learning_amd.utils.checkpoint import state_dict_with_marker
     torch.manual_seed(77)
     donor = get_networks("synthetic_cifar10", "SSLResNet18")
     ck = str(tmp_path / "ssl.pth")
-    torch.save({"state_dict": donor.state_dict()}, ck)
+    torch.save({"state_dict": state_dict_with_marker(donor)}, ck)
 
     s = helpers.make_strategy(RandomSampler, ckpt_path=str(tmp_path),
                               freeze_feature=True)
