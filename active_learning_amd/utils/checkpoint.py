@@ -62,6 +62,13 @@ def load_pretrained_weights(net, path, replace_key=None, skip_key=None, required
                 return name.replace(k, v)
         return name
 
+    # accept bare strings for the key filters (a string would otherwise be
+    # iterated per-character and match almost every key)
+    if isinstance(skip_key, str):
+        skip_key = [skip_key]
+    if isinstance(required_key, str):
+        required_key = [required_key]
+
     def skip(name):
         if skip_key and any(s in name for s in skip_key):
             return True

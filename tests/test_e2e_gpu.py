@@ -120,7 +120,7 @@ def test_pretrained_linear_eval_round_gpu(tmp_path):
                               freeze_feature=True)
     s.net.freeze_feature = True  # detach embedding (resnet_simclr.py:36-37)
     s.train_args["init_pretrained_ckpt_path"] = ck
-    s.train_args["skip_key"] = "linear"  # reinit head survives (surgery path)
+    s.train_args["skip_key"] = ["linear"]  # reinit head survives (surgery)
     s.n_epoch = 3
     s.update(np.arange(40), 40)
     s.init_network_weights()
