@@ -75,6 +75,8 @@ void al_scatter_s2(const void* tmp, void* dx, int N, int H, int W, int C, int P,
                    int Q, int stride, void* stream);
 void al_badge_gram(const float* a, const float* e, const float* d, float* out,
                    long N, int Ka, int Ke, void* stream);
+int al_pairwise_sqdist(const void* f, const float* sq, float* out,
+                       const void* zero, long N, int M, void* stream);
 int al_linear_needs_zero(int M_, int N_, int K_);
 void al_linear_fwd(const float* x, const float* w, const float* bias, float* out,
                    int B, int M, int C, void* stream);
@@ -536,6 +538,20 @@ Tensor scatter_s2(const Tensor& tmp, int64_t H, int64_t W, int64_t stride) {
   return dx;
 }
 
+Tensor pairwise_sqdist_dev(const Tensor& f, const Tensor& sq) {
+  check_bf16_contig(f, "features");
+  TORCH_CHECK(sq.scalar_type() == torch::kFloat32 && sq.is_contiguous());
+  const long N = f.size(0);
+  const int M = (int)f.size(1);
+  TORCH_CHECK(sq.numel() == N && M % 64 == 0);
+  Tensor out = torch::empty({N, N}, sq.options());
+  int rc = al_pairwise_sqdist(f.data_ptr(), sq.data_ptr<float>(),
+                              out.data_ptr<float>(), zero_page(f).data_ptr(),
+                              N, M, cur_stream());
+  TORCH_CHECK(rc == 0, "pairwise_sqdist: M % 64 != 0");
+  return out;
+}
+
 Tensor badge_gram(const Tensor& a, const Tensor& e, const Tensor& d) {
   TORCH_CHECK(a.is_cuda() && a.scalar_type() == torch::kFloat32 && a.is_contiguous());
   TORCH_CHECK(e.scalar_type() == torch::kFloat32 && e.is_contiguous());
@@ -613,6 +629,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("kcenter_greedy_dev", &kcenter_greedy_dev);
   m.def("scatter_s2", &scatter_s2);
   m.def("badge_gram", &badge_gram);
+  m.def("pairwise_sqdist_dev", &pairwise_sqdist_dev);
   m.def("linear_fwd", &linear_fwd);
   m.def("linear_bwd", &linear_bwd);
   m.def("adam_step", &adam_step);

@@ -210,6 +210,123 @@ extern "C" void al_badge_gram(const float* a, const float* e, const float* d,
                      stream, a, e, d, out, N, Ka, Ke);
 }
 
+// ---------------------------------------------------------------------------
+// Pairwise squared distances on bf16 MFMA:
+//   out[i][j] = sq[i] + sq[j] - 2 * <f_i, f_j>     (out fp32, f bf16)
+// The coreset N x N matrix (68 GB at N=130k, resident in HBM). The fp32
+// rocBLAS composition runs at the fp32 MFMA rate (~115 TF); the dot products
+// here run on mfma_f32_16x16x32_bf16 with fp32 accumulation and fp32
+// norms/out (PARITY.md: the embeddings themselves are bf16-computed).
+// 128x128x64 tile, 4 waves (2x2), 2-barrier double-buffered glds staging
+// with the igemm XOR source swizzle; grid-strided over output tiles.
+// ---------------------------------------------------------------------------
+
+typedef __bf16 pw_bf16x8 __attribute__((ext_vector_type(8)));
+
+__launch_bounds__(256)
+__global__ void pairwise_kernel(const bf16* __restrict__ F,
+                                const float* __restrict__ sq,
+                                float* __restrict__ out,
+                                const bf16* __restrict__ zero,
+                                long N, int M) {
+  extern __shared__ __attribute__((aligned(16))) char pw_smem[];
+  bf16* As = (bf16*)pw_smem;         // [2][128][64]
+  bf16* Bs = As + 2 * 128 * 64;
+  const int tid = threadIdx.x;
+  const int wid = tid >> 6, lane = tid & 63;
+  const int wr = wid >> 1, wc = wid & 1;
+  const int l15 = lane & 15, l4 = lane >> 4;
+  const int KT = M / 64;
+  const long gm = (N + 127) / 128;
+
+  for (long t = blockIdx.x; t < gm * gm; t += gridDim.x) {
+    const long r0 = (t % gm) * 128, n0 = (t / gm) * 128;
+    f32x4 acc[4][4];
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+    __syncthreads();  // previous tile's reads complete before restaging
+
+    auto stage = [&](int buf, int kt) {
+      bf16* ab = As + buf * 128 * 64;
+      bf16* bb = Bs + buf * 128 * 64;
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {  // 4 A-chunks + 4 B-chunks per thread
+        const int tt = (wid * 4 + i) * 64 + lane;
+        const int row = tt >> 3, u = tt & 7;
+        const int usw = u ^ (row & 7);
+        const long ra = r0 + row, rb = n0 + row;
+        const bf16* sa = (ra < N) ? F + ra * M + (long)kt * 64 + usw * 8 : zero;
+        const bf16* sb = (rb < N) ? F + rb * M + (long)kt * 64 + usw * 8 : zero;
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) void*)sa,
+            (__attribute__((address_space(3))) void*)(ab + (wid * 4 + i) * 512),
+            16, 0, 0);
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) void*)sb,
+            (__attribute__((address_space(3))) void*)(bb + (wid * 4 + i) * 512),
+            16, 0, 0);
+      }
+    };
+
+    stage(0, 0);
+    int buf = 0;
+    for (int kt = 0; kt < KT; ++kt) {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __syncthreads();
+      if (kt + 1 < KT) stage(buf ^ 1, kt + 1);
+      const bf16* ab = As + buf * 128 * 64;
+      const bf16* bb = Bs + buf * 128 * 64;
+#pragma unroll
+      for (int kc = 0; kc < 2; ++kc) {
+        pw_bf16x8 af[4], bf[4];
+#pragma unroll
+        for (int f = 0; f < 4; ++f) {
+          const int ar = wr * 64 + f * 16 + l15;
+          af[f] = *(const pw_bf16x8*)(ab + ar * 64 + (((kc * 4 + l4) ^ (ar & 7)) * 8));
+          const int br = wc * 64 + f * 16 + l15;
+          bf[f] = *(const pw_bf16x8*)(bb + br * 64 + (((kc * 4 + l4) ^ (br & 7)) * 8));
+        }
+#pragma unroll
+        for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+          for (int ni = 0; ni < 4; ++ni)
+            acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
+      }
+      buf ^= 1;
+    }
+
+#pragma unroll
+    for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        const long col = n0 + wc * 64 + ni * 16 + l15;
+        if (col >= N) continue;
+        const float sc = sq[col];
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const long row = r0 + wr * 64 + mi * 16 + l4 * 4 + r;
+          if (row < N)
+            out[row * N + col] = sq[row] + sc - 2.f * acc[mi][ni][r];
+        }
+      }
+  }
+}
+
+extern "C" int al_pairwise_sqdist(const void* f, const float* sq, float* out,
+                                  const void* zero, long N, int M,
+                                  hipStream_t stream) {
+  if (M % 64 != 0) return -1;
+  const long gm = (N + 127) / 128;
+  const long blocks = min(gm * gm, (long)8192);
+  const size_t lds = 4 * 128 * 64 * sizeof(bf16);  // 64 KiB (2 bufs x A+B)
+  hipLaunchKernelGGL(pairwise_kernel, dim3((unsigned)blocks), dim3(256), lds,
+                     stream, (const bf16*)f, sq, out, (const bf16*)zero, N, M);
+  return 0;
+}
+
 // column sums of dY (B,C) -> db (C)
 __global__ void colsum_kernel(const float* __restrict__ dy, float* __restrict__ db,
                               int B, int C) {

@@ -95,11 +95,24 @@ def mase_margins(logits: torch.Tensor, weight: torch.Tensor):
 # --------------------------------------------------------------------------- #
 
 def pairwise_sqdist(features: torch.Tensor, chunk=8192) -> torch.Tensor:
-    """||x_i - x_j||^2 as norms + GEMM (coreset_sampler.py:59-64), chunked so
-    the N x M read stays streaming while the N x N output is materialized
-    (fp32; N=130k -> 68 GB, resident in the 288 GB HBM)."""
+    """||x_i - x_j||^2 as norms + dot products (coreset_sampler.py:59-64);
+    the N x N fp32 output is materialized (N=130k -> 68 GB, resident in the
+    288 GB HBM).
+
+    GPU default: the bf16-MFMA pairwise kernel (linear.hip) — dots on
+    mfma_f32_16x16x32_bf16 with fp32 accumulation and fp32 norms/output.
+    The embeddings are bf16-computed to begin with (PARITY.md);
+    AL_PAIRWISE_BF16=0 falls back to the exact-fp32 rocBLAS composition."""
     f = features.float()
     n = f.shape[0]
+    if (f.is_cuda and f.shape[1] % 64 == 0
+            and os.environ.get("AL_PAIRWISE_BF16", "1") == "1"):
+        from .extension import require_extension
+        fb = f.to(torch.bfloat16).contiguous()
+        # norms from the SAME rounded values the dots use: the result is a
+        # true distance matrix of the bf16-rounded points (near-zero diag)
+        sqb = (fb.float() ** 2).sum(dim=1).contiguous()
+        return require_extension().pairwise_sqdist_dev(fb, sqb)
     sq = (f * f).sum(dim=1)
     out = torch.empty((n, n), dtype=torch.float32, device=f.device)
     for i0 in range(0, n, chunk):

@@ -244,6 +244,26 @@ def test_badge_gram_kernel(dims):
     assert relerr(got, ref) < 1e-5, f"badge_gram {dims}"
 
 
+@pytest.mark.parametrize("nm", [(1000, 2048), (777, 256), (130, 64)])
+def test_pairwise_bf16_kernel(nm):
+    """bf16-MFMA pairwise distances vs the exact fp32 composition. Tolerance
+    reflects the bf16 dot-product quantization (inputs are bf16-computed
+    embeddings in the real pipeline)."""
+    import os
+    from active_learning_amd.ops.scoring import pairwise_sqdist
+    n, m = nm
+    torch.manual_seed(6)
+    f = torch.randn(n, m, device="cuda")
+    os.environ["AL_PAIRWISE_BF16"] = "0"
+    ref = pairwise_sqdist(f)
+    os.environ["AL_PAIRWISE_BF16"] = "1"
+    got = pairwise_sqdist(f)
+    os.environ.pop("AL_PAIRWISE_BF16")
+    assert relerr(got, ref) < 5e-3, f"pairwise bf16 {nm}: {relerr(got, ref)}"
+    # diagonal must be ~0 (sq[i] + sq[i] - 2<f_i,f_i> with fp32 norms)
+    assert got.diagonal().abs().max().item() < 0.05 * ref.max().item()
+
+
 def test_kcenter_persistent_kernel():
     """The cooperative persistent k-center kernel must select exactly the
     same points as the torch reference loop (deterministic mode), and produce

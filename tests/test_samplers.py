@@ -244,6 +244,39 @@ def test_sharded_forward_pool_matches_single(monkeypatch, tmp_path):
     assert torch.allclose(logits2, ref_logits, atol=1e-6)
 
 
+def test_base_sampler_matches_reference_sequential():
+    """The vectorized BASE selection (score-matrix scatter + per-class masked
+    top-k) must pick exactly what the reference's sequential per-class loop
+    picks (base_sampler.py:21-35)."""
+    import torch
+    from active_learning_amd.strategies import BASESampler
+    from helpers import make_strategy
+
+    s = make_strategy(BASESampler)
+    s.update(np.arange(10), 10)
+    budget = 17
+    pool = s.available_query_idxs(boolean=False, shuffle=False)
+    min_m, per_class, pred, _ = s.compute_margins(pool)
+
+    # literal transcription of the reference's loop
+    labeled_idxs = []
+    C = s.num_classes
+    for c in range(C):
+        take = budget // C + int(c < budget % C)
+        if take == 0:
+            continue
+        dist_c = torch.where(pred == c, min_m, per_class[:, c]).clone()
+        if labeled_idxs:
+            dist_c[torch.tensor(labeled_idxs)] = float("inf")
+        order = torch.sort(dist_c, descending=False).indices
+        labeled_idxs += order[:take].tolist()
+    expected = sorted(np.asarray(pool)[labeled_idxs].tolist())
+
+    got, n = s.query(budget)
+    assert n == budget
+    assert sorted(got) == expected
+
+
 def test_graphed_inference_cpu_fallback():
     """GraphedInference on CPU (or odd shapes) must transparently run the
     wrapped fn eagerly."""
