@@ -764,7 +764,9 @@ __global__ void wgrad4_kernel(const bf16* __restrict__ dy,
 
   for (int g = 0; g < KT; ++g) {
     const int buf = g & 1;
+    // drain + barrier: cross-wave retirement guarantee (see wgrad5_kernel)
     asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
 #pragma unroll
     for (int ph = 0; ph < 4; ++ph) {
       const int qm = ph >> 1, qn = ph & 1;
@@ -969,7 +971,14 @@ __global__ void wgrad5_kernel(const bf16* __restrict__ dy,
 
   for (int g = 0; g < KT; ++g) {
     const int buf = g & 1;
+    // cross-wave handshake: EVERY wave drains to <=2 outstanding loads and
+    // THEN barriers, so all waves' contributions to this group's slots are
+    // retired before any wave reads them. Without the barrier the trailing
+    // B refill (issued at the END of the previous group) has only ~1 phase
+    // of slack and a fast wave reads a half-landed slot (race screen caught
+    // intermittent nan/inf at KT=11 shapes).
     asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
 #pragma unroll
     for (int qm = 0; qm < 2; ++qm) {
       const bf16* ah = As + (buf * 2 + qm) * HALF;
