@@ -289,6 +289,16 @@ def test_kcenter_persistent_kernel():
     assert len(set(sel)) == budget, "duplicate selection in k-means++ mode"
     assert not labeled[torch.tensor(sel)].any().item(), "selected labeled point"
 
+    # determinism under a fixed RNG stream: the cooperative kernel must give
+    # identical selections for identical uniforms (a mismatch would indicate
+    # a scheduling race in its 3-grid-sync iteration)
+    for trial in range(3):
+        torch.manual_seed(123)
+        a = kcenter_greedy(dist, labeled, budget, randomize=True)
+        torch.manual_seed(123)
+        b = kcenter_greedy(dist, labeled, budget, randomize=True)
+        assert a == b, f"k-means++ kernel nondeterministic (trial {trial})"
+
 
 # --------------------------------------------------------------------------- #
 # bn / pool through the autograd Functions (GPU path vs CPU path)
