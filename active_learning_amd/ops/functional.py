@@ -94,38 +94,48 @@ def _igemm_eligible(C, KD):
     return C % 8 == 0 and KD % 64 == 0
 
 
-def _wpad_cached(w_c, kdpad):
-    """Zero-padded (K,1,1,kdpad) copy of the stem weight, cached on the bf16
-    weight (rebuilt only when the bf16 copy is recreated)."""
+def _stem_pads(R, S, C):
+    """Row-padded im2col geometry: each filter row (S*C values) padded to an
+    8-aligned rowpad so every 16B chunk of the packed matrix lies within one
+    row and the pack kernel's reads are lane-contiguous."""
+    rowpad = ((S * C + 7) // 8) * 8
+    kdpad = ((R * rowpad + 63) // 64) * 64
+    return rowpad, kdpad
+
+
+def _wpad_cached(w_c, kdpad, rowpad):
+    """Zero-padded (K,1,1,kdpad) copy of the stem weight in the row-padded
+    im2col layout, cached on the bf16 weight (rebuilt only when the bf16
+    copy is recreated)."""
     wpad = getattr(w_c, "_al_wpad", None)
-    K = w_c.shape[0]
-    KD = w_c.numel() // K
+    K, R, S, C = w_c.shape
     if wpad is None:
         wpad = torch.zeros(K, 1, 1, kdpad, dtype=w_c.dtype, device=w_c.device)
         try:
             w_c._al_wpad = wpad
         except Exception:
             pass
-    wpad.view(K, kdpad)[:, :KD] = w_c.reshape(K, KD)
+    wpad.view(K, kdpad)[:, :R * rowpad].view(K, R, rowpad)[:, :, :S * C] = \
+        w_c.reshape(K, R, S * C)
     return wpad
 
 
 def _gpu_conv_fwd_packed(ext, x, w_c, stride, padding):
-    """Stem path: A = im2col(x) zero-padded to kdpad cols, conv as 1x1 igemm.
+    """Stem path: A = im2col(x) in the row-padded layout, conv as 1x1 igemm.
     Returns (y, apack); apack is reused by the backward wgrad."""
     K, R, S, C = w_c.shape
-    KD = R * S * C
-    kdpad = ((KD + 63) // 64) * 64
-    apack = ext.im2col_pack(x, R, S, stride, padding, kdpad)
-    return ext.conv2d_fwd(apack, _wpad_cached(w_c, kdpad), 1, 0), apack
+    rowpad, kdpad = _stem_pads(R, S, C)
+    apack = ext.im2col_pack(x, R, S, stride, padding, kdpad, rowpad)
+    return ext.conv2d_fwd(apack, _wpad_cached(w_c, kdpad, rowpad), 1, 0), apack
 
 
 def _gpu_conv_wgrad_packed(ext, dy, apack, w_shape):
     K, R, S, C = w_shape
-    KD = R * S * C
-    kdpad = apack.shape[-1]
+    rowpad, kdpad = _stem_pads(R, S, C)
+    assert kdpad == apack.shape[-1]
     dwpad = ext.conv2d_bwd_weight(dy, apack, 1, 1, 1, 0)  # (K,1,1,kdpad) fp32
-    return dwpad.view(K, kdpad)[:, :KD].reshape(K, R, S, C).contiguous()
+    return (dwpad.view(K, kdpad)[:, :R * rowpad].view(K, R, rowpad)
+            [:, :, :S * C].reshape(K, R, S, C).contiguous())
 
 
 class _GradArena:
@@ -389,10 +399,12 @@ class Conv2dNHWCStats(Function):
                                                 buf if buf is not None else empty)
             else:
                 K, R, S, C = w_c.shape
-                kdpad = ((R * S * C + 63) // 64) * 64
-                apack = ext.im2col_pack(x, R, S, stride, padding, kdpad)
+                rowpad, kdpad = _stem_pads(R, S, C)
+                apack = ext.im2col_pack(x, R, S, stride, padding, kdpad, rowpad)
                 ctx.apack = apack
-                y, s, ss = ext.conv2d_fwd_stats(apack, _wpad_cached(w_c, kdpad), 1, 0,
+                y, s, ss = ext.conv2d_fwd_stats(apack,
+                                                _wpad_cached(w_c, kdpad, rowpad),
+                                                1, 0,
                                                 buf if buf is not None else empty)
         else:
             xc, wc = _cpu_conv_args(x, w_c)

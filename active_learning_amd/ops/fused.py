@@ -16,7 +16,7 @@ from .extension import extension_available, load_extension
 
 _DISABLED = os.environ.get("AL_AMD_DISABLE_FUSED_EVAL") == "1"
 _STATS_DISABLED = os.environ.get("AL_AMD_DISABLE_CONV_STATS") == "1"
-from .functional import _igemm_eligible, _wpad_cached, cast_cached
+from .functional import _igemm_eligible, _stem_pads, _wpad_cached, cast_cached
 
 
 def _tick(t):
@@ -65,8 +65,8 @@ def conv_bn_act(conv, bn, x, residual=None):
         if _igemm_eligible(C, R * S * C):
             return ext.conv2d_fwd_fused(x, w_c, conv.stride, conv.padding,
                                         scale, shift, bn.relu, res)
-        kdpad = ((R * S * C + 63) // 64) * 64
-        apack = ext.im2col_pack(x, R, S, conv.stride, conv.padding, kdpad)
-        return ext.conv2d_fwd_fused(apack, _wpad_cached(w_c, kdpad), 1, 0,
+        rowpad, kdpad = _stem_pads(R, S, C)
+        apack = ext.im2col_pack(x, R, S, conv.stride, conv.padding, kdpad, rowpad)
+        return ext.conv2d_fwd_fused(apack, _wpad_cached(w_c, kdpad, rowpad), 1, 0,
                                     scale, shift, bn.relu, res)
     return bn(conv(x), residual=residual)

@@ -57,7 +57,7 @@ void al_ce_bwd(const float* probs, const long* targets, const float* scale,
                float* dlogits, int B, int C, void* stream);
 void al_im2col_pack(const void* x, void* out, int N, int H, int W, int C, int R,
                     int S, int P, int Q, int stride, int pad, int kdpad,
-                    void* stream);
+                    int rowpad, void* stream);
 void al_sgd_step(float* p, const float* g, float* buf, float lr, float momentum,
                  float wd, long n, void* shadow, void* stream);
 void al_adam_step(float* p, const float* g, float* m, float* v, float lr, float b1,
@@ -251,13 +251,14 @@ Tensor conv2d_bwd_weight_into(const Tensor& dy, const Tensor& x, long R, long S,
 }
 
 Tensor im2col_pack(const Tensor& x, long R, long S, long stride, long pad,
-                   long kdpad) {
+                   long kdpad, long rowpad) {
   check_bf16_contig(x, "x");
   const int N = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
   const int P = out_dim(H, R, stride, pad), Q = out_dim(W, S, stride, pad);
+  TORCH_CHECK(rowpad % 8 == 0 && rowpad >= S * C && R * rowpad <= kdpad);
   auto out = torch::empty({N, P, Q, kdpad}, x.options());
   al_im2col_pack(x.data_ptr(), out.data_ptr(), N, H, W, C, (int)R, (int)S, P, Q,
-                 (int)stride, (int)pad, (int)kdpad, cur_stream());
+                 (int)stride, (int)pad, (int)kdpad, (int)rowpad, cur_stream());
   return out;
 }
 

@@ -1172,57 +1172,52 @@ extern "C" void al_scatter_s2(const void* tmp, void* dx, int N, int H, int W,
 // ---------------------------------------------------------------------------
 
 __global__ void im2col_pack_kernel(const bf16* __restrict__ x, bf16* __restrict__ out,
-                                   ConvShape sh, int kdpad) {
-  // thread handles 16 consecutive kd of one m row: one coordinate decode,
-  // carry-incremented across the segment, one s16x8-pair store
-  const long total16 = sh.M * (kdpad / 16);
-  for (long i = grid_stride_begin(); i < total16; i += grid_stride_step()) {
-    const int seg = (int)(i % (kdpad / 16));
-    const long m = i / (kdpad / 16);
-    int kd = seg * 16;
-    // decode m once
+                                   ConvShape sh, int kdpad, int rowpad) {
+  // row-padded layout: kd = r*rowpad + (s*C + c), each filter row padded to
+  // rowpad (8-aligned) so every 16B chunk lies within ONE (r)-row — the
+  // gather becomes lane-consecutive contiguous reads (the old 16-element-
+  // per-thread serial gather was latency-bound, ~1.4 TB/s).
+  const int sc = sh.S * sh.C;
+  const int kdrp = sh.R * rowpad;
+  for (long m = blockIdx.x; m < sh.M; m += gridDim.x) {
     const int q = (int)(m % sh.Q);
     long t = m / sh.Q;
     const int p = (int)(t % sh.P);
     const int n = (int)(t / sh.P);
-    const long pix_base = (long)n * sh.H * sh.W;
     const int h0 = p * sh.stride - sh.pad;
     const int w0 = q * sh.stride - sh.pad;
-    int c = kd % sh.C;
-    int rs = kd / sh.C;
-    int s = rs % sh.S;
-    int r = rs / sh.S;
-    s16x8 o[2] = {{0,0,0,0,0,0,0,0},{0,0,0,0,0,0,0,0}};
-    short* os = (short*)o;
-    // fully unrolled: the 16 gathers are independent — issue them all and
-    // let the carry chain (cheap VALU) overlap the loads
-#pragma unroll
-    for (int j = 0; j < 16; ++j) {
-      if (kd + j < sh.KD) {
-        const int h = h0 + r, w = w0 + s;
-        if ((unsigned)h < (unsigned)sh.H && (unsigned)w < (unsigned)sh.W) {
-          const bf16 v = x[(pix_base + (long)h * sh.W + w) * sh.C + c];
-          os[j] = *(const short*)&v;
+    const long nbase = (long)n * sh.H * sh.W * sh.C;
+    bf16* orow = out + m * kdpad;
+    for (int kd = threadIdx.x; kd < kdpad; kd += blockDim.x) {
+      short v = 0;
+      if (kd < kdrp) {
+        const int r = kd / rowpad;
+        const int off = kd - r * rowpad;
+        if (off < sc) {
+          const int h = h0 + r;
+          const int wi = off / sh.C;
+          const int w = w0 + wi;
+          if ((unsigned)h < (unsigned)sh.H && (unsigned)w < (unsigned)sh.W) {
+            const bf16 val = x[nbase + ((long)h * sh.W + w0) * sh.C + off];
+            v = *(const short*)&val;
+          }
         }
-        if (++c == sh.C) { c = 0; if (++s == sh.S) { s = 0; ++r; } }
       }
+      *(short*)&orow[kd] = v;
     }
-    ((s16x8*)out)[i * 2] = o[0];
-    ((s16x8*)out)[i * 2 + 1] = o[1];
   }
 }
 
 extern "C" void al_im2col_pack(const void* x, void* out, int N, int H, int W, int C,
                                int R, int S, int P, int Q, int stride, int pad,
-                               int kdpad, hipStream_t stream) {
+                               int kdpad, int rowpad, hipStream_t stream) {
   ConvShape sh;
   sh.N = N; sh.H = H; sh.W = W; sh.C = C; sh.K = 0; sh.R = R; sh.S = S;
   sh.P = P; sh.Q = Q; sh.stride = stride; sh.pad = pad;
   sh.M = (long)N * P * Q;
   sh.Nout = 0;
   sh.KD = R * S * C;
-  long total = sh.M * kdpad;
-  int blocks = (int)min((total + 255) / 256, (long)8192);
+  int blocks = (int)min(sh.M, (long)32768);
   hipLaunchKernelGGL(im2col_pack_kernel, dim3(blocks), dim3(256), 0, stream,
-                     (const bf16*)x, (bf16*)out, sh, kdpad);
+                     (const bf16*)x, (bf16*)out, sh, kdpad, rowpad);
 }
