@@ -261,6 +261,17 @@ _stats_arena = _StatsArena()
 # saves. Kept opt-in for shapes/hardware where the tradeoff differs.
 _BNBACK_ON = os.environ.get("AL_BNBACK_FUSE", "0") == "1"
 
+# Residual-gradient fan-in fusion: a residual join gives the block input TWO
+# gradient contributions (the BN's dres and the block's conv1 dx), which
+# autograd sums in a separate elementwise pass (~1.3 ms/step of
+# CUDAFunctor_add at B=256). Instead the BN backward attaches dres to the
+# shared tensor and returns None for the residual grad; the consuming conv's
+# bwd-data accumulates it in its epilogue (one extra coalesced read vs a
+# whole 3-pass add). Only engages when the residual tensor IS a conv input
+# (marked at forward) — transition blocks (residual = downsample output)
+# keep the normal path.
+_RESBACK_ON = os.environ.get("AL_RESBACK_FUSE", "1") == "1"
+
 
 def _claim_bnback(x):
     """Claim the upstream BatchNorm's backward side-channel (set by
@@ -291,6 +302,11 @@ class Conv2dNHWC(Function):
         ctx.stride, ctx.padding = stride, padding
         ctx.weight_dtype = weight.dtype
         ctx.bnb = _claim_bnback(x)
+        if _RESBACK_ON and _native_ok(x) and x.requires_grad:
+            try:
+                x._al_res_consumer = True
+            except Exception:
+                pass
         if _native_ok(x):
             ext = require_extension()
             _grad_arena.mark_step()
@@ -311,6 +327,16 @@ class Conv2dNHWC(Function):
         w_c = ctx.w_c
         dy = dy.contiguous()
         dx = dw = None
+        # residual-grad side channel: the upstream BN's backward already ran
+        # (topological order) and may have parked dres on our input tensor
+        dres_s = None
+        if ctx.needs_input_grad[0]:
+            dres_s = getattr(x, "_al_dres", None)
+            if dres_s is not None:
+                try:
+                    x._al_dres = None  # consume once
+                except Exception:
+                    pass
         if _native_ok(x):
             ext = require_extension()
             K, R, S, C = w_c.shape
@@ -339,6 +365,12 @@ class Conv2dNHWC(Function):
                         dx._al_bnsums = (mask, s, sx, dx._version)
                     except Exception:
                         pass
+                elif (dres_s is not None and dres_s.dtype == dy.dtype
+                      and dres_s.is_contiguous()):
+                    dx = ext.conv2d_bwd_data_res(dy, wt_cached(w_c), ctx.stride,
+                                                 ctx.padding, x.shape[1],
+                                                 x.shape[2], dres_s)
+                    dres_s = None
                 else:
                     dx = ext.conv2d_bwd_data(dy, wt_cached(w_c), ctx.stride,
                                              ctx.padding, x.shape[1], x.shape[2])
@@ -365,6 +397,8 @@ class Conv2dNHWC(Function):
                 dw = torch.nn.grad.conv2d_weight(xc.float(), list(wc.shape), dyc,
                                                  stride=ctx.stride, padding=ctx.padding)
                 dw = dw.permute(0, 2, 3, 1).contiguous()
+        if dres_s is not None and dx is not None:
+            dx = dx + dres_s  # routes that could not fuse the accumulation
         if dw is not None:
             dw = dw.to(ctx.weight_dtype)
         return dx, dw, None, None
@@ -387,6 +421,11 @@ class Conv2dNHWCStats(Function):
         ctx.stride, ctx.padding = stride, padding
         ctx.weight_dtype = weight.dtype
         ctx.bnb = _claim_bnback(x)
+        if _RESBACK_ON and _native_ok(x) and x.requires_grad:
+            try:
+                x._al_res_consumer = True
+            except Exception:
+                pass
         if _native_ok(x):
             ext = require_extension()
             _grad_arena.mark_step()
@@ -550,6 +589,8 @@ class BatchNormAct(Function):
         ctx.use_batch_stats = use_batch_stats
         ctx.relu = relu
         ctx.has_residual = residual is not None
+        ctx.res_ref = residual if (_RESBACK_ON and residual is not None
+                                   and _native_ok(x)) else None
         ctx.pg = pg
         ctx.n = n
         if (fuse_backward and relu and residual is None and relu_mask is not None
@@ -634,6 +675,15 @@ class BatchNormAct(Function):
             else:
                 dx = g * dyf
             dx = dx.to(x.dtype)
+        if (dres is not None and getattr(ctx, "res_ref", None) is not None
+                and getattr(ctx.res_ref, "_al_res_consumer", False)):
+            # hand dres to the conv that shares this tensor: its bwd-data
+            # epilogue accumulates it, and autograd skips the fan-in add
+            try:
+                ctx.res_ref._al_dres = dres
+                dres = None
+            except Exception:
+                pass
         return (dx, dgamma, dbeta, None, None, None, None, None, None, dres,
                 None, None, None)
 

@@ -187,6 +187,27 @@ Tensor conv2d_bwd_data(const Tensor& dy, const Tensor& wt, long stride, long pad
   return dx;
 }
 
+Tensor conv2d_bwd_data_res(const Tensor& dy, const Tensor& wt, long stride,
+                           long pad, long H, long W, const Tensor& res) {
+  // bwd-data with the upstream residual GRADIENT accumulated in the
+  // epilogue: dx = conv_bwd(dy) + res — replaces autograd's fan-in add pass
+  // for residual joins (ops/functional.py RESBACK side channel).
+  check_bf16_contig(dy, "dy");
+  check_bf16_contig(wt, "wt");
+  check_bf16_contig(res, "res");
+  const int N = dy.size(0), P = dy.size(1), Q = dy.size(2), K = dy.size(3);
+  const int C = wt.size(0), R = wt.size(1), S = wt.size(2);
+  TORCH_CHECK(wt.size(3) == K, "channel mismatch (wt must be CRSK)");
+  TORCH_CHECK(res.numel() == (long)N * H * W * C, "res size");
+  auto dx = torch::empty({N, (long)H, (long)W, C}, dy.options());
+  al_conv2d_mm(1, dy.data_ptr(), wt.data_ptr(), dx.data_ptr(),
+               zero_page(dy).data_ptr(), N, (int)H, (int)W, C, K, R, S, P, Q,
+               (int)stride, (int)pad, nullptr, nullptr, res.data_ptr(), 0,
+               nullptr, nullptr, nullptr, nullptr, nullptr, nullptr,
+               cur_stream());
+  return dx;
+}
+
 std::vector<Tensor> conv2d_bwd_data_bn(const Tensor& dy, const Tensor& wt,
                                        long stride, long pad, long H, long W,
                                        const Tensor& mask, const Tensor& xbn,
@@ -609,6 +630,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv2d_fwd_stats", &conv2d_fwd_stats);
   m.def("conv2d_bwd_data", &conv2d_bwd_data);
   m.def("conv2d_bwd_data_bn", &conv2d_bwd_data_bn);
+  m.def("conv2d_bwd_data_res", &conv2d_bwd_data_res);
   m.def("conv2d_bwd_weight", &conv2d_bwd_weight);
   m.def("conv2d_bwd_weight_into", &conv2d_bwd_weight_into);
   m.def("bn_stats", &bn_stats);

@@ -437,6 +437,13 @@ __global__ void igemm_kernel(const bf16* __restrict__ A, const bf16* __restrict_
           if (epi_relu) f = fmaxf(f, 0.f);
           v[j] = f2bits(f);
         }
+      } else if (epi_res) {
+        // plain residual-gradient accumulation (bwd-data): dx += dres,
+        // replacing autograd's separate fan-in add pass
+        const s16x8 rv = *(const s16x8*)(epi_res + off);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          v[j] = f2bits(bits2f(v[j]) + bits2f(rv[j]));
       }
       *(s16x8*)(out + off) = v;
     } else {  // column tail: elementwise, all epi reads guarded
@@ -446,6 +453,8 @@ __global__ void igemm_kernel(const bf16* __restrict__ A, const bf16* __restrict_
           f = f * epi_scale[col0 + j] + epi_shift[col0 + j];
           if (epi_res) f += bf2f(epi_res[off + j]);
           if (epi_relu) f = fmaxf(f, 0.f);
+        } else if (epi_res) {
+          f += bf2f(epi_res[off + j]);
         }
         short sj = f2bits(f);
         out[off + j] = *(bf16*)&sj;
@@ -856,6 +865,11 @@ __global__ void gemm256_nt_kernel(const bf16* __restrict__ A,
         if (epi_relu) f = fmaxf(f, 0.f);
         v[j] = f2bits(f);
       }
+    } else if (epi_res) {  // plain residual-grad accumulation (bwd-data)
+      const s16x8 rv = *(const s16x8*)(epi_res + off);
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        v[j] = f2bits(bits2f(v[j]) + bits2f(rv[j]));
     }
     *(s16x8*)(out + off) = v;
   }
@@ -926,6 +940,8 @@ __global__ void conv_direct_kernel(const bf16* __restrict__ A,
       acc = acc * epi_scale[j] + epi_shift[j];
       if (epi_res) acc += bf2f(epi_res[i]);
       if (epi_relu) acc = fmaxf(acc, 0.f);
+    } else if (epi_res) {
+      acc += bf2f(epi_res[i]);
     }
     if (stat_sum) {
       atomicAdd(&stat_sum[j], acc);

@@ -164,6 +164,57 @@ def test_conv_bwd_data_bn_fused(case):
     assert relerr(sx, sx_ref) < 1e-3, "fused sum_dy_xhat mismatch"
 
 
+def test_conv_bwd_data_res_fused():
+    """conv2d_bwd_data_res == conv2d_bwd_data + dres (plain epilogue add)."""
+    n, h, w, c, k, r = 4, 16, 16, 64, 64, 3
+    torch.manual_seed(8)
+    dy = torch.randn(n, h, w, k).cuda().to(torch.bfloat16)
+    wt = (torch.randn(k, r, r, c) * 0.05).cuda().to(torch.bfloat16)
+    wt_perm = wt.permute(3, 1, 2, 0).contiguous()
+    res = torch.randn(n, h, w, c).cuda().to(torch.bfloat16)
+    ref = EXT.conv2d_bwd_data(dy, wt_perm, 1, 1, h, w).float() + res.float()
+    got = EXT.conv2d_bwd_data_res(dy, wt_perm, 1, 1, h, w, res)
+    assert relerr(got, ref) < 2e-2
+
+
+def test_resnet_resback_fusion_matches_disabled():
+    """Model-level: the residual-grad fan-in fusion must not change
+    gradients (vs autograd's separate add), within the atomics
+    nondeterminism floor."""
+    import os
+    from active_learning_amd.models import get_networks
+    from active_learning_amd.ops import functional as AFn
+    torch.manual_seed(12)
+    net = get_networks("synthetic_cifar10", "SSLResNet18").cuda()
+    x = torch.randn(8, 3, 16, 16, device="cuda")
+    y = torch.randint(0, 10, (8,), device="cuda")
+
+    def run(fuse):
+        old = AFn._RESBACK_ON
+        AFn._RESBACK_ON = fuse
+        try:
+            for p_ in net.parameters():
+                p_.grad = None
+            net.train()
+            out = net(x)
+            torch.nn.functional.cross_entropy(out, y).backward()
+            return {n_: p_.grad.detach().clone() for n_, p_ in
+                    net.named_parameters() if p_.grad is not None}
+        finally:
+            AFn._RESBACK_ON = old
+
+    g_off = run(False)
+    g_off2 = run(False)
+    g_on = run(True)
+    assert g_on.keys() == g_off.keys()
+    for n_ in g_off:
+        base = relerr(g_off2[n_], g_off[n_])
+        err = relerr(g_on[n_], g_off[n_])
+        tol = max(5e-3, 10 * base)
+        assert err < tol, (f"resback fusion changed grad of {n_}: {err} "
+                           f"(floor {base})")
+
+
 def test_resnet_bnback_fusion_matches_disabled():
     """Model-level: gradients with the BN-backward fusion active must match
     the unfused path (same kernels otherwise)."""
