@@ -93,9 +93,17 @@ __global__ void kcenter_kernel(const float* __restrict__ dist,
         partial[nb + blockIdx.x] = b;
       }
       grid.sync();
-      // block 0, wave 0: pick mode and locate the target block with a
-      // wave-parallel prefix over the <=512 per-block partials
-      if (blockIdx.x == 0 && tid < kWave) {
+      // EVERY block's wave 0 redundantly computes the CDF prefix over the
+      // <=512 per-block partials and decides whether the target falls in
+      // ITS range — identical arithmetic on identical inputs picks exactly
+      // one owner with NO second grid.sync (the former block-0-then-sync
+      // scheme cost one more sync per iteration, ~30% of the loop).
+      __shared__ int s_tb;
+      __shared__ float s_target;
+      __shared__ int s_mode;
+      if (tid == 0) s_tb = -1;
+      __syncthreads();
+      if (tid < kWave) {
         const int lane = tid;
         float wsum = 0.f, csum = 0.f;
         for (int b = lane; b < nb; b += kWave) {
@@ -137,16 +145,18 @@ __global__ void kcenter_kernel(const float* __restrict__ dist,
         }
         if (tb < 0) tb = nb - 1;  // rounding tail
         if (lane == 0) {
-          st->g_mode = mode;
-          st->g_target = target;
-          st->g_found = -1 - tb;   // negative encoding: block tb scans
+          s_tb = tb;
+          s_target = target;
+          s_mode = mode;
         }
       }
-      grid.sync();
-      // pass 2: the selected block finds its in-range index with a
+      __syncthreads();
+      // pass 2: the owning block finds its in-range index with a
       // block-parallel prefix (per-thread contiguous sub-ranges, LDS scan)
-      if (-1 - st->g_found == blockIdx.x) {
-        const int mode = st->g_mode;
+      if (s_tb == (int)blockIdx.x) {
+        if (tid == 0) st->g_found = 0;  // owner-only: reset claim flag
+        __syncthreads();
+        const int mode = s_mode;
         const long span = i1 - i0;
         const long per = (span + blockDim.x - 1) / blockDim.x;
         const long j0 = i0 + tid * per;
@@ -166,7 +176,7 @@ __global__ void kcenter_kernel(const float* __restrict__ dist,
           s_pref[tid] += add;
           __syncthreads();
         }
-        const float target = st->g_target;
+        const float target = s_target;
         const float excl = (tid == 0) ? 0.f : s_pref[tid - 1];
         const bool mine = (mysum > 0.f) && (target >= excl) &&
                           (target < excl + mysum);
@@ -188,11 +198,11 @@ __global__ void kcenter_kernel(const float* __restrict__ dist,
             sel[t] = pick;
             labeled[pick] = 1;
             st->g_j = (int)pick;
-            st->g_found = (int)1;  // claimed
+            st->g_found = 1;  // claimed (owner-block scope)
           }
         }
         __syncthreads();
-        if (tid == 0 && st->g_found < 0) {
+        if (tid == 0 && st->g_found != 1) {
           // degenerate: every sub-range empty — fall back to first unlabeled
           for (long i = 0; i < n; ++i)
             if (!labeled[i]) {
