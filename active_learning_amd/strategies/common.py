@@ -241,7 +241,13 @@ def _sharded_forward_pool(strategy, dataset, idxs, want_embedding, keep_device):
     net.to(strategy.device)
     logits0, emb0, labels0 = _infer_shard(net, strategy.device, dataset,
                                           shards[0], loader_args, want_embedding)
-    results = _query_pool.collect(n_tasks)
+    try:
+        results = _query_pool.collect(n_tasks)
+    except Exception:
+        # a failed/timed-out worker may leave stale results in the queue;
+        # tear the pool down so the next query starts clean
+        _query_pool.shutdown()
+        raise
 
     dev = strategy.device if keep_device else torch.device("cpu")
     logits_parts = [logits0.to(dev)]
