@@ -11,6 +11,8 @@
 // All element passes are memory-bound: bf16x8 (16 B/lane) vectorized, one
 // read of each operand, grid-stride (G11/G13).
 
+#include <cstdlib>
+
 #include "al_common.h"
 
 // ---------------------------------------------------------------------------
@@ -66,6 +68,29 @@ __global__ void bn_stats_kernel(const bf16* __restrict__ x, float* __restrict__ 
   }
 }
 
+// Nontemporal load/store for the element passes when the streams are far
+// bigger than L2 (32 MiB across the 8 XCDs): measured +10-12% on the
+// >=64 MiB/stream shapes and -17% on the L2-reachable 7x7x2048 shape
+// (tools/bn_ab.py), so the launchers pick per launch by footprint.
+template <bool NT>
+AL_DEV s16x8 ld8(const s16x8* p) {
+  if constexpr (NT) return __builtin_nontemporal_load(p);
+  else return *p;
+}
+template <bool NT>
+AL_DEV void st8(s16x8* p, s16x8 v) {
+  if constexpr (NT) __builtin_nontemporal_store(v, p);
+  else *p = v;
+}
+
+static inline bool bn_nt(long rows, int C) {
+  static const bool on = [] {
+    const char* e = getenv("AL_BN_NT");
+    return !(e && e[0] == '0');
+  }();
+  return on && rows * (long)C * 2 >= (64L << 20);
+}
+
 static inline int bn_cg_per_block(int C) {
   int c8 = C / 8;
   int cg = 1;
@@ -109,7 +134,7 @@ extern "C" int al_bn_reduce_blocks(long rows, int C) {
 // and no parameter reloads (the v1 grid-stride form re-read mean/invstd/
 // gamma/beta per chunk and spent a 64-bit modulo per 16B — measured ~3.2
 // TB/s; this form matches the stats kernels' ~5+ TB/s streaming pattern).
-template <bool RELU, bool RES>
+template <bool RELU, bool RES, bool NT>
 __global__ void bn_norm2_kernel(const bf16* __restrict__ x, bf16* __restrict__ y,
                                 const float* __restrict__ mean,
                                 const float* __restrict__ invstd,
@@ -135,9 +160,9 @@ __global__ void bn_norm2_kernel(const bf16* __restrict__ x, bf16* __restrict__ y
   const long step = (long)gridDim.x * rows_per_block;
   #pragma unroll 2
   for (long r = row0; r < rows; r += step) {
-    s16x8 v = ((const s16x8*)x)[r * C8 + c8];
+    s16x8 v = ld8<NT>((const s16x8*)x + r * C8 + c8);
     s16x8 rv;
-    if (RES) rv = ((const s16x8*)res)[r * C8 + c8];
+    if (RES) rv = ld8<NT>((const s16x8*)res + r * C8 + c8);
     s16x8 o;
     unsigned char mb = 0;
 #pragma unroll
@@ -150,7 +175,7 @@ __global__ void bn_norm2_kernel(const bf16* __restrict__ x, bf16* __restrict__ y
       }
       o[j] = f2bits(f);
     }
-    ((s16x8*)y)[r * C8 + c8] = o;
+    st8<NT>((s16x8*)y + r * C8 + c8, o);
     // one mask BIT per element: the backward passes read this instead of
     // re-streaming y (drops a full activation read from both)
     if (RELU && relu_mask) relu_mask[r * C8 + c8] = mb;
@@ -167,20 +192,23 @@ extern "C" void al_bn_norm_fwd(const void* x, void* y, const float* mean,
   const int rpb = 256 / cg;
   int row_blocks = (int)min((rows + rpb - 1) / rpb, (long)4096);
   dim3 grid(row_blocks, (C / 8 + cg - 1) / cg), block(256);
-#define CASE(RELU_, RES_) \
-  hipLaunchKernelGGL((bn_norm2_kernel<RELU_, RES_>), grid, block, 0, stream, \
+#define CASE1(RELU_, RES_, NT_) \
+  hipLaunchKernelGGL((bn_norm2_kernel<RELU_, RES_, NT_>), grid, block, 0, stream, \
                      (const bf16*)x, (bf16*)y, mean, invstd, gamma, beta, r, \
                      (unsigned char*)relu_mask, rows, C, cg)
+#define CASE(RELU_, RES_) \
+  do { if (bn_nt(rows, C)) CASE1(RELU_, RES_, true); else CASE1(RELU_, RES_, false); } while (0)
   if (relu) { if (r) CASE(true, true); else CASE(true, false); }
   else      { if (r) CASE(false, true); else CASE(false, false); }
 #undef CASE
+#undef CASE1
 }
 
 // ---------------------------------------------------------------------------
 // backward reduce: sum_dy, sum_dy_xhat per channel (dy masked by y>0 if RELU)
 // ---------------------------------------------------------------------------
 
-template <bool RELU>
+template <bool RELU, bool NT>
 __global__ void bn_bwd_reduce_kernel(const bf16* __restrict__ dy,
                                      const bf16* __restrict__ x,
                                      const unsigned char* __restrict__ relu_mask,
@@ -206,8 +234,8 @@ __global__ void bn_bwd_reduce_kernel(const bf16* __restrict__ dy,
   const long step = (long)gridDim.x * rows_per_block;
   for (long r = row0; r < rows; r += step) {
     const long off = r * C8 + c8;
-    s16x8 gv = ((const s16x8*)dy)[off];
-    s16x8 xv = ((const s16x8*)x)[off];
+    s16x8 gv = ld8<NT>((const s16x8*)dy + off);
+    s16x8 xv = ld8<NT>((const s16x8*)x + off);
     unsigned char mb = 0xff;
     if (RELU) mb = relu_mask[off];
 #pragma unroll
@@ -249,14 +277,17 @@ extern "C" void al_bn_bwd_reduce(const void* dy, const void* x, const void* relu
   const int rpb = 256 / cg;
   int row_blocks = (int)min((rows + rpb - 1) / rpb, (long)bn_row_block_cap(C));
   dim3 grid(row_blocks, (C / 8 + cg - 1) / cg);
-  if (relu)
-    hipLaunchKernelGGL((bn_bwd_reduce_kernel<true>), grid, block, 0, stream,
-                       (const bf16*)dy, (const bf16*)x, (const unsigned char*)relu_mask, mean, invstd,
-                       sum_dy, sum_dy_xhat, rows, C, cg);
-  else
-    hipLaunchKernelGGL((bn_bwd_reduce_kernel<false>), grid, block, 0, stream,
-                       (const bf16*)dy, (const bf16*)x, (const unsigned char*)relu_mask, mean, invstd,
-                       sum_dy, sum_dy_xhat, rows, C, cg);
+#define CASE1(RELU_, NT_) \
+  hipLaunchKernelGGL((bn_bwd_reduce_kernel<RELU_, NT_>), grid, block, 0, stream, \
+                     (const bf16*)dy, (const bf16*)x, \
+                     (const unsigned char*)relu_mask, mean, invstd, sum_dy, \
+                     sum_dy_xhat, rows, C, cg)
+#define CASE(RELU_) \
+  do { if (bn_nt(rows, C)) CASE1(RELU_, true); else CASE1(RELU_, false); } while (0)
+  if (relu) CASE(true);
+  else CASE(false);
+#undef CASE
+#undef CASE1
 }
 
 // ---------------------------------------------------------------------------
@@ -269,7 +300,7 @@ extern "C" void al_bn_bwd_reduce(const void* dy, const void* x, const void* relu
 // v2 backward elementwise: same fixed-channel-strip structure; the batch
 // terms fold to dx = gi*g - t2*x + c0 (two fma) with per-channel (gi, t2,
 // c0) hoisted into registers.
-template <bool RELU, bool BATCH, bool RES>
+template <bool RELU, bool BATCH, bool RES, bool NT>
 __global__ void bn_bwd2_kernel(const bf16* __restrict__ dy, const bf16* __restrict__ x,
                                const unsigned char* __restrict__ relu_mask,
                                const float* __restrict__ mean,
@@ -300,9 +331,9 @@ __global__ void bn_bwd2_kernel(const bf16* __restrict__ dy, const bf16* __restri
   #pragma unroll 2
   for (long r = row0; r < rows; r += step) {
     const long i = r * C8 + c8;
-    s16x8 gv = ((const s16x8*)dy)[i];
+    s16x8 gv = ld8<NT>((const s16x8*)dy + i);
     s16x8 xv;
-    if (BATCH) xv = ((const s16x8*)x)[i];
+    if (BATCH) xv = ld8<NT>((const s16x8*)x + i);
     unsigned char mb = 0xff;
     if (RELU) mb = relu_mask[i];
     s16x8 odx, ores;
@@ -316,8 +347,8 @@ __global__ void bn_bwd2_kernel(const bf16* __restrict__ dy, const bf16* __restri
       else v = gi[j] * g;
       odx[j] = f2bits(v);
     }
-    ((s16x8*)dx)[i] = odx;
-    if (RES) ((s16x8*)dres)[i] = ores;
+    st8<NT>((s16x8*)dx + i, odx);
+    if (RES) st8<NT>((s16x8*)dres + i, ores);
   }
 }
 
@@ -331,11 +362,14 @@ extern "C" void al_bn_bwd(const void* dy, const void* x, const void* relu_mask,
   int row_blocks = (int)min((rows + rpb - 1) / rpb, (long)4096);
   dim3 grid(row_blocks, (C / 8 + cg - 1) / cg), block(256);
   float inv_n = 1.0f / n;
-#define CASE(RELU_, BATCH_, RES_) \
-  hipLaunchKernelGGL((bn_bwd2_kernel<RELU_, BATCH_, RES_>), grid, block, 0, stream, \
+#define CASE1(RELU_, BATCH_, RES_, NT_) \
+  hipLaunchKernelGGL((bn_bwd2_kernel<RELU_, BATCH_, RES_, NT_>), grid, block, 0, stream, \
                      (const bf16*)dy, (const bf16*)x, (const unsigned char*)relu_mask, mean, invstd, \
                      gamma, sum_dy, sum_dy_xhat, inv_n, (bf16*)dx, (bf16*)dres, \
                      rows, C, cg)
+#define CASE(RELU_, BATCH_, RES_) \
+  do { if (bn_nt(rows, C)) CASE1(RELU_, BATCH_, RES_, true); \
+       else CASE1(RELU_, BATCH_, RES_, false); } while (0)
   if (relu) {
     if (use_batch_stats) { if (has_res) CASE(true, true, true); else CASE(true, true, false); }
     else                 { if (has_res) CASE(true, false, true); else CASE(true, false, false); }
@@ -344,6 +378,7 @@ extern "C" void al_bn_bwd(const void* dy, const void* x, const void* relu_mask,
     else                 { if (has_res) CASE(false, false, true); else CASE(false, false, false); }
   }
 #undef CASE
+#undef CASE1
 }
 
 // ---------------------------------------------------------------------------
