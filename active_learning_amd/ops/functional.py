@@ -415,6 +415,10 @@ class Conv2dNHWCStats(Function):
 
     @staticmethod
     def forward(ctx, x, weight, stride, padding):
+        # without this, autograd MATERIALIZES a zero [K] tensor for each of
+        # the two non-differentiable outputs (s, ss) on every backward:
+        # 2 x 53 FillFunctor launches/step at B=256 (tools/fill_audit.py)
+        ctx.set_materialize_grads(False)
         w_c = cast_cached(weight, x.dtype)
         ctx.save_for_backward(x)
         ctx.w_c = w_c
@@ -457,6 +461,8 @@ class Conv2dNHWCStats(Function):
 
     @staticmethod
     def backward(ctx, dy, _ds, _dss):
+        if dy is None:  # only the (never-used) stats outputs were consumed
+            return None, None, None, None
         return Conv2dNHWC.backward(ctx, dy)
 
 
