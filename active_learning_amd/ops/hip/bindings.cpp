@@ -65,6 +65,7 @@ void al_adam_step(float* p, const float* g, float* m, float* v, float lr, float 
                   void* stream);
 void al_sgd_step_multi(const void* table, int nchunks, float lr, float momentum,
                        float wd, int zero_grad, void* stream);
+void al_wt_refresh(const void* table, int nrows, void* stream);
 void al_sgd_step_multi_dev(const void* table, int nchunks, const float* hyper,
                            int zero_grad, void* stream);
 int al_kcenter_greedy(const float* dist, float* min_dist, unsigned char* labeled,
@@ -515,6 +516,15 @@ void sgd_step_multi_dev(const Tensor& table, int64_t nchunks,
                         hyper.data_ptr<float>(), zero_grad ? 1 : 0, cur_stream());
 }
 
+void wt_refresh_multi(const Tensor& table, int64_t nrows) {
+  // one 64x64 (k, c) tile per row: batched (K,R,S,C)->(C,R,S,K) bf16 weight
+  // transpose refresh after the fused SGD update (see optim.hip)
+  TORCH_CHECK(table.is_cuda() && table.scalar_type() == torch::kInt64 &&
+              table.is_contiguous());
+  TORCH_CHECK(table.numel() >= nrows * 4, "wt table too small");
+  al_wt_refresh(table.data_ptr<int64_t>(), (int)nrows, cur_stream());
+}
+
 void adam_step(Tensor& p, const Tensor& g, Tensor& m, Tensor& v, double lr, double b1,
                double b2, double eps, double wd, double bc1, double bc2) {
   TORCH_CHECK(p.is_cuda() && p.scalar_type() == torch::kFloat32);
@@ -649,6 +659,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sgd_step", &sgd_step);
   m.def("sgd_step_multi", &sgd_step_multi);
   m.def("sgd_step_multi_dev", &sgd_step_multi_dev);
+  m.def("wt_refresh_multi", &wt_refresh_multi);
   m.def("kcenter_greedy_dev", &kcenter_greedy_dev);
   m.def("scatter_s2", &scatter_s2);
   m.def("badge_gram", &badge_gram);

@@ -127,6 +127,54 @@ extern "C" void al_sgd_step_multi_dev(const void* table, int nchunks,
                        stream, (const SgdChunk*)table, hyper);
 }
 
+// ---------------------------------------------------------------------------
+// Batched (K,R,S,C) -> (C,R,S,K) bf16 weight transpose: refreshes every
+// conv's cached bwd-data permutation in ONE launch right after the fused SGD
+// update (the per-conv ATen permute+clone path cost ~53 launches / 0.35 ms
+// per step at B=256, tools/fill_audit.py). One workgroup per 64x64 (k, c)
+// tile of one rs plane; host enqueues only full tiles (K, C % 64 == 0 —
+// the igemm-eligible conv weights). Table row: 4 int64 =
+// (src, dst, k0 | c0<<16 | rs<<32, K | C<<16 | RS<<32).
+__global__ void wt_refresh_kernel(const long* __restrict__ table) {
+  const long* row = table + (long)blockIdx.x * 4;
+  const bf16* __restrict__ src = (const bf16*)row[0];
+  bf16* __restrict__ dst = (bf16*)row[1];
+  const long m1 = row[2], m2 = row[3];
+  const int k0 = (int)(m1 & 0xffff), c0 = (int)((m1 >> 16) & 0xffff);
+  const int rs = (int)(m1 >> 32);
+  const int K = (int)(m2 & 0xffff), C = (int)((m2 >> 16) & 0xffff);
+  const int RS = (int)(m2 >> 32);
+  __shared__ short tile[64][68];  // 68: de-phase LDS banks across rows
+  // load: thread -> (k row = t/4, 16-col strip = (t%4)*16); 2 x 16B per lane
+  const int kl = threadIdx.x >> 2, cs = (threadIdx.x & 3) << 4;
+  {
+    const s16x8* sp =
+        (const s16x8*)(src + ((long)(k0 + kl) * RS + rs) * C + c0 + cs);
+#pragma unroll
+    for (int u = 0; u < 2; ++u) {
+      s16x8 v = sp[u];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) tile[kl][cs + u * 8 + j] = v[j];
+    }
+  }
+  __syncthreads();
+  // store transposed: thread -> (c row = t/4, 16-k strip)
+  const int cl = threadIdx.x >> 2, ks = (threadIdx.x & 3) << 4;
+  s16x8* dp = (s16x8*)(dst + ((long)(c0 + cl) * RS + rs) * K + k0 + ks);
+#pragma unroll
+  for (int u = 0; u < 2; ++u) {
+    s16x8 v;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) v[j] = tile[ks + u * 8 + j][cl];
+    dp[u] = v;
+  }
+}
+
+extern "C" void al_wt_refresh(const void* table, int nrows, hipStream_t stream) {
+  hipLaunchKernelGGL(wt_refresh_kernel, dim3(nrows), dim3(256), 0, stream,
+                     (const long*)table);
+}
+
 __global__ void adam_kernel(float* __restrict__ p, const float* __restrict__ g,
                             float* __restrict__ m, float* __restrict__ v, float lr,
                             float b1, float b2, float eps, float wd, float bc1,

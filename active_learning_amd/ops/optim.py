@@ -39,6 +39,7 @@ class FusedSGD(torch.optim.Optimizer):
                         dampening=dampening)
         super().__init__(params, defaults)
         self._mt_cache = {}  # group idx -> (fingerprint, device table, nchunks)
+        self._wt_cache = {}  # group idx -> (fingerprint, tile table, nrows)
         self._mt_keepalive = []  # pinned/device tables referenced by live graphs
         self._hyper_dev = None   # device (lr, momentum, wd); graph-capture mode
         self._hyper_host = None
@@ -174,14 +175,54 @@ class FusedSGD(torch.optim.Optimizer):
         else:
             ext.sgd_step_multi(entry[1], entry[2], lr, momentum, wd,
                                self._persistent)
+        wt_fp, wt_entries = [], []
         for p, shadow in post:
             if shadow is not None:
-                if getattr(shadow, "_al_wt", None) is not None:
-                    shadow._al_wt = None
+                wt = getattr(shadow, "_al_wt", None)
+                if wt is not None:
+                    if (shadow.dim() == 4 and shadow.shape[0] % 64 == 0
+                            and shadow.shape[3] % 64 == 0):
+                        # refreshed by ONE batched transpose kernel below
+                        # instead of a per-conv ATen permute+clone
+                        wt_fp.append((shadow.data_ptr(), wt.data_ptr())
+                                     + tuple(shadow.shape))
+                        wt_entries.append(shadow)
+                    else:
+                        shadow._al_wt = None
                 p._al_cast = (p._version, shadow)
             else:
                 _invalidate(p)
             bump_tick(p)
+        if wt_fp:
+            self._launch_wt_refresh(ext, gi, tuple(wt_fp), wt_entries)
+
+    def _launch_wt_refresh(self, ext, gi, fp, shadows):
+        """Refresh every cached (C,R,S,K) bwd-data weight permutation in one
+        launch (64x64 tiles; only full tiles are enqueued — the host gates on
+        K, C % 64). Table cached by pointer fingerprint like the sgd chunk
+        table, so steady-state (and captured) steps launch with no H2D."""
+        entry = self._wt_cache.get(gi)
+        if entry is None or entry[0] != fp:
+            if torch.cuda.is_current_stream_capturing():
+                for shadow in shadows:
+                    shadow._al_wt = None  # lazy ATen rebuild for this step
+                return
+            rows = []
+            for (sp, dp, K, R, S, C) in fp:
+                RS = R * S
+                for rs in range(RS):
+                    for k0 in range(0, K, 64):
+                        for c0 in range(0, C, 64):
+                            rows.append((sp, dp,
+                                         k0 | (c0 << 16) | (rs << 32),
+                                         K | (C << 16) | (RS << 32)))
+            host = torch.tensor(rows, dtype=torch.int64).pin_memory()
+            dev = torch.empty_like(host, device=shadows[0].device)
+            dev.copy_(host, non_blocking=True)
+            self._mt_keepalive.append((host, dev))
+            entry = (fp, dev, len(rows))
+            self._wt_cache[gi] = entry
+        ext.wt_refresh_multi(entry[1], entry[2])
 
     def _per_tensor_fallback(self, ext, params, lr, momentum, wd):
         for p in params:

@@ -500,6 +500,48 @@ def test_sgd_multi_tensor_gpu():
     assert len(o_g._mt_cache) == 1 and o_g._mt_cache[0][2] >= 10
 
 
+def test_wt_refresh_batched_gpu():
+    """The batched 64x64-tile weight-transpose refresh (one launch after the
+    multi-tensor SGD update) must leave every cached (C,R,S,K) permutation
+    bitwise equal to the ATen permute of the refreshed bf16 shadow."""
+    from active_learning_amd.ops.functional import cast_cached, wt_cached
+    from active_learning_amd.ops.optim import FusedSGD
+    torch.manual_seed(5)
+    shapes = [(64, 3, 3, 64), (256, 1, 1, 64), (128, 3, 3, 128),
+              (2048, 1, 1, 512), (64, 1, 1, 256)]
+    ps = [torch.nn.Parameter(torch.randn(*s, device="cuda")) for s in shapes]
+    opt = FusedSGD(ps, lr=0.05, momentum=0.9, weight_decay=1e-4)
+    for it in range(4):
+        for p in ps:
+            # touch the caches the way a conv forward/backward would
+            wt_cached(cast_cached(p, torch.bfloat16))
+            p.grad = torch.randn_like(p)
+        opt.step()
+    torch.cuda.synchronize()
+    # steady state: the batched table exists and covers every weight
+    assert 0 in opt._wt_cache and opt._wt_cache[0][2] > 0
+    for p, s in zip(ps, shapes):
+        shadow = p._al_cast[1]
+        wt = shadow._al_wt
+        assert wt is not None, s
+        ref = shadow.permute(3, 1, 2, 0).contiguous()
+        assert torch.equal(wt, ref), f"wt mismatch for {s}"
+        # and the shadow itself matches a fresh cast of the master
+        assert torch.equal(shadow, p.detach().to(torch.bfloat16)), s
+
+    # non-64-multiple weights keep the lazy per-tensor path (wt invalidated)
+    q = torch.nn.Parameter(torch.randn(60, 3, 3, 64, device="cuda"))
+    opt2 = FusedSGD([q], lr=0.05, momentum=0.9)
+    wt_cached(cast_cached(q, torch.bfloat16))
+    q.grad = torch.randn_like(q)
+    opt2.step()
+    torch.cuda.synchronize()
+    shadow = q._al_cast[1]
+    assert getattr(shadow, "_al_wt", None) is None
+    wt = wt_cached(shadow)
+    assert torch.equal(wt, shadow.permute(3, 1, 2, 0).contiguous())
+
+
 # --------------------------------------------------------------------------- #
 # model-level: ResNet-18 forward/backward GPU bf16 vs CPU fp32
 # --------------------------------------------------------------------------- #

@@ -51,6 +51,35 @@ def main():
     for s, n in shapes.most_common(25):
         print(f"{n:5d}  {s}")
 
+    # copies and any high-count ATen ops (launch-bound suspects)
+    copies = collections.Counter()
+    counts = collections.Counter()
+    for e in prof.events():
+        if e.name.startswith("aten::"):
+            counts[e.name] += 1
+            if e.name in ("aten::copy_", "aten::cat", "aten::clone",
+                          "aten::contiguous", "aten::to", "aten::_to_copy"):
+                copies[f"{e.name} {e.input_shapes}"] += 1
+    print("\nhigh-count aten ops (>=40/step):")
+    for nm, n in counts.most_common(25):
+        if n >= 40:
+            print(f"{n:5d}  {nm}")
+    print("\ncopy-ish ops by shape:")
+    for s, n in copies.most_common(20):
+        print(f"{n:5d}  {s}")
+
+    # where the copy/cast GPU time actually goes
+    ka = prof.key_averages()
+    print("\naten op CUDA totals (copy/cast/clone/transpose suspects):")
+    for row in ka:
+        if row.key in ("aten::copy_", "aten::to", "aten::_to_copy",
+                       "aten::clone", "aten::contiguous", "aten::permute",
+                       "aten::cat") or "copyBuffer" in row.key \
+                or "elementwise" in row.key or "Fill" in row.key:
+            cuda_us = getattr(row, "self_device_time_total",
+                              getattr(row, "self_cuda_time_total", 0))
+            print(f"{row.count:6d}  {cuda_us/1e3:8.3f} ms  {row.key[:90]}")
+
 
 if __name__ == "__main__":
     main()
