@@ -1237,7 +1237,9 @@ __global__ void im2col_pack_kernel(const bf16* __restrict__ x, bf16* __restrict_
         }
       }
     }
-    ((s16x8*)out)[i] = o;
+    // nontemporal: apack is ~16x the input (1.2 GB at the B=256 stem), far
+    // beyond L2 — don't let the stream evict the x rows being gathered
+    __builtin_nontemporal_store(o, (s16x8*)out + i);
     // advance (c8, q, p, n) by `step` items without dividing
     c8 += dc8;
     int mc_ = c8 >= c8n;
