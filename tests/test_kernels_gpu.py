@@ -417,6 +417,41 @@ def test_bn_function_gpu_matches_cpu(training, relu, residual):
         assert relerr(resg.grad, resc.grad) < 0.03
 
 
+def test_bn_large_stream_nontemporal_path():
+    """BN element passes at a >=64 MiB/stream shape: exercises the
+    nontemporal (NT=true) template instantiations the footprint gate picks
+    for the big layers (bn.hip bn_nt()) — the small-shape tests above only
+    ever run the temporal variants. Values must match plain fp32 torch math
+    on the same bf16-rounded inputs."""
+    rows, c = 1 << 21, 64  # 256 MiB per bf16 stream -> NT path
+    torch.manual_seed(3)
+    x = torch.randn(rows, 1, 1, c, device="cuda").to(torch.bfloat16)
+    dy = torch.randn_like(x)
+    mean = torch.randn(c, device="cuda")
+    invstd = torch.rand(c, device="cuda") + 0.5
+    g = torch.rand(c, device="cuda") + 0.5
+    b = torch.randn(c, device="cuda")
+    e = x.new_empty(0)
+
+    y, mask = EXT.bn_norm_fwd(x, mean, invstd, g, b, True, e, True)
+    xf = x.float()
+    y_ref = torch.relu((xf - mean) * invstd * g + b)
+    assert relerr(y, y_ref) < 5e-3
+
+    s, sx = EXT.bn_bwd_reduce(dy, x, mask, mean, invstd, True)
+    dym = dy.float() * (y_ref > 0)
+    s_ref = dym.sum((0, 1, 2))
+    sx_ref = (dym * (xf - mean) * invstd).sum((0, 1, 2))
+    assert relerr(s, s_ref) < 5e-3 and relerr(sx, sx_ref) < 5e-3
+
+    n = float(rows)
+    dx, _ = EXT.bn_bwd(dy, x, mask, mean, invstd, g, s_ref, sx_ref, n,
+                       True, True, False)
+    xhat = (xf - mean) * invstd
+    dx_ref = g * invstd * (dym - s_ref / n - xhat * (sx_ref / n))
+    assert relerr(dx, dx_ref) < 5e-3
+
+
 def test_maxpool_gpu():
     # bf16-quantized input: CPU/GPU agree on argmax except exact ties
     x = torch.randn(2, 17, 17, 64).to(torch.bfloat16).float()
