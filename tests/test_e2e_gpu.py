@@ -132,3 +132,20 @@ def test_pretrained_linear_eval_round_gpu(tmp_path):
     torch.cuda.synchronize()
     import os
     assert os.path.exists(s.generate_weight_paths()["best_ckpt"])
+
+
+@pytest.mark.timeout(240)
+def test_six_round_cache_lifecycle_gpu(tmp_path):
+    """Six AL rounds on the GPU: every round re-initializes the weights
+    (fresh bf16 shadows + wt permutations, grad/stats arena eviction, new
+    fused-SGD chunk + wt-refresh tables). Guards the cross-round lifecycle
+    of all pointer-fingerprinted caches."""
+    from active_learning_amd.cli import get_args
+    from active_learning_amd.main_al import main
+    args = get_args([
+        "--dataset", "synthetic_cifar10", "--rounds", "6", "--round_budget", "4",
+        "--n_epoch", "2", "--early_stop_patience", "2", "--debug_mode",
+        "--ckpt_path", str(tmp_path / "ckpt"), "--log_dir", str(tmp_path / "logs"),
+        "--model", "SSLResNet18", "--strategy", "BADGESampler"])
+    s = main(args)
+    assert s.idxs_lb.sum() == 5 + 5 * 4
